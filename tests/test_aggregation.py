@@ -1,0 +1,191 @@
+"""Aggregation oracle vs reference expectations + numpy brute force.
+
+CKMS expectations from src/aggregator/aggregation/quantile/cm/stream_test.go;
+counter/gauge/timer semantics from aggregation/{counter,gauge,timer}.go;
+window semantics from aggregator/{generic_elem,list}.go (cites in the oracle).
+"""
+import numpy as np
+import pytest
+
+import oracle
+
+TQ = [0.5, 0.9, 0.99]
+
+
+def test_ckms_empty_one_three():
+    q, mn, mx = oracle.ckms_quantiles([], TQ, eps=0.01)
+    assert list(q) == [0, 0, 0] and mn == 0 and mx == 0
+    q, mn, mx = oracle.ckms_quantiles([100.0], TQ, eps=0.01)
+    assert list(q) == [100, 100, 100] and mn == 100 and mx == 100
+    q, mn, mx = oracle.ckms_quantiles([-100.0], TQ, eps=0.01)
+    assert list(q) == [-100, -100, -100]
+    # stream_test.go:84-98: {100,200,300} -> p50=200, p90=p99=300
+    q, mn, mx = oracle.ckms_quantiles([100.0, 200.0, 300.0], TQ, eps=0.01)
+    assert mn == 100 and mx == 300 and list(q) == [200.0, 300.0, 300.0]
+
+
+@pytest.mark.parametrize("every", [10**9, 100])
+@pytest.mark.parametrize("dist", ["inc", "dec", "rand"])
+def test_ckms_eps_bounds_100k(dist, every):
+    """stream_test.go:155-230: quantile rank error <= eps*n at n=100k."""
+    n = 100000
+    if dist == "inc":
+        vals = np.arange(float(n))
+    elif dist == "dec":
+        vals = np.arange(float(n))[::-1]
+    else:
+        vals = np.random.default_rng(7).random(n) * 1e6
+    q, mn, mx = oracle.ckms_quantiles(vals, TQ, eps=0.01, every=every)
+    s = np.sort(vals)
+    assert mn == s[0] and mx == s[-1]
+    for qi, qq in zip(q, TQ):
+        rank = np.searchsorted(s, qi)
+        assert abs(rank - qq * n) <= n * 0.01 + 1, (dist, every, qq)
+
+
+def ckms_small_n_expected(sorted_vals, qs_sorted):
+    """Reference CKMS result for n <= ~250 samples under the m3 production
+    defaults (eps=1e-3, insertAndCompressEvery=1024): no compression occurs
+    (every merge threshold is 0 < testVal) so the sample list is the sorted
+    values with numRanks=1, delta=0, and calcQuantiles (stream.go:231-277)
+    reduces to this walk. NB: the walk emits at most ONE quantile per sample,
+    so colliding ranks mid-walk shift later quantiles one sample right —
+    exactly the reference behavior, reproduced by the GPU rollup kernel.
+      n <= 3: sorted[min(int(q*n), n-1)]  (quantilesFromBuf :210-229)."""
+    s = sorted_vals
+    n = len(s)
+    out = np.empty(len(qs_sorted))
+    if n == 0:
+        return out * 0
+    if n <= 3:
+        for i, q in enumerate(qs_sorted):
+            out[i] = s[min(int(q * n), n - 1)]
+        return out
+    ranks = [int(np.ceil(q * n)) for q in qs_sorted]  # thresholds are 0
+    idx = 0
+    min_rank = 0
+    prev = 0
+    max_rank = 0
+    for k in range(n):  # curr = sample k
+        if idx >= len(ranks):
+            break
+        max_rank = min_rank + 1
+        if max_rank > ranks[idx] or min_rank > ranks[idx]:
+            out[idx] = s[prev]
+            idx += 1
+        min_rank += 1
+        prev = k
+    for i in range(idx, len(ranks)):
+        if max_rank >= ranks[i] or min_rank > ranks[i]:
+            out[i] = s[prev]
+    return out
+
+
+def test_ckms_small_n_exact_walk():
+    """With the m3 production defaults and n <= 64, CKMS quantiles equal the
+    exact no-compression walk above. This is the contract the fused GPU
+    rollup kernel implements."""
+    rng = np.random.default_rng(3)
+    qs = [0.1, 0.25, 0.5, 0.75, 0.95, 0.99, 0.999]
+    for _ in range(800):
+        n = int(rng.integers(1, 65))
+        vals = rng.random(n) * 1000
+        q, mn, mx = oracle.ckms_quantiles(vals, qs)
+        s = np.sort(vals)
+        assert mn == s[0] and mx == s[-1]
+        exp = ckms_small_n_expected(s, qs)
+        assert np.array_equal(q, exp), (n, list(q), list(exp))
+
+
+WINDOW = 60 * 10**9  # 1m buckets
+START = (1427162462 * 10**9 // WINDOW) * WINDOW
+
+
+def _mkbatch(rng, nseries=8, npts=240, cadence_s=10):
+    ts = START + np.arange(npts, dtype=np.int64) * cadence_s * 10**9
+    ts = np.broadcast_to(ts, (nseries, npts)).copy()
+    vals = np.round(rng.random((nseries, npts)) * 1e4, 3)
+    counts = np.full(nseries, npts, np.uint32)
+    return ts, vals, counts
+
+
+def test_rollup_gauge_brute_force():
+    rng = np.random.default_rng(21)
+    ts, vals, counts = _mkbatch(rng)
+    vals[0, 5] = np.nan  # NaN rules: count yes, sum/min/max skip
+    nbuckets = 40
+    aggs = ["last", "min", "max", "mean", "count", "sum", "sumsq", "stdev"]
+    out, wts = oracle.rollup_batch(ts, vals, counts, oracle.METRIC_GAUGE,
+                                   WINDOW, nbuckets, aggs)
+    npts_per = WINDOW // (10 * 10**9)  # 6
+    for i in range(ts.shape[0]):
+        for b in range(nbuckets):
+            sel = vals[i, b * npts_per:(b + 1) * npts_per]
+            good = sel[~np.isnan(sel)]
+            row = dict(zip(aggs, out[i, b]))
+            assert wts[i, b] == START + (b + 1) * WINDOW  # window END ts
+            assert row["last"] == sel[-1] or (np.isnan(sel[-1]) and row["last"] != row["last"])
+            assert row["count"] == len(sel)
+            assert row["min"] == good.min() and row["max"] == good.max()
+            assert row["sum"] == np.sum(good, dtype=np.float64) or abs(row["sum"] - good.sum()) < 1e-9
+            # mean = sum/count counts NaN points in the denominator (gauge.go:108-113)
+            assert row["mean"] == row["sum"] / len(sel)
+            n = len(sel)
+            num = n * row["sumsq"] - row["sum"] ** 2
+            exp_stdev = np.sqrt(num / (n * (n - 1))) if n > 1 else 0.0
+            assert abs(row["stdev"] - exp_stdev) < 1e-9
+
+
+def test_rollup_counter_brute_force():
+    rng = np.random.default_rng(22)
+    ts, vals, counts = _mkbatch(rng)
+    vals = np.floor(vals)  # integral counter values
+    nbuckets = 40
+    aggs = ["sum", "min", "max", "count", "mean", "sumsq", "stdev"]
+    out, _ = oracle.rollup_batch(ts, vals, counts, oracle.METRIC_COUNTER,
+                                 WINDOW, nbuckets, aggs)
+    per = 6
+    for i in range(4):
+        for b in range(nbuckets):
+            sel = vals[i, b * per:(b + 1) * per].astype(np.int64)
+            row = dict(zip(aggs, out[i, b]))
+            assert row["sum"] == sel.sum()
+            assert row["min"] == sel.min() and row["max"] == sel.max()
+            assert row["count"] == len(sel)
+            assert row["sumsq"] == float((sel.astype(object) ** 2).sum())
+
+
+def test_rollup_timer_quantiles_exact():
+    """Timer p50/p95/p99 on 6-sample 1m buckets == the verified small-n CKMS
+    semantics: sorted[ceil(q*n)] 1-indexed."""
+    rng = np.random.default_rng(23)
+    ts, vals, counts = _mkbatch(rng)
+    nbuckets = 40
+    aggs = ["sum", "mean", "count", "min", "max", "median", "p95", "p99", "stdev"]
+    out, _ = oracle.rollup_batch(ts, vals, counts, oracle.METRIC_TIMER,
+                                 WINDOW, nbuckets, aggs)
+    per = 6
+    for i in range(4):
+        for b in range(nbuckets):
+            sel = np.sort(vals[i, b * per:(b + 1) * per])
+            row = dict(zip(aggs, out[i, b]))
+            n = len(sel)
+            assert row["min"] == sel[0] and row["max"] == sel[-1]
+            assert row["median"] == sel[int(np.ceil(0.5 * n)) - 1]
+            assert row["p95"] == sel[min(int(np.ceil(0.95 * n)), n) - 1]
+            assert row["p99"] == sel[min(int(np.ceil(0.99 * n)), n) - 1]
+            assert abs(row["sum"] - sel.sum()) < 1e-9
+            assert row["count"] == n
+
+
+def test_rollup_gauge_last_equal_timestamps():
+    """gauge.go:87-92: last only replaced when timestamp.After(lastAt) —
+    equal timestamps keep the FIRST value."""
+    ts = np.array([[START, START, START + WINDOW]], dtype=np.int64)
+    vals = np.array([[1.0, 2.0, 3.0]])
+    counts = np.array([3], np.uint32)
+    out, _ = oracle.rollup_batch(ts, vals, counts, oracle.METRIC_GAUGE,
+                                 WINDOW, 2, ["last", "count"])
+    assert out[0, 0, 0] == 1.0  # first wins on equal ts
+    assert out[0, 0, 1] == 2.0
+    assert out[0, 1, 0] == 3.0
