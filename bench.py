@@ -1,0 +1,233 @@
+#!/usr/bin/env python3
+"""bench.py — headline benchmark: batched M3TSZ decode on MI355X.
+
+Measures BASELINE.json's metric ("datapoints/sec M3TSZ decode (1M series x
+1440 pts) + HBM GB/s vs peak") on BASELINE.json configs[1] — the largest
+single-GPU configuration — with synthetic data (m3_amd/workload.py shapes,
+SURVEY.md §8d) encoded on-device by the product encoder. A "step" = one
+batched decode of the full resident batch into SoA (ts, val) rows.
+
+Multi-GPU (--gpus N, launched via torch.distributed.run): the series shard
+trivially (independent series, zero exchange — SURVEY.md §8e): each rank
+owns nseries/N series, weak scaling, no data-path collective; value is the
+whole-job aggregate.
+
+cpu_baseline: the oracle (C restatement of the reference decoder; the only
+permitted oracle use in this file) timed on host cores over a bounded sample
+of the same streams. roofline: HIP-event kernel timing of the dominant
+(decode) kernel vs the 8 TB/s HBM3E peak.
+"""
+import argparse
+import json
+import os
+import sys
+import time
+
+import numpy as np
+
+REPO = os.path.dirname(os.path.abspath(__file__))
+sys.path.insert(0, REPO)
+
+HBM_PEAK_GBS = 8000.0  # 8 TB/s spec peak (MI355X_MICROARCH.md)
+
+
+def shard_range(nseries, world, rank):
+    """Contiguous series range owned by `rank` (equal counts; series are
+    statically sharded — no exchange, mirroring the reference's shard-hash
+    data distribution at aggregator/sharding/hash.go:89)."""
+    per = nseries // world
+    extra = nseries % world
+    lo = rank * per + min(rank, extra)
+    hi = lo + per + (1 if rank < extra else 0)
+    return lo, hi
+
+
+def parse_args():
+    p = argparse.ArgumentParser()
+    p.add_argument("--gpus", type=int, default=1)
+    p.add_argument("--steps", type=int, default=10)
+    p.add_argument("--warmup", type=int, default=3)
+    p.add_argument("--nseries", type=int, default=1_000_000)
+    p.add_argument("--npts", type=int, default=1440)
+    p.add_argument("--chunk", type=int, default=65536)
+    p.add_argument("--cpu-sample-series", type=int, default=0,
+                   help="series in the cpu_baseline sample (0 = auto-size to ~15s)")
+    p.add_argument("--skip-cpu-baseline", action="store_true")
+    return p.parse_args()
+
+
+def main():
+    args = parse_args()
+    import torch
+    import m3_amd
+    from m3_amd import engine, workload
+
+    rank = int(os.environ.get("RANK", "0"))
+    world = int(os.environ.get("WORLD_SIZE", "1"))
+    local_rank = int(os.environ.get("LOCAL_RANK", str(rank)))
+    dist = None
+    if world > 1:
+        import torch.distributed as dist_mod
+        dist = dist_mod
+        dist.init_process_group("nccl")
+    device = f"cuda:{local_rank}"
+    torch.cuda.set_device(device)
+
+    lo, hi = shard_range(args.nseries, world, rank)
+    n_local = hi - lo
+    t0 = time.time()
+    d_blob, d_offsets, d_lens, enc_bytes = workload.encode_on_device(
+        torch, n_local, args.npts, chunk=args.chunk, device=device,
+        rank_offset=lo, verbose=(rank == 0))
+    torch.cuda.synchronize()
+    if rank == 0:
+        print(f"[bench] built {n_local} series x {args.npts} pts: "
+              f"{enc_bytes/1e9:.3f} GB encoded "
+              f"({enc_bytes/(n_local*args.npts):.2f} B/pt) "
+              f"in {time.time()-t0:.1f}s", flush=True)
+
+    npts_total_local = n_local * args.npts
+    out_ts = torch.empty((n_local, args.npts), dtype=torch.int64, device=device)
+    out_vals = torch.empty((n_local, args.npts), dtype=torch.float64, device=device)
+    out_counts = torch.empty(n_local, dtype=torch.int32, device=device)
+    out_errs = torch.empty(n_local, dtype=torch.int32, device=device)
+
+    def step():
+        engine.decode_batch_dev(d_blob, d_offsets, d_lens, out_ts, out_vals,
+                                out_counts, out_errs)
+
+    # correctness gate outside the timed region: every series decodes fully
+    step()
+    torch.cuda.synchronize()
+    if int(out_errs.abs().sum().item()) != 0 or \
+       int((out_counts != args.npts).sum().item()) != 0:
+        raise SystemExit("[bench] decode errors in workload — aborting")
+
+    for _ in range(args.warmup):
+        step()
+    torch.cuda.synchronize()
+
+    # timed region: K steps, barrier + sync on both sides, per-launch HIP
+    # events on the decode kernel for the roofline
+    ev = [(torch.cuda.Event(enable_timing=True), torch.cuda.Event(enable_timing=True))
+          for _ in range(args.steps)]
+    if dist:
+        dist.barrier()
+    torch.cuda.synchronize()
+    wall0 = time.time()
+    for k in range(args.steps):
+        ev[k][0].record()
+        step()
+        ev[k][1].record()
+    torch.cuda.synchronize()
+    wall1 = time.time()
+    if dist:
+        dist.barrier()
+
+    elapsed = wall1 - wall0
+    if dist:
+        t = torch.tensor([elapsed], device=device)
+        dist.all_reduce(t, op=dist.ReduceOp.MAX)
+        elapsed = float(t.item())
+
+    kernel_ms = float(np.mean([a.elapsed_time(b) for a, b in ev]))
+    npts_global = args.nseries * args.npts
+    dps = npts_global * args.steps / elapsed  # datapoints/sec, whole job
+
+    # roofline on the dominant kernel (decode): algorithmic bytes per launch
+    algo_bytes = enc_bytes + 16 * npts_total_local  # stream in + (ts,val) out
+    achieved_gbs = algo_bytes / (kernel_ms / 1e3) / 1e9
+    read_only_gbs = enc_bytes / (kernel_ms / 1e3) / 1e9
+    traffic = None
+    tpath = os.path.join(REPO, "profiles", "traffic.json")
+    if rank == 0 and os.path.exists(tpath):
+        try:
+            with open(tpath) as f:
+                tj = json.load(f)
+            if tj.get("kernel") == "k_decode_batch" and \
+               tj.get("nseries") == n_local and tj.get("npts") == args.npts:
+                traffic = tj.get("bytes_per_launch")
+        except Exception:
+            pass
+
+    cpu_baseline = None
+    if rank == 0 and not args.skip_cpu_baseline:
+        cpu_baseline = run_cpu_baseline(args, d_blob, d_offsets, d_lens, n_local)
+
+    if rank == 0:
+        line = {
+            "metric": "datapoints/sec M3TSZ decode (1M series x 1440 pts) + HBM GB/s vs peak",
+            "value": dps,
+            "unit": "datapoints/sec",
+            "n_gpus": world,
+            "steps": args.steps,
+            "warmup": args.warmup,
+            "ms_per_step": elapsed / args.steps * 1e3,
+            "higher_is_better": True,
+            "scaling": "weak",
+            "vs_baseline": None,
+            "dtype": "f64",
+            "data": "synthetic",
+            "config": {
+                "workload": "1M series x 1440 pts batched M3TSZ decode, HBM-bandwidth microbench",
+                "nseries": args.nseries,
+                "npts": args.npts,
+                "encoded_bytes_per_pt": enc_bytes / npts_total_local,
+                "parallelism": f"series-sharded x{world} (no collectives)",
+            },
+            "roofline": {
+                "bound": "hbm",
+                "achieved": achieved_gbs,
+                "peak": HBM_PEAK_GBS,
+                "unit": "GB/s",
+                "frac": achieved_gbs / HBM_PEAK_GBS,
+                "read_only_gbs": read_only_gbs,
+                "read_only_frac": read_only_gbs / HBM_PEAK_GBS,
+                "traffic": traffic,
+            },
+            "cpu_baseline": cpu_baseline,
+        }
+        print(json.dumps(line), flush=True)
+
+    if dist:
+        dist.destroy_process_group()
+
+
+def run_cpu_baseline(args, d_blob, d_offsets, d_lens, n_local):
+    """Oracle (C restatement of the reference decoder, kind='port') on the
+    host cores, over a bounded sample of the same streams (~10-30 s)."""
+    import oracle  # permitted: cpu_baseline leg only
+
+    cores = os.cpu_count()
+    # probe with 2k series to size the sample
+    probe_n = min(2000, n_local)
+    off = d_offsets[:probe_n + 1].cpu().numpy().astype(np.uint64)
+    lens = d_lens[:probe_n].cpu().numpy().astype(np.uint32)
+    blob = d_blob[: int(off[-1])].cpu().numpy()
+    t0 = time.time()
+    oracle.decode_batch(blob, off, stride=args.npts, nthreads=cores)
+    probe_t = time.time() - t0
+    target_series = probe_n
+    if args.cpu_sample_series:
+        target_series = min(args.cpu_sample_series, n_local)
+    else:
+        per_series = probe_t / probe_n
+        target_series = int(min(n_local, max(probe_n, 15.0 / per_series)))
+    off = d_offsets[:target_series + 1].cpu().numpy().astype(np.uint64)
+    lens = d_lens[:target_series].cpu().numpy().astype(np.uint32)
+    blob = d_blob[: int(off[-1])].cpu().numpy()
+    t0 = time.time()
+    _, _, counts = oracle.decode_batch(blob, off, stride=args.npts, nthreads=cores)
+    dt = time.time() - t0
+    pts = int(counts.astype(np.int64).sum())
+    return {
+        "value": pts / dt,
+        "unit": "datapoints/sec",
+        "cores": cores,
+        "kind": "port",
+        "sample": f"{target_series} of {n_local} series ({pts} pts) in {dt:.1f}s",
+    }
+
+
+if __name__ == "__main__":
+    main()

@@ -1,0 +1,149 @@
+/*
+ * m3gpu.h — C ABI of the MI355X-native M3TSZ + rollup engine (libm3gpu.so).
+ *
+ * This is the drop-in boundary for the reference's pluggable codec/rollup
+ * hot path. On the Go side the reference injects codec constructors into
+ * pools (src/dbnode/storage/options.go:498-510, dbnode/server/server.go:
+ * 1786-1793, dbnode/client/options.go:504, query/pools/query_pools.go:208)
+ * and creates per-window aggregations via typeSpecificElemBase.NewAggregation
+ * (aggregator/aggregator/generic_elem.go:94-95). A cgo wrapper implementing
+ * encoding.Encoder / encoding.ReaderIterator (dbnode/encoding/types.go:39-96,
+ * :197-203) and the aggregation batch path binds to the entry points below —
+ * see INTEGRATION.md for the cgo stubs a maintainer would add.
+ *
+ * Conventions:
+ *  - All functions return 0 on success, a negative M3GPU_ERR_* on failure;
+ *    m3gpu_last_error() gives a thread-local message.
+ *  - `_dev` entry points take DEVICE pointers and an optional hipStream_t
+ *    (as void*); they enqueue async work (caller synchronizes the stream).
+ *    Host-pointer convenience forms (no suffix) do alloc+H2D+kernel+D2H
+ *    internally and block — these are what a cgo caller uses directly.
+ *  - Encoded streams are the reference's on-disk M3TSZ block format
+ *    (src/dbnode/encoding/m3tsz, incl. the EOS-marker tail), bit-exact.
+ *  - Packed stream layout: blobs[offsets[i] .. offsets[i]+lens[i]) is series
+ *    i's stream. offsets[i] MUST be 8-byte aligned (pad between streams with
+ *    zero bytes; the total buffer padded to a multiple of 8). lens[] are the
+ *    true stream lengths.
+ *  - Time units use the reference's xtime.Unit byte values
+ *    (src/x/time/unit.go:30-42): 1=s, 2=ms, 3=us, 4=ns.
+ */
+#ifndef M3GPU_H
+#define M3GPU_H
+
+#include <stdint.h>
+
+#ifdef __cplusplus
+extern "C" {
+#endif
+
+enum {
+    M3GPU_OK = 0,
+    M3GPU_ERR_HIP = -1,        /* HIP runtime failure (see m3gpu_last_error) */
+    M3GPU_ERR_BADARG = -2,     /* invalid argument */
+    /* per-series error codes written to out_errs[i] (positive) */
+    M3GPU_SERIES_OK = 0,
+    M3GPU_SERIES_EOF = 1,          /* truncated stream */
+    M3GPU_SERIES_DOD_OVERFLOW = 2, /* timestamp_encoder.go:216-221 */
+    M3GPU_SERIES_NO_SCHEME = 3,    /* errNoTimeSchemaForUnit */
+    M3GPU_SERIES_INVALID_MULT = 4, /* errInvalidMultiplier */
+    M3GPU_SERIES_ANNOTATION = 5,   /* bad annotation */
+    M3GPU_SERIES_CAPACITY = 6,     /* stride/out buffer too small */
+    M3GPU_SERIES_UNSORTED = 7,     /* rollup input crosses buckets backwards */
+    M3GPU_SERIES_BUCKET_OVERFLOW = 8, /* >64 values in one rollup bucket */
+};
+
+/* metric types for the rollup entry (aggregator/aggregation) */
+enum { M3GPU_METRIC_COUNTER = 0, M3GPU_METRIC_GAUGE = 1, M3GPU_METRIC_TIMER = 2 };
+
+/* aggregation type ids — the reference's metrics/aggregation Type enum
+ * (src/metrics/aggregation/type.go:31-56) */
+enum {
+    M3GPU_AGG_LAST = 1, M3GPU_AGG_MIN = 2, M3GPU_AGG_MAX = 3, M3GPU_AGG_MEAN = 4,
+    M3GPU_AGG_MEDIAN = 5, M3GPU_AGG_COUNT = 6, M3GPU_AGG_SUM = 7,
+    M3GPU_AGG_SUMSQ = 8, M3GPU_AGG_STDEV = 9,
+    M3GPU_AGG_P10 = 10, M3GPU_AGG_P20 = 11, M3GPU_AGG_P30 = 12, M3GPU_AGG_P40 = 13,
+    M3GPU_AGG_P50 = 14, M3GPU_AGG_P60 = 15, M3GPU_AGG_P70 = 16, M3GPU_AGG_P80 = 17,
+    M3GPU_AGG_P90 = 18, M3GPU_AGG_P95 = 19, M3GPU_AGG_P99 = 20,
+    M3GPU_AGG_P999 = 21, M3GPU_AGG_P9999 = 22, M3GPU_AGG_P25 = 23, M3GPU_AGG_P75 = 24,
+};
+
+int m3gpu_init(int device);
+void m3gpu_shutdown(void);
+const char* m3gpu_last_error(void);
+
+/* -------- batched decode (replaces m3tsz ReaderIterator bulk reads:
+ * src/dbnode/encoding/m3tsz/iterator.go:81-219 over xio.Reader64) --------
+ * One series per wavefront. Outputs SoA rows: series i writes
+ * out_ts[i*stride .. i*stride+count) and out_vals likewise;
+ * out_counts[i] = decoded points; out_errs[i] = M3GPU_SERIES_*. */
+int m3gpu_decode_batch_dev(
+    const uint8_t* d_blobs, const uint64_t* d_offsets, const uint32_t* d_lens,
+    uint32_t nseries, int int_optimized, uint8_t default_unit,
+    int64_t* d_out_ts, double* d_out_vals, uint32_t* d_out_counts,
+    int32_t* d_out_errs, uint32_t stride, void* hip_stream);
+
+int m3gpu_decode_batch(
+    const uint8_t* blobs, uint64_t blobs_len,
+    const uint64_t* offsets, const uint32_t* lens,
+    uint32_t nseries, int int_optimized, uint8_t default_unit,
+    int64_t* out_ts, double* out_vals, uint32_t* out_counts,
+    int32_t* out_errs, uint32_t stride);
+
+/* -------- batched encode (replaces m3tsz.Encoder bulk writes:
+ * src/dbnode/encoding/m3tsz/encoder.go:89-250) --------
+ * Series i encodes counts[i] points from ts/vals rows (stride elements per
+ * row); start time = first timestamp (as dbnode series buffers do). Output:
+ * stream bytes at d_out_bytes + i*out_stride (finalized, EOS tail included),
+ * length in out_lens[i]. out_stride must be a multiple of 8 and hold the
+ * worst case (~20 B/pt + 16). Fixed unit, no annotations (bulk path). */
+int m3gpu_encode_batch_dev(
+    const int64_t* d_ts, const double* d_vals, const uint32_t* d_counts,
+    uint32_t nseries, uint32_t stride, int int_optimized, uint8_t unit,
+    uint8_t* d_out_bytes, uint32_t out_stride, uint32_t* d_out_lens,
+    int32_t* d_out_errs, void* hip_stream);
+
+int m3gpu_encode_batch(
+    const int64_t* ts, const double* vals, const uint32_t* counts,
+    uint32_t nseries, uint32_t stride, int int_optimized, uint8_t unit,
+    uint8_t* out_bytes, uint32_t out_stride, uint32_t* out_lens,
+    int32_t* out_errs);
+
+/* Pack strided encoder output into the tight 8B-aligned blob layout:
+ * series i's first lens[i] bytes (rounded up to whole 8B words, whose pad
+ * bytes the encoder zeroes) move from d_src + i*src_stride to
+ * d_dst + d_dst_offsets[i]. d_dst must be pre-zeroed. */
+int m3gpu_compact_dev(
+    const uint8_t* d_src, uint32_t src_stride, const uint32_t* d_lens,
+    const uint64_t* d_dst_offsets, uint32_t nseries, uint8_t* d_dst,
+    void* hip_stream);
+
+/* -------- fused decode -> windowed rollup (replaces decode +
+ * elem.AddValue/Consume: aggregator/aggregator/generic_elem.go:219-235,
+ * 424-485 with aggregation/{counter,gauge,timer}.go semantics) --------
+ * Decodes each stream and aggregates into ceil-aligned windows of window_ns:
+ * bucket b covers [base + b*window, base + (b+1)*window) where base =
+ * truncate(first timestamp). Emits naggs doubles per bucket in agg_types[]
+ * order (d_out[i*nbuckets*naggs + b*naggs + k]) and the window-END timestamp
+ * (list.go:541-543) in d_out_window_ts[i*nbuckets + b]. Quantiles use the
+ * production CKMS semantics (exact for <=64 values per bucket; more sets
+ * M3GPU_SERIES_BUCKET_OVERFLOW). agg_types is a HOST pointer (naggs <= 16). */
+int m3gpu_rollup_batch_dev(
+    const uint8_t* d_blobs, const uint64_t* d_offsets, const uint32_t* d_lens,
+    uint32_t nseries, int int_optimized, uint8_t default_unit,
+    int metric_type, int64_t window_ns, uint32_t nbuckets,
+    const int32_t* agg_types, int naggs,
+    double* d_out, int64_t* d_out_window_ts, int32_t* d_out_errs,
+    void* hip_stream);
+
+int m3gpu_rollup_batch(
+    const uint8_t* blobs, uint64_t blobs_len,
+    const uint64_t* offsets, const uint32_t* lens,
+    uint32_t nseries, int int_optimized, uint8_t default_unit,
+    int metric_type, int64_t window_ns, uint32_t nbuckets,
+    const int32_t* agg_types, int naggs,
+    double* out, int64_t* out_window_ts, int32_t* out_errs);
+
+#ifdef __cplusplus
+}
+#endif
+#endif

@@ -1,0 +1,1475 @@
+/*
+ * m3tsz_kernels.hip — MI355X (gfx950/CDNA4) native M3TSZ block codec and
+ * fused decode->rollup engine, plus the C-ABI host layer (include/m3gpu.h).
+ *
+ * This is a from-scratch GPU design of the reference's hot path (reference
+ * cites per function; format: src/dbnode/encoding/m3tsz + aggregator
+ * semantics), NOT a port of its Go code:
+ *
+ *  - One SERIES PER WAVEFRONT. M3TSZ is a sequentially-dependent
+ *    variable-length code, so all 64 lanes run the (wave-uniform) bit parser
+ *    redundantly — branches cost s_branch, not divergence — and the wave's
+ *    parallelism is spent on the memory system:
+ *      decode: lane (n & 63) captures point n in registers; every 64 points
+ *        the wave stores 64x(8B ts + 8B val) contiguously — fully coalesced
+ *        512B transactions. The compressed stream is read as aligned
+ *        big-endian u64 words at one wave-uniform address (L1 broadcast).
+ *      encode: lane (w & 63) captures output word w; every 64 words the wave
+ *        stores 512B coalesced.
+ *      rollup: bucket values are staged to LDS, quantiles picked by a
+ *        rank-select over lanes (no sort network needed), reproducing the
+ *        reference CKMS walk exactly (see ckms_small_n semantics below).
+ *  - Streams are decoded from 8-byte-aligned offsets (layout contract in
+ *    m3gpu.h) so every refill is one aligned dwordx2 load.
+ *  - Wave64 only; no CUDA shims, no hipified code.
+ *
+ * Compile: hipcc --offload-arch=gfx950 -O3 -ffp-contract=off (bit-exact f64:
+ * convertToIntFloat m3tsz.go:78-119, decode accumulation iterator.go:168-175).
+ */
+#include <hip/hip_runtime.h>
+#include <math.h>
+#include <string.h>
+#include <stdio.h>
+#include "../../include/m3gpu.h"
+
+/* ========================= shared constants ========================= */
+/* m3tsz.go:28-62, scheme.go:28-52, x/time/unit.go:30-42 */
+
+#define WAVE 64
+#define WAVES_PER_BLOCK 4
+#define BLOCK_THREADS (WAVE * WAVES_PER_BLOCK)
+
+namespace m3 {
+
+__device__ __constant__ int64_t UNIT_NS_D[9] = {
+    0, 1000000000LL, 1000000LL, 1000LL, 1LL,
+    60000000000LL, 3600000000000LL, 86400000000000LL, 31536000000000000LL};
+
+#define MARKER_OPCODE 0x100ULL
+#define MARKER_BITS 11 /* 9 opcode + 2 value */
+#define MARKER_EOS 0
+#define MARKER_ANNOTATION 1
+#define MARKER_TIMEUNIT 2
+
+#define MAX_MULT 6
+#define NUM_SIG_BITS 6
+#define NUM_MULT_BITS 3
+#define SIG_DIFF_THRESHOLD 3
+#define SIG_REPEAT_THRESHOLD 5
+
+__device__ __forceinline__ int unit_valid(uint8_t u) { return u > 0 && u < 9; }
+
+/* scheme.go:42-52: {7,9,12} bucket value bits; default 32 (s/ms) / 64 (us/ns) */
+__device__ __forceinline__ int scheme_default_bits(uint8_t unit) {
+    if (unit == 1 || unit == 2) return 32;
+    if (unit == 3 || unit == 4) return 64;
+    return 0;
+}
+
+__device__ __forceinline__ uint8_t num_sig(uint64_t v) { /* encoding.go:29-31 */
+    return (uint8_t)(64 - (v ? __builtin_clzll(v) : 64));
+}
+__device__ __forceinline__ int64_t sign_extend(uint64_t v, uint32_t nbits) {
+    uint32_t sh = 64 - nbits;
+    return ((int64_t)(v << sh)) >> sh;
+}
+/* Go float64->int64 (amd64 CVTTSD2SQ): out-of-range/NaN -> INT64_MIN */
+__device__ __forceinline__ int64_t go_f2i(double v) {
+    if (!(v >= -9223372036854775808.0 && v < 9223372036854775808.0)) return INT64_MIN;
+    return (int64_t)v;
+}
+__device__ __forceinline__ uint64_t f2bits(double v) { return __double_as_longlong(v); }
+__device__ __forceinline__ double bits2f(uint64_t b) { return __longlong_as_double((long long)b); }
+
+/* Go math.Modf: Modf(+-Inf) = (+-Inf, NaN) */
+__device__ __forceinline__ double go_modf(double v, double* ip) {
+    if (isinf(v)) { *ip = v; return __longlong_as_double(0x7ff8000000000000LL); }
+    return modf(v, ip);
+}
+
+/* ===================== device bit reader ===================== */
+/* istream.go:73-115 over reader64.go:40-80, with the m3gpu.h layout
+ * contract: stream starts 8B-aligned, buffer zero-padded to 8B. */
+
+struct BitReader {
+    const uint64_t* words; /* aligned start of this stream */
+    int64_t len;           /* true byte length */
+    int64_t index;         /* byte index of next word (multiple of 8) */
+    uint64_t current;      /* left-aligned buffered bits */
+    uint32_t remaining;    /* valid bits in current */
+
+    __device__ void init(const uint8_t* base, uint64_t off, uint32_t l) {
+        words = (const uint64_t*)(base + off);
+        len = l;
+        index = 0;
+        current = 0;
+        remaining = 0;
+    }
+    /* reader64.Read64: word + bit count; zero-padded tail comes for free */
+    __device__ __forceinline__ int read64(uint64_t* w, uint32_t* nbits) {
+        if (index >= len) return M3GPU_SERIES_EOF;
+        uint64_t v = __builtin_bswap64(words[index >> 3]);
+        int64_t avail = len - index;
+        if (avail >= 8) { *nbits = 64; }
+        else { *nbits = (uint32_t)(8 * avail); }
+        index += 8;
+        *w = v;
+        return 0;
+    }
+    __device__ __forceinline__ int read_bits(uint32_t n, uint64_t* out) {
+        uint64_t res = n ? (current >> (64 - n)) : 0;
+        if (n <= remaining) {
+            current = (n >= 64) ? 0 : (current << n);
+            remaining -= n;
+            *out = res;
+            return 0;
+        }
+        uint32_t needed = n - remaining;
+        uint64_t w; uint32_t nb;
+        int err = read64(&w, &nb);
+        if (err) return err;
+        if (nb < needed) return M3GPU_SERIES_EOF;
+        current = (needed >= 64) ? 0 : (w << needed);
+        remaining = nb - needed;
+        *out = res | (w >> (64 - needed));
+        return 0;
+    }
+    __device__ __forceinline__ int peek_bits(uint32_t n, uint64_t* out) {
+        if (n <= remaining) { *out = n ? (current >> (64 - n)) : 0; return 0; }
+        uint64_t res = n ? (current >> (64 - n)) : 0;
+        uint32_t needed = n - remaining;
+        if (index >= len) return M3GPU_SERIES_EOF;
+        uint64_t w = __builtin_bswap64(words[index >> 3]);
+        int64_t avail = len - index;
+        uint32_t nb = avail >= 8 ? 64 : (uint32_t)(8 * avail);
+        if (nb < needed) return M3GPU_SERIES_EOF;
+        *out = res | (w >> (64 - needed));
+        return 0;
+    }
+};
+
+/* ===================== device decoder state ===================== */
+/* Port of the reference decode state machine: iterator.go:47-219,
+ * timestamp_iterator.go:41-361, float_encoder_iterator.go:105-165. */
+
+struct Decoder {
+    BitReader r;
+    int64_t prev_time, prev_time_delta;
+    double int_val;
+    uint64_t prev_float_bits, prev_xor;
+    uint8_t time_unit, scheme_unit, mult, sig;
+    uint8_t default_unit;
+    bool have_scheme, tu_changed, done, is_float;
+    bool int_optimized;
+
+    __device__ void init(const uint8_t* base, uint64_t off, uint32_t len,
+                         bool intopt, uint8_t dunit) {
+        r.init(base, off, len);
+        prev_time = 0; prev_time_delta = 0;
+        int_val = 0; prev_float_bits = 0; prev_xor = 0;
+        time_unit = 0; scheme_unit = 0; mult = 0; sig = 0;
+        default_unit = dunit;
+        have_scheme = false; tu_changed = false; done = false; is_float = false;
+        int_optimized = intopt;
+    }
+
+    /* timestamp_iterator.go:115-135 */
+    __device__ int read_time_unit() {
+        uint64_t tu_bits;
+        int err = r.read_bits(8, &tu_bits);
+        if (err) return err;
+        uint8_t tu = (uint8_t)tu_bits;
+        if (unit_valid(tu) && tu != time_unit) {
+            tu_changed = true;
+            if (scheme_default_bits(tu)) { have_scheme = true; scheme_unit = tu; }
+        }
+        time_unit = tu;
+        return 0;
+    }
+
+    /* binary.ReadVarint via ReadByte */
+    __device__ int read_varint(int64_t* out) {
+        uint64_t ux = 0;
+        int shift = 0;
+        for (int i = 0; i < 10; i++) {
+            uint64_t b;
+            int err = r.read_bits(8, &b);
+            if (err) return err;
+            ux |= (b & 0x7f) << shift;
+            if (!(b & 0x80)) {
+                int64_t x = (int64_t)(ux >> 1);
+                if (ux & 1) x = ~x;
+                *out = x;
+                return 0;
+            }
+            shift += 7;
+        }
+        return M3GPU_SERIES_ANNOTATION;
+    }
+
+    /* timestamp_iterator.go:327-356 (annotation bytes are skipped — the bulk
+     * decode surface does not return annotations) */
+    __device__ int skip_annotation() {
+        int64_t alen;
+        int err = read_varint(&alen);
+        if (err) return err;
+        alen += 1;
+        if (alen <= 0) return M3GPU_SERIES_ANNOTATION;
+        for (int64_t i = 0; i < alen; i++) {
+            uint64_t b;
+            err = r.read_bits(8, &b);
+            if (err) return err;
+        }
+        return 0;
+    }
+
+    /* timestamp_iterator.go:307-325 */
+    __device__ int read_full_timestamp(int64_t* dod) {
+        if (!scheme_default_bits(time_unit)) return M3GPU_SERIES_NO_SCHEME;
+        have_scheme = true;
+        scheme_unit = time_unit;
+        uint64_t bits;
+        int err = r.read_bits(64, &bits);
+        if (err) return err;
+        *dod = (int64_t)bits;
+        return 0;
+    }
+
+    /* timestamp_iterator.go:250-305 */
+    __device__ int read_dod(int64_t* out) {
+        if (tu_changed) return read_full_timestamp(out);
+        if (!have_scheme) return M3GPU_SERIES_NO_SCHEME;
+        uint64_t cb;
+        int err = r.read_bits(1, &cb);
+        if (err) return err;
+        if (cb == 0) { *out = 0; return 0; }
+        const uint32_t opcodes[3] = {0x2, 0x6, 0xe};
+        const uint32_t vbits[3] = {7, 9, 12};
+        for (int i = 0; i < 3; i++) {
+            uint64_t nxt;
+            err = r.read_bits(1, &nxt);
+            if (err) { *out = 0; return 0; } /* swallowed (:271-274) */
+            cb = (cb << 1) | nxt;
+            if (cb == opcodes[i]) {
+                uint64_t db;
+                err = r.read_bits(vbits[i], &db);
+                if (err) return err;
+                if (!unit_valid(time_unit)) { *out = 0; return 0; }
+                *out = sign_extend(db, vbits[i]) * UNIT_NS_D[time_unit];
+                return 0;
+            }
+        }
+        uint32_t dbits = (uint32_t)scheme_default_bits(scheme_unit);
+        uint64_t db;
+        err = r.read_bits(dbits, &db);
+        if (err) return err;
+        if (!unit_valid(time_unit)) { *out = 0; return 0; }
+        *out = sign_extend(db, dbits) * UNIT_NS_D[time_unit];
+        return 0;
+    }
+
+    /* readMarkerOrDeltaOfDelta (:237-248) with tryReadMarker (:174-235)
+     * unrolled into a loop (annotation/timeunit markers chain). */
+    __device__ int read_marker_or_dod(int64_t* out) {
+        for (;;) {
+            uint64_t ov;
+            if (r.peek_bits(MARKER_BITS, &ov) != 0) break; /* peek error => not a marker */
+            if ((ov >> 2) != MARKER_OPCODE) break;
+            uint64_t marker = ov & 0x3;
+            uint64_t discard;
+            int err;
+            if (marker == MARKER_EOS) {
+                err = r.read_bits(MARKER_BITS, &discard);
+                if (err) return err;
+                done = true;
+                *out = 0;
+                return 0;
+            } else if (marker == MARKER_ANNOTATION) {
+                err = r.read_bits(MARKER_BITS, &discard);
+                if (err) return err;
+                err = skip_annotation();
+                if (err) return err;
+                continue;
+            } else if (marker == MARKER_TIMEUNIT) {
+                err = r.read_bits(MARKER_BITS, &discard);
+                if (err) return err;
+                err = read_time_unit();
+                if (err) return err;
+                continue;
+            } else {
+                break; /* unknown marker value: parse as dod (:232-234) */
+            }
+        }
+        return read_dod(out);
+    }
+
+    /* timestamp_iterator.go:137-161 + initialTimeUnit */
+    __device__ int read_first_timestamp() {
+        uint64_t nt_bits;
+        int err = r.read_bits(64, &nt_bits);
+        if (err) return err;
+        int64_t nt = (int64_t)nt_bits;
+        if (time_unit == 0 && unit_valid(default_unit) &&
+            nt % UNIT_NS_D[default_unit] == 0) {
+            time_unit = default_unit;
+        }
+        if (scheme_default_bits(time_unit)) { have_scheme = true; scheme_unit = time_unit; }
+        int64_t dod;
+        err = read_marker_or_dod(&dod);
+        if (err) return err;
+        if (!done) prev_time_delta += dod;
+        prev_time = nt + prev_time_delta;
+        return 0;
+    }
+
+    /* timestamp_iterator.go:80-113 */
+    __device__ int read_timestamp(bool* first) {
+        *first = false;
+        int err;
+        if (prev_time != 0) {
+            int64_t dod;
+            err = read_marker_or_dod(&dod);
+            if (err == 0 && !done) {
+                prev_time_delta += dod;
+                prev_time += prev_time_delta;
+            }
+        } else {
+            *first = true;
+            err = read_first_timestamp();
+        }
+        if (err) return err;
+        if (tu_changed) { prev_time_delta = 0; tu_changed = false; }
+        return 0;
+    }
+
+    /* float_encoder_iterator.go:105-165 */
+    __device__ int read_full_float() {
+        uint64_t vb;
+        int err = r.read_bits(64, &vb);
+        if (err) return err;
+        prev_float_bits = vb;
+        prev_xor = vb;
+        return 0;
+    }
+    __device__ int read_next_float() {
+        uint64_t cb;
+        int err = r.read_bits(1, &cb);
+        if (err) return err;
+        if (cb == 0) { prev_xor = 0; return 0; }
+        uint64_t nxt;
+        err = r.read_bits(1, &nxt);
+        if (err) return err;
+        cb = (cb << 1) | nxt;
+        if (cb == 0x2) { /* contained */
+            uint32_t lead = prev_xor ? __builtin_clzll(prev_xor) : 64;
+            uint32_t trail = prev_xor ? __builtin_ctzll(prev_xor) : 0;
+            uint32_t nmean = 64 - lead - trail;
+            uint64_t mb;
+            err = r.read_bits(nmean, &mb);
+            if (err) return err;
+            prev_xor = mb << trail;
+            prev_float_bits ^= prev_xor;
+            return 0;
+        }
+        uint64_t lm;
+        err = r.read_bits(12, &lm);
+        if (err) return err;
+        uint64_t lead = (lm & 4032) >> 6;
+        uint64_t nmean = (lm & 63) + 1;
+        uint64_t mb;
+        err = r.read_bits((uint32_t)nmean, &mb);
+        if (err) return err;
+        uint64_t trail = 64 - lead - nmean;
+        prev_xor = mb << trail;
+        prev_float_bits ^= prev_xor;
+        return 0;
+    }
+
+    /* iterator.go:178-219 */
+    __device__ int read_int_sig_mult() {
+        uint64_t b;
+        int err = r.read_bits(1, &b);
+        if (err) return err;
+        if (b == 1) { /* opcodeUpdateSig */
+            err = r.read_bits(1, &b);
+            if (err) return err;
+            if (b == 0) sig = 0;
+            else {
+                uint64_t s;
+                err = r.read_bits(NUM_SIG_BITS, &s);
+                if (err) return err;
+                sig = (uint8_t)s + 1;
+            }
+        }
+        err = r.read_bits(1, &b);
+        if (err) return err;
+        if (b == 1) { /* opcodeUpdateMult */
+            uint64_t m;
+            err = r.read_bits(NUM_MULT_BITS, &m);
+            if (err) return err;
+            mult = (uint8_t)m;
+            if (mult > MAX_MULT) return M3GPU_SERIES_INVALID_MULT;
+        }
+        return 0;
+    }
+    __device__ int read_int_val_diff() {
+        if (sig == 64) { /* readIntValDiffSlow */
+            uint64_t sb;
+            int err = r.read_bits(1, &sb);
+            if (err) return err;
+            double sgn = (sb == 1) ? 1.0 : -1.0;
+            uint64_t bits;
+            err = r.read_bits(sig, &bits);
+            if (err) return err;
+            int_val += sgn * (double)bits;
+            return 0;
+        }
+        uint64_t bits;
+        int err = r.read_bits((uint32_t)sig + 1, &bits);
+        if (err) return err;
+        double sgn = -1.0;
+        if ((bits >> sig) == 1) { sgn = 1.0; bits ^= (1ULL << sig); }
+        int_val += sgn * (double)bits;
+        return 0;
+    }
+
+    /* iterator.go:108-176 */
+    __device__ int read_first_value() {
+        if (!int_optimized) return read_full_float();
+        uint64_t b;
+        int err = r.read_bits(1, &b);
+        if (err) return err;
+        if (b == 1) { /* float mode */
+            err = read_full_float();
+            if (err) return err;
+            is_float = true;
+            return 0;
+        }
+        err = read_int_sig_mult();
+        if (err) return err;
+        return read_int_val_diff();
+    }
+    __device__ int read_next_value() {
+        if (!int_optimized) return read_next_float();
+        uint64_t b;
+        int err = r.read_bits(1, &b);
+        if (err) return err;
+        if (b == 0) { /* opcodeUpdate */
+            err = r.read_bits(1, &b);
+            if (err) return err;
+            if (b == 1) return 0; /* repeat */
+            err = r.read_bits(1, &b);
+            if (err) return err;
+            if (b == 1) { /* -> float mode */
+                err = read_full_float();
+                if (err) return err;
+                is_float = true;
+                return 0;
+            }
+            err = read_int_sig_mult();
+            if (err) return err;
+            err = read_int_val_diff();
+            if (err) return err;
+            is_float = false;
+            return 0;
+        }
+        if (is_float) return read_next_float();
+        return read_int_val_diff();
+    }
+
+    /* One point. Returns 1 = value in (*t,*v), 0 = done, -err on error. */
+    __device__ int next(int64_t* t, double* v) {
+        if (done) return 0;
+        bool first;
+        int err = read_timestamp(&first);
+        if (err) return -err;
+        if (done) return 0;
+        err = first ? read_first_value() : read_next_value();
+        if (err) return -err;
+        *t = prev_time;
+        if (!int_optimized || is_float) *v = bits2f(prev_float_bits);
+        else *v = (mult == 0) ? int_val : int_val / exp10_table(mult);
+        return 1;
+    }
+    __device__ __forceinline__ double exp10_table(uint8_t m) {
+        const double mult_tab[7] = {1.0, 10.0, 100.0, 1000.0, 10000.0, 100000.0, 1000000.0};
+        return mult_tab[m];
+    }
+};
+
+/* ========================= decode kernel ========================= */
+
+__global__ void __launch_bounds__(BLOCK_THREADS)
+k_decode_batch(const uint8_t* __restrict__ blobs,
+               const uint64_t* __restrict__ offsets,
+               const uint32_t* __restrict__ lens,
+               uint32_t nseries, int int_optimized, uint8_t default_unit,
+               int64_t* __restrict__ out_ts, double* __restrict__ out_vals,
+               uint32_t* __restrict__ out_counts, int32_t* __restrict__ out_errs,
+               uint32_t stride) {
+    const uint32_t wave = threadIdx.x / WAVE;
+    const uint32_t lane = threadIdx.x % WAVE;
+    const uint32_t series = blockIdx.x * WAVES_PER_BLOCK + wave;
+    if (series >= nseries) return;
+
+    Decoder d;
+    d.init(blobs, offsets[series], lens[series], int_optimized != 0, default_unit);
+
+    int64_t* row_ts = out_ts + (uint64_t)series * stride;
+    double* row_vals = out_vals + (uint64_t)series * stride;
+
+    uint32_t cnt = 0;
+    int err = 0;
+    int64_t my_ts = 0;
+    double my_val = 0;
+
+    for (;;) {
+        int64_t t;
+        double v;
+        int rstat = d.next(&t, &v);
+        if (rstat <= 0) { err = -rstat; break; }
+        if (cnt >= stride) { err = M3GPU_SERIES_CAPACITY; break; }
+        /* lane (cnt & 63) captures this point; coalesced flush every 64 */
+        if ((cnt & 63) == lane) { my_ts = t; my_val = v; }
+        cnt++;
+        if ((cnt & 63) == 0) {
+            uint32_t base = cnt - 64;
+            row_ts[base + lane] = my_ts;
+            row_vals[base + lane] = my_val;
+        }
+    }
+    uint32_t rem = cnt & 63;
+    if (lane < rem) {
+        uint32_t base = cnt - rem;
+        row_ts[base + lane] = my_ts;
+        row_vals[base + lane] = my_val;
+    }
+    if (lane == 0) {
+        out_counts[series] = cnt;
+        out_errs[series] = err;
+    }
+}
+
+/* ===================== device encoder ===================== */
+/* Word-based big-endian bit emitter producing the byte-identical stream of
+ * ostream.go:133-221; lane (w & 63) stages word w, coalesced 512B flushes. */
+
+struct BitWriter {
+    uint64_t acc;       /* left-aligned bit accumulator */
+    uint32_t used;      /* bits used in acc */
+    uint32_t nwords;    /* full words emitted */
+    uint64_t staged;    /* this lane's staged word */
+    uint64_t* out;      /* aligned output row */
+    uint32_t cap_words;
+    uint32_t lane;
+    int err;
+
+    __device__ void init(uint8_t* row, uint32_t cap_bytes, uint32_t l) {
+        acc = 0; used = 0; nwords = 0; staged = 0;
+        out = (uint64_t*)row;
+        cap_words = cap_bytes / 8;
+        lane = l;
+        err = 0;
+    }
+    __device__ __forceinline__ void emit_word(uint64_t w) {
+        if ((nwords & 63) == lane) staged = __builtin_bswap64(w);
+        nwords++;
+        if ((nwords & 63) == 0) {
+            uint32_t base = nwords - 64;
+            if (nwords > cap_words) { err = M3GPU_SERIES_CAPACITY; return; }
+            out[base + lane] = staged;
+        }
+    }
+    __device__ __forceinline__ void write_bits(uint64_t v, uint32_t n) {
+        if (n == 0 || err) return;
+        v = (n >= 64) ? v : (v & ((1ULL << n) - 1)); /* low n bits */
+        uint32_t space = 64 - used;
+        if (n <= space) {
+            acc |= (space - n >= 64) ? 0 : (v << (space - n));
+            used += n;
+            if (used == 64) { emit_word(acc); acc = 0; used = 0; }
+        } else {
+            uint32_t hi = space;          /* bits that fit */
+            uint32_t lo = n - space;      /* remainder */
+            if (hi) acc |= v >> lo;
+            emit_word(acc);
+            acc = (lo >= 64) ? 0 : (v << (64 - lo));
+            used = lo;
+        }
+    }
+    __device__ __forceinline__ void write_bit(uint32_t b) { write_bits(b, 1); }
+    /* Flush staged words + partial tail. Returns total byte length. */
+    __device__ uint32_t finish() {
+        uint32_t rem_words = nwords & 63;
+        uint32_t base = nwords - rem_words;
+        if (nwords > cap_words) { err = M3GPU_SERIES_CAPACITY; return 0; }
+        if (lane < rem_words) out[base + lane] = staged;
+        uint32_t nbytes = nwords * 8;
+        if (used > 0) {
+            if (nwords + 1 > cap_words) { err = M3GPU_SERIES_CAPACITY; return 0; }
+            if (lane == 0) out[nwords] = __builtin_bswap64(acc); /* zero-padded */
+            nbytes += (used + 7) / 8;
+        }
+        return nbytes;
+    }
+};
+
+/* m3tsz.go:78-119 convertToIntFloat — bit-sensitive: -ffp-contract=off */
+__device__ int convert_to_int_float(double v, uint8_t cur_max_mult,
+                                    double* out_val, uint8_t* out_mult, bool* out_is_float) {
+    const double mult_tab[7] = {1.0, 10.0, 100.0, 1000.0, 10000.0, 100000.0, 1000000.0};
+    const double MAXINT = 9223372036854775808.0;
+    if (cur_max_mult == 0 && v < MAXINT) {
+        double i, r;
+        r = go_modf(v, &i);
+        if (r == 0) { *out_val = i; *out_mult = 0; *out_is_float = false; return 0; }
+    }
+    if (cur_max_mult > MAX_MULT) return M3GPU_SERIES_INVALID_MULT;
+    double sign = 1.0;
+    if (v < 0) sign = -1.0;
+    for (uint8_t m = cur_max_mult; m <= MAX_MULT; m++) {
+        double val = v * mult_tab[m] * sign;
+        if (val >= 1e13) break;
+        double i, r;
+        r = go_modf(val, &i);
+        if (r == 0) { *out_val = sign * i; *out_mult = m; *out_is_float = false; return 0; }
+        else if (r < 0.1) {
+            if (nextafter(val, 0.0) <= i) { *out_val = sign * i; *out_mult = m; *out_is_float = false; return 0; }
+        } else if (r > 0.9) {
+            double nxt = i + 1;
+            if (nextafter(val, nxt) >= nxt) { *out_val = sign * nxt; *out_mult = m; *out_is_float = false; return 0; }
+        }
+    }
+    *out_val = v; *out_mult = 0; *out_is_float = true;
+    return 0;
+}
+
+/* Encoder state: encoder.go:42-61 + TimestampEncoder + sig tracker. The bulk
+ * path has a fixed unit and no annotations (m3gpu.h contract). */
+struct Encoder {
+    BitWriter w;
+    int64_t prev_time, prev_time_delta;
+    uint64_t prev_xor, prev_float_bits;
+    double int_val;
+    uint8_t time_unit, max_mult;
+    uint8_t num_sig_state, cur_highest_lower_sig, num_lower_sig;
+    bool has_written_first, is_float, int_optimized;
+    uint32_t num_encoded;
+
+    __device__ void init(uint8_t* row, uint32_t cap_bytes, uint32_t lane,
+                         int64_t start_ns, bool intopt, uint8_t default_unit) {
+        w.init(row, cap_bytes, lane);
+        prev_time = start_ns;
+        prev_time_delta = 0;
+        prev_xor = 0; prev_float_bits = 0;
+        int_val = 0;
+        /* initialTimeUnit (timestamp_encoder.go:248-259) */
+        time_unit = (unit_valid(default_unit) &&
+                     start_ns % UNIT_NS_D[default_unit] == 0) ? default_unit : 0;
+        max_mult = 0;
+        num_sig_state = 0; cur_highest_lower_sig = 0; num_lower_sig = 0;
+        has_written_first = false; is_float = false;
+        int_optimized = intopt;
+        num_encoded = 0;
+    }
+
+    __device__ void write_marker(uint32_t marker) {
+        w.write_bits(MARKER_OPCODE, 9);
+        w.write_bits(marker, 2);
+    }
+
+    /* timestamp_encoder.go:205-246 */
+    __device__ int write_dod_unchanged(int64_t prev_delta, int64_t cur_delta, uint8_t unit) {
+        if (!unit_valid(unit)) return M3GPU_SERIES_NO_SCHEME;
+        int64_t u = UNIT_NS_D[unit];
+        int64_t dod = (cur_delta - prev_delta) / u;
+        if (unit == 1 || unit == 2) {
+            if ((int64_t)(int32_t)dod != dod) return M3GPU_SERIES_DOD_OVERFLOW;
+        }
+        int dbits = scheme_default_bits(unit);
+        if (!dbits) return M3GPU_SERIES_NO_SCHEME;
+        if (dod == 0) { w.write_bits(0, 1); return 0; }
+        const uint32_t opcodes[3] = {0x2, 0x6, 0xe};
+        const uint32_t obits[3] = {2, 3, 4};
+        const uint32_t vbits[3] = {7, 9, 12};
+        for (int i = 0; i < 3; i++) {
+            int64_t bmin = -((int64_t)1 << (vbits[i] - 1));
+            int64_t bmax = ((int64_t)1 << (vbits[i] - 1)) - 1;
+            if (dod >= bmin && dod <= bmax) {
+                w.write_bits(opcodes[i], obits[i]);
+                w.write_bits((uint64_t)dod, vbits[i]);
+                return 0;
+            }
+        }
+        w.write_bits(0xf, 4);
+        w.write_bits((uint64_t)dod, dbits);
+        return 0;
+    }
+
+    /* timestamp_encoder.go:72-129 (fixed unit, no annotations) */
+    __device__ int write_time(int64_t cur_time, uint8_t unit) {
+        if (!has_written_first) {
+            w.write_bits((uint64_t)prev_time, 64);
+            has_written_first = true;
+        }
+        bool tu_changed = false;
+        if (unit_valid(unit) && unit != time_unit) { /* maybeWriteTimeUnitChange */
+            write_marker(MARKER_TIMEUNIT);
+            w.write_bits(unit, 8);
+            time_unit = unit;
+            tu_changed = true;
+        }
+        int64_t time_delta = cur_time - prev_time;
+        prev_time = cur_time;
+        if (tu_changed) {
+            w.write_bits((uint64_t)(time_delta - prev_time_delta), 64);
+            prev_time_delta = 0;
+            return 0;
+        }
+        int err = write_dod_unchanged(prev_time_delta, time_delta, unit);
+        prev_time_delta = time_delta;
+        return err;
+    }
+
+    /* float_encoder_iterator.go:69-103 */
+    __device__ void write_full_float(uint64_t val) {
+        prev_float_bits = val;
+        prev_xor = val;
+        w.write_bits(val, 64);
+    }
+    __device__ void write_xor(uint64_t cur_xor) {
+        if (cur_xor == 0) { w.write_bits(0, 1); return; }
+        uint32_t pl = prev_xor ? __builtin_clzll(prev_xor) : 64;
+        uint32_t pt = prev_xor ? __builtin_ctzll(prev_xor) : 0;
+        uint32_t cl = __builtin_clzll(cur_xor);
+        uint32_t ct = __builtin_ctzll(cur_xor);
+        if (cl >= pl && ct >= pt) {
+            w.write_bits(0x2, 2);
+            w.write_bits(cur_xor >> pt, 64 - pl - pt);
+            return;
+        }
+        w.write_bits(0x3, 2);
+        w.write_bits(cl, 6);
+        uint32_t nmean = 64 - cl - ct;
+        w.write_bits(nmean - 1, 6);
+        w.write_bits(cur_xor >> ct, nmean);
+    }
+    __device__ void write_next_float(uint64_t val) {
+        uint64_t x = prev_float_bits ^ val;
+        write_xor(x);
+        prev_xor = x;
+        prev_float_bits = val;
+    }
+
+    /* int_sig_bits_tracker.go:35-91 */
+    __device__ void tracker_write_int_val_diff(uint64_t val_bits, bool neg) {
+        w.write_bit(neg ? 1 : 0);
+        w.write_bits(val_bits, num_sig_state);
+    }
+    __device__ void tracker_write_int_sig(uint8_t s) {
+        if (num_sig_state != s) {
+            w.write_bit(1);
+            if (s == 0) w.write_bit(0);
+            else { w.write_bit(1); w.write_bits((uint64_t)(s - 1), NUM_SIG_BITS); }
+        } else {
+            w.write_bit(0);
+        }
+        num_sig_state = s;
+    }
+    __device__ uint8_t tracker_track_new_sig(uint8_t nsig) {
+        uint8_t new_sig = num_sig_state;
+        if (nsig > num_sig_state) {
+            new_sig = nsig;
+        } else if (num_sig_state - nsig >= SIG_DIFF_THRESHOLD) {
+            if (num_lower_sig == 0) cur_highest_lower_sig = nsig;
+            else if (nsig > cur_highest_lower_sig) cur_highest_lower_sig = nsig;
+            num_lower_sig++;
+            if (num_lower_sig >= SIG_REPEAT_THRESHOLD) {
+                new_sig = cur_highest_lower_sig;
+                num_lower_sig = 0;
+            }
+        } else {
+            num_lower_sig = 0;
+        }
+        return new_sig;
+    }
+
+    /* encoder.go:233-250 */
+    __device__ void write_int_sig_mult(uint8_t s, uint8_t m, bool float_changed) {
+        tracker_write_int_sig(s);
+        if (m > max_mult) {
+            w.write_bit(1);
+            w.write_bits(m, NUM_MULT_BITS);
+            max_mult = m;
+        } else if (num_sig_state == s && max_mult == m && float_changed) {
+            w.write_bit(1);
+            w.write_bits(max_mult, NUM_MULT_BITS);
+        } else {
+            w.write_bit(0);
+        }
+    }
+
+    /* encoder.go:112-146 */
+    __device__ int write_first_value(double v) {
+        if (!int_optimized) { write_full_float(f2bits(v)); return 0; }
+        double val; uint8_t m; bool isf;
+        int err = convert_to_int_float(v, 0, &val, &m, &isf);
+        if (err) return err;
+        if (isf) {
+            w.write_bit(1);
+            write_full_float(f2bits(v));
+            is_float = true;
+            max_mult = m;
+            return 0;
+        }
+        w.write_bit(0);
+        int_val = val;
+        bool neg_diff = true;
+        if (val < 0) { neg_diff = false; val = -1 * val; }
+        uint64_t val_bits = (uint64_t)go_f2i(val);
+        uint8_t nsig = num_sig(val_bits);
+        write_int_sig_mult(nsig, m, false);
+        tracker_write_int_val_diff(val_bits, neg_diff);
+        return 0;
+    }
+
+    /* encoder.go:174-231 */
+    __device__ void write_float_val(uint64_t val, uint8_t m) {
+        if (!is_float) {
+            w.write_bit(0); w.write_bit(0); w.write_bit(1);
+            write_full_float(val);
+            is_float = true;
+            max_mult = m;
+            return;
+        }
+        if (val == prev_float_bits) { w.write_bit(0); w.write_bit(1); return; }
+        w.write_bit(1);
+        write_next_float(val);
+    }
+    __device__ void write_int_val(double val, uint8_t m, bool isf, double val_diff) {
+        if (val_diff == 0 && isf == is_float && m == max_mult) {
+            w.write_bit(0); w.write_bit(1);
+            return;
+        }
+        bool neg = false;
+        if (val_diff < 0) { neg = true; val_diff = -1 * val_diff; }
+        uint64_t diff_bits = (uint64_t)go_f2i(val_diff);
+        uint8_t nsig = num_sig(diff_bits);
+        uint8_t new_sig = tracker_track_new_sig(nsig);
+        bool float_changed = (isf != is_float);
+        if (m > max_mult || num_sig_state != new_sig || float_changed) {
+            w.write_bit(0); w.write_bit(0); w.write_bit(0);
+            write_int_sig_mult(new_sig, m, float_changed);
+            tracker_write_int_val_diff(diff_bits, neg);
+            is_float = false;
+        } else {
+            w.write_bit(1);
+            tracker_write_int_val_diff(diff_bits, neg);
+        }
+        int_val = val;
+    }
+
+    /* encoder.go:148-172 */
+    __device__ int write_next_value(double v) {
+        if (!int_optimized) { write_next_float(f2bits(v)); return 0; }
+        double val; uint8_t m; bool isf;
+        int err = convert_to_int_float(v, max_mult, &val, &m, &isf);
+        if (err) return err;
+        double val_diff = 0;
+        if (!isf) val_diff = int_val - val;
+        if (isf || val_diff >= 9223372036854775808.0 || val_diff <= -9223372036854775808.0) {
+            write_float_val(f2bits(val), m);
+            return 0;
+        }
+        write_int_val(val, m, isf, val_diff);
+        return 0;
+    }
+
+    __device__ int encode(int64_t t, double v, uint8_t unit) {
+        int err = write_time(t, unit);
+        if (err) return err;
+        err = num_encoded == 0 ? write_first_value(v) : write_next_value(v);
+        if (err == 0) num_encoded++;
+        return err ? err : w.err;
+    }
+
+    /* Finalize: appending the EOS marker to the live bitstream produces
+     * exactly head[:len-1] + Tail(lastByte, pos) (scheme.go:198-212). */
+    __device__ uint32_t finalize() {
+        if (num_encoded == 0 && w.nwords == 0 && w.used == 0) return 0;
+        write_marker(MARKER_EOS);
+        return w.finish();
+    }
+};
+
+__global__ void __launch_bounds__(BLOCK_THREADS)
+k_encode_batch(const int64_t* __restrict__ ts, const double* __restrict__ vals,
+               const uint32_t* __restrict__ counts, uint32_t nseries,
+               uint32_t stride, int int_optimized, uint8_t unit,
+               uint8_t* __restrict__ out_bytes, uint32_t out_stride,
+               uint32_t* __restrict__ out_lens, int32_t* __restrict__ out_errs) {
+    const uint32_t wave = threadIdx.x / WAVE;
+    const uint32_t lane = threadIdx.x % WAVE;
+    const uint32_t series = blockIdx.x * WAVES_PER_BLOCK + wave;
+    if (series >= nseries) return;
+
+    const int64_t* row_ts = ts + (uint64_t)series * stride;
+    const double* row_vals = vals + (uint64_t)series * stride;
+    uint32_t n = counts[series];
+
+    Encoder e;
+    e.init(out_bytes + (uint64_t)series * out_stride, out_stride, lane,
+           n ? row_ts[0] : 0, int_optimized != 0, unit);
+
+    int err = 0;
+    for (uint32_t j = 0; j < n; j++) {
+        err = e.encode(row_ts[j], row_vals[j], unit);
+        if (err) break;
+    }
+    uint32_t len = 0;
+    if (!err) {
+        len = e.finalize();
+        err = e.w.err;
+    }
+    if (lane == 0) {
+        out_lens[series] = err ? 0 : len;
+        out_errs[series] = err;
+    }
+}
+
+/* ===================== fused decode -> rollup kernel ===================== */
+/* Reference semantics:
+ *  bucket assignment  generic_elem.go:219-221 (truncate to window)
+ *  output timestamp   list.go:541-543 (window END)
+ *  Counter            counter.go:52-131  (int64 sum/min/max/count/sumSq)
+ *  Gauge              gauge.go:73-165    (NaN rules :87-98; last by ts order)
+ *  Timer              timer.go:56-153 over the CKMS stream; for <= 64 values
+ *                     per bucket the production-default CKMS (eps=1e-3,
+ *                     insertAndCompressEvery=1024) equals the no-compression
+ *                     calcQuantiles walk (stream.go:231-277), implemented
+ *                     here exactly (incl. the one-emission-per-sample shift
+ *                     for colliding ranks); >64 values flags
+ *                     M3GPU_SERIES_BUCKET_OVERFLOW.
+ */
+
+#define QCAP 64
+#define MAX_AGGS 16
+
+struct RollupPlan {
+    int32_t agg_types[MAX_AGGS];
+    int8_t qidx[MAX_AGGS];   /* index into qs, or -1 */
+    double qs[MAX_AGGS];     /* sorted unique quantiles */
+    int32_t nq;
+    int32_t naggs;
+};
+
+struct BucketState {
+    int64_t isum, imin, imax, isumsq;
+    double fsum, fsumsq, fmin, fmax, last;
+    int64_t last_at;
+    int64_t count;
+
+    __device__ void reset() {
+        isum = 0; isumsq = 0;
+        imin = INT64_MAX; imax = INT64_MIN;   /* counter.go:44-47 */
+        fsum = 0; fsumsq = 0;
+        fmin = __longlong_as_double(0x7ff8000000000000LL); /* NaN, gauge.go:56-59 */
+        fmax = __longlong_as_double(0x7ff8000000000000LL);
+        last = 0; last_at = 0;
+        count = 0;
+    }
+};
+
+__device__ double m3_stdev(int64_t count, double sum_sq, double sum) { /* common.go:29-36 */
+    int64_t div = count * (count - 1);
+    if (div == 0) return 0.0;
+    return sqrt(((double)count * sum_sq - sum * sum) / (double)div);
+}
+
+__global__ void __launch_bounds__(BLOCK_THREADS)
+k_rollup_batch(const uint8_t* __restrict__ blobs,
+               const uint64_t* __restrict__ offsets,
+               const uint32_t* __restrict__ lens,
+               uint32_t nseries, int int_optimized, uint8_t default_unit,
+               int metric_type, int64_t window_ns, uint32_t nbuckets,
+               RollupPlan plan,
+               double* __restrict__ out, int64_t* __restrict__ out_window_ts,
+               int32_t* __restrict__ out_errs) {
+    const uint32_t wave = threadIdx.x / WAVE;
+    const uint32_t lane = threadIdx.x % WAVE;
+    const uint32_t series = blockIdx.x * WAVES_PER_BLOCK + wave;
+    /* per-wave bucket value staging for quantiles (timer) */
+    __shared__ double qvals_all[WAVES_PER_BLOCK][QCAP];
+    __shared__ double sorted_all[WAVES_PER_BLOCK][QCAP];
+    double* qvals = qvals_all[wave];
+    double* sorted = sorted_all[wave];
+    if (series >= nseries) return;
+
+    Decoder d;
+    d.init(blobs, offsets[series], lens[series], int_optimized != 0, default_unit);
+
+    double* out_row = out + (uint64_t)series * nbuckets * plan.naggs;
+    int64_t* wts_row = out_window_ts ? out_window_ts + (uint64_t)series * nbuckets : nullptr;
+
+    BucketState bs;
+    bs.reset();
+    int64_t base = 0;
+    int64_t cur_bucket = -1;
+    uint32_t nq_in_bucket = 0;
+    int err = 0;
+
+    auto emit_bucket = [&](int64_t b) {
+        if (b < 0 || b >= (int64_t)nbuckets) return;
+        if (wts_row && lane == 0) wts_row[b] = base + (b + 1) * window_ns;
+        uint32_t n = nq_in_bucket;
+        /* quantile targets: the no-compression calcQuantiles walk closed
+         * form: k_i = max(rank_i, k_{i-1}+1), value = sorted[min(k_i,n)-1];
+         * n<=3: sorted[min(int(q*n), n-1)] (quantilesFromBuf) */
+        if (metric_type == M3GPU_METRIC_TIMER && plan.nq > 0 && n > 0) {
+            /* rank-select: lane l < n computes the rank of qvals[l].
+             * LDS ops from one wave retire in order; the fences only stop
+             * compiler reordering around the cross-lane LDS use. */
+            __builtin_amdgcn_wave_barrier();
+            __threadfence_block();
+            if (lane < n) {
+                double v = qvals[lane];
+                uint32_t rank = 0;
+                for (uint32_t j = 0; j < n; j++) {
+                    double o = qvals[j];
+                    rank += (o < v) || (o == v && j < lane);
+                }
+                sorted[rank] = v;
+            }
+            __builtin_amdgcn_wave_barrier();
+            __threadfence_block();
+        }
+        if (lane < (uint32_t)plan.naggs) {
+            int32_t t = plan.agg_types[lane];
+            int8_t qi = plan.qidx[lane];
+            double r = 0;
+            if (metric_type == M3GPU_METRIC_COUNTER) { /* counter.go:112-131 */
+                switch (t) {
+                case M3GPU_AGG_MIN: r = (double)bs.imin; break;
+                case M3GPU_AGG_MAX: r = (double)bs.imax; break;
+                case M3GPU_AGG_MEAN: r = bs.count ? (double)bs.isum / (double)bs.count : 0; break;
+                case M3GPU_AGG_COUNT: r = (double)bs.count; break;
+                case M3GPU_AGG_SUM: r = (double)bs.isum; break;
+                case M3GPU_AGG_SUMSQ: r = (double)bs.isumsq; break;
+                case M3GPU_AGG_STDEV: r = m3_stdev(bs.count, (double)bs.isumsq, (double)bs.isum); break;
+                default: r = 0; break;
+                }
+            } else if (metric_type == M3GPU_METRIC_GAUGE) { /* gauge.go:144-165 */
+                switch (t) {
+                case M3GPU_AGG_LAST: r = bs.last; break;
+                case M3GPU_AGG_MIN: r = bs.fmin; break;
+                case M3GPU_AGG_MAX: r = bs.fmax; break;
+                case M3GPU_AGG_MEAN: r = bs.count ? bs.fsum / (double)bs.count : 0.0; break;
+                case M3GPU_AGG_COUNT: r = (double)bs.count; break;
+                case M3GPU_AGG_SUM: r = bs.fsum; break;
+                case M3GPU_AGG_SUMSQ: r = bs.fsumsq; break;
+                case M3GPU_AGG_STDEV: r = m3_stdev(bs.count, bs.fsumsq, bs.fsum); break;
+                default: r = 0; break;
+                }
+            } else { /* timer.go:131-153 */
+                if (qi >= 0) {
+                    if (n == 0) r = 0.0; /* empty stream Quantile -> 0 */
+                    else if (n <= 3) { /* quantilesFromBuf :210-229 */
+                        int idx = (int)(plan.qs[qi] * (double)n);
+                        if (idx >= (int)n) idx = n - 1;
+                        r = sorted[idx];
+                    } else {
+                        /* calcQuantiles walk closed form over the sorted
+                         * unique quantile list (one emission per sample) */
+                        int k = 0;
+                        for (int i = 0; i <= qi; i++) {
+                            int rank = (int)ceil(plan.qs[i] * (double)n);
+                            k = (i == 0) ? rank : ((rank > k + 1) ? rank : k + 1);
+                        }
+                        if (k > (int)n) k = n;
+                        r = sorted[k - 1];
+                    }
+                } else {
+                    switch (t) {
+                    case M3GPU_AGG_MIN: r = n ? sorted[0] : 0.0; break;   /* Quantile(0) */
+                    case M3GPU_AGG_MAX: r = n ? sorted[n - 1] : 0.0; break;
+                    case M3GPU_AGG_MEAN: r = bs.count ? bs.fsum / (double)bs.count : 0.0; break;
+                    case M3GPU_AGG_COUNT: r = (double)bs.count; break;
+                    case M3GPU_AGG_SUM: r = bs.fsum; break;
+                    case M3GPU_AGG_SUMSQ: r = bs.fsumsq; break;
+                    case M3GPU_AGG_STDEV: r = m3_stdev(bs.count, bs.fsumsq, bs.fsum); break;
+                    default: r = 0; break;
+                    }
+                }
+            }
+            out_row[(uint64_t)b * plan.naggs + lane] = r;
+        }
+    };
+
+    for (;;) {
+        int64_t t;
+        double v;
+        int rstat = d.next(&t, &v);
+        if (rstat < 0) { err = -rstat; break; }
+        bool have = rstat == 1;
+        int64_t b = -1;
+        if (have) {
+            if (cur_bucket < 0) base = (t / window_ns) * window_ns; /* Truncate */
+            if (t < base) { err = M3GPU_SERIES_UNSORTED; break; }
+            b = (t - base) / window_ns;
+            if (b >= (int64_t)nbuckets) { err = M3GPU_SERIES_CAPACITY; break; }
+            if (b < cur_bucket) { err = M3GPU_SERIES_UNSORTED; break; }
+        }
+        if (!have || b != cur_bucket) {
+            if (cur_bucket >= 0) emit_bucket(cur_bucket);
+            /* emit empty buckets in any gap (and the tail after the stream) */
+            int64_t stop = have ? b : (int64_t)nbuckets;
+            for (int64_t eb = (cur_bucket < 0 ? 0 : cur_bucket + 1); eb < stop; eb++) {
+                bs.reset();
+                nq_in_bucket = 0;
+                emit_bucket(eb);
+            }
+            if (!have) break;
+            bs.reset();
+            nq_in_bucket = 0;
+            cur_bucket = b;
+        }
+        /* accumulate */
+        if (metric_type == M3GPU_METRIC_COUNTER) { /* counter.go:52-78 */
+            int64_t iv = go_f2i(v);
+            bs.isum += iv;
+            bs.count++;
+            if (bs.imax < iv) bs.imax = iv;
+            if (bs.imin > iv) bs.imin = iv;
+            bs.isumsq += iv * iv;
+        } else if (metric_type == M3GPU_METRIC_GAUGE) { /* gauge.go:73-103 */
+            if (bs.last_at == 0 || t > bs.last_at) { bs.last_at = t; bs.last = v; }
+            bs.count++;
+            if (!isnan(v)) {
+                bs.fsum += v;
+                if (isnan(bs.fmax) || bs.fmax < v) bs.fmax = v;
+                if (isnan(bs.fmin) || bs.fmin > v) bs.fmin = v;
+                bs.fsumsq += v * v;
+            }
+        } else { /* timer.go:56-75 */
+            bs.count++;
+            bs.fsum += v;
+            bs.fsumsq += v * v;
+            if (plan.nq > 0) {
+                if (nq_in_bucket >= QCAP) { err = M3GPU_SERIES_BUCKET_OVERFLOW; break; }
+                if (lane == 0) qvals[nq_in_bucket] = v;
+                nq_in_bucket++;
+            }
+        }
+    }
+    if (lane == 0) out_errs[series] = err;
+}
+
+/* ===================== stream compaction kernel ===================== */
+/* Packs strided encoder output rows into the tight 8B-aligned blob layout
+ * (m3gpu.h): one series per wavefront, u64 copies, coalesced within a row. */
+__global__ void __launch_bounds__(BLOCK_THREADS)
+k_compact(const uint8_t* __restrict__ src, uint32_t src_stride,
+          const uint32_t* __restrict__ lens,
+          const uint64_t* __restrict__ dst_offsets, uint32_t nseries,
+          uint8_t* __restrict__ dst) {
+    const uint32_t wave = threadIdx.x / WAVE;
+    const uint32_t lane = threadIdx.x % WAVE;
+    const uint32_t series = blockIdx.x * WAVES_PER_BLOCK + wave;
+    if (series >= nseries) return;
+    const uint64_t* s = (const uint64_t*)(src + (uint64_t)series * src_stride);
+    uint64_t* d = (uint64_t*)(dst + dst_offsets[series]);
+    uint32_t nwords = (lens[series] + 7) / 8;
+    for (uint32_t w = lane; w < nwords; w += WAVE) d[w] = s[w];
+}
+
+} // namespace m3
+
+/* ============================ C-ABI host layer ============================ */
+
+#include <mutex>
+
+static __thread char g_err[512];
+static int g_device = 0;
+static bool g_inited = false;
+
+static int set_hip_err(const char* what, hipError_t e) {
+    snprintf(g_err, sizeof(g_err), "%s: %s", what, hipGetErrorString(e));
+    return M3GPU_ERR_HIP;
+}
+#define HIP_TRY(expr) do { hipError_t _e = (expr); if (_e != hipSuccess) return set_hip_err(#expr, _e); } while (0)
+
+extern "C" {
+
+const char* m3gpu_last_error(void) { return g_err; }
+
+int m3gpu_init(int device) {
+    HIP_TRY(hipSetDevice(device));
+    g_device = device;
+    g_inited = true;
+    return M3GPU_OK;
+}
+
+void m3gpu_shutdown(void) { g_inited = false; }
+
+static inline uint32_t grid_for(uint32_t nseries) {
+    return (nseries + WAVES_PER_BLOCK - 1) / WAVES_PER_BLOCK;
+}
+
+int m3gpu_decode_batch_dev(
+    const uint8_t* d_blobs, const uint64_t* d_offsets, const uint32_t* d_lens,
+    uint32_t nseries, int int_optimized, uint8_t default_unit,
+    int64_t* d_out_ts, double* d_out_vals, uint32_t* d_out_counts,
+    int32_t* d_out_errs, uint32_t stride, void* hip_stream) {
+    if (!nseries) return M3GPU_OK;
+    hipStream_t s = (hipStream_t)hip_stream;
+    hipLaunchKernelGGL(m3::k_decode_batch, dim3(grid_for(nseries)),
+                       dim3(BLOCK_THREADS), 0, s,
+                       d_blobs, d_offsets, d_lens, nseries, int_optimized,
+                       default_unit, d_out_ts, d_out_vals, d_out_counts,
+                       d_out_errs, stride);
+    HIP_TRY(hipGetLastError());
+    return M3GPU_OK;
+}
+
+int m3gpu_encode_batch_dev(
+    const int64_t* d_ts, const double* d_vals, const uint32_t* d_counts,
+    uint32_t nseries, uint32_t stride, int int_optimized, uint8_t unit,
+    uint8_t* d_out_bytes, uint32_t out_stride, uint32_t* d_out_lens,
+    int32_t* d_out_errs, void* hip_stream) {
+    if (!nseries) return M3GPU_OK;
+    if (out_stride % 8) {
+        snprintf(g_err, sizeof(g_err), "out_stride must be a multiple of 8");
+        return M3GPU_ERR_BADARG;
+    }
+    hipStream_t s = (hipStream_t)hip_stream;
+    hipLaunchKernelGGL(m3::k_encode_batch, dim3(grid_for(nseries)),
+                       dim3(BLOCK_THREADS), 0, s,
+                       d_ts, d_vals, d_counts, nseries, stride, int_optimized,
+                       unit, d_out_bytes, out_stride, d_out_lens, d_out_errs);
+    HIP_TRY(hipGetLastError());
+    return M3GPU_OK;
+}
+
+int m3gpu_compact_dev(
+    const uint8_t* d_src, uint32_t src_stride, const uint32_t* d_lens,
+    const uint64_t* d_dst_offsets, uint32_t nseries, uint8_t* d_dst,
+    void* hip_stream) {
+    if (!nseries) return M3GPU_OK;
+    hipStream_t s = (hipStream_t)hip_stream;
+    hipLaunchKernelGGL(m3::k_compact, dim3(grid_for(nseries)),
+                       dim3(BLOCK_THREADS), 0, s,
+                       d_src, src_stride, d_lens, d_dst_offsets, nseries, d_dst);
+    HIP_TRY(hipGetLastError());
+    return M3GPU_OK;
+}
+
+int m3gpu_rollup_batch_dev(
+    const uint8_t* d_blobs, const uint64_t* d_offsets, const uint32_t* d_lens,
+    uint32_t nseries, int int_optimized, uint8_t default_unit,
+    int metric_type, int64_t window_ns, uint32_t nbuckets,
+    const int32_t* agg_types, int naggs,
+    double* d_out, int64_t* d_out_window_ts, int32_t* d_out_errs,
+    void* hip_stream) {
+    if (!nseries) return M3GPU_OK;
+    if (naggs < 1 || naggs > MAX_AGGS || window_ns <= 0) {
+        snprintf(g_err, sizeof(g_err), "bad naggs/window");
+        return M3GPU_ERR_BADARG;
+    }
+    m3::RollupPlan plan;
+    memset(&plan, 0, sizeof(plan));
+    plan.naggs = naggs;
+    /* sorted unique quantile list + per-agg mapping (timer stream is
+     * registered with the sorted unique quantile set) */
+    double qs[MAX_AGGS];
+    int nq = 0;
+    for (int i = 0; i < naggs; i++) {
+        plan.agg_types[i] = agg_types[i];
+        plan.qidx[i] = -1;
+        double q = -1;
+        switch (agg_types[i]) {
+        case M3GPU_AGG_MEDIAN: q = 0.5; break;
+        case M3GPU_AGG_P10: q = 0.1; break;
+        case M3GPU_AGG_P20: q = 0.2; break;
+        case M3GPU_AGG_P25: q = 0.25; break;
+        case M3GPU_AGG_P30: q = 0.3; break;
+        case M3GPU_AGG_P40: q = 0.4; break;
+        case M3GPU_AGG_P50: q = 0.5; break;
+        case M3GPU_AGG_P60: q = 0.6; break;
+        case M3GPU_AGG_P70: q = 0.7; break;
+        case M3GPU_AGG_P75: q = 0.75; break;
+        case M3GPU_AGG_P80: q = 0.8; break;
+        case M3GPU_AGG_P90: q = 0.9; break;
+        case M3GPU_AGG_P95: q = 0.95; break;
+        case M3GPU_AGG_P99: q = 0.99; break;
+        case M3GPU_AGG_P999: q = 0.999; break;
+        case M3GPU_AGG_P9999: q = 0.9999; break;
+        default: break;
+        }
+        if (q >= 0) { qs[nq++] = q; }
+    }
+    /* sort + dedupe */
+    for (int i = 1; i < nq; i++) {
+        double k = qs[i];
+        int j = i - 1;
+        while (j >= 0 && qs[j] > k) { qs[j + 1] = qs[j]; j--; }
+        qs[j + 1] = k;
+    }
+    int m = 0;
+    for (int i = 0; i < nq; i++)
+        if (m == 0 || qs[m - 1] != qs[i]) qs[m++] = qs[i];
+    nq = m;
+    plan.nq = nq;
+    for (int i = 0; i < nq; i++) plan.qs[i] = qs[i];
+    for (int i = 0; i < naggs; i++) {
+        double q = -1;
+        switch (agg_types[i]) {
+        case M3GPU_AGG_MEDIAN: case M3GPU_AGG_P50: q = 0.5; break;
+        case M3GPU_AGG_P10: q = 0.1; break;
+        case M3GPU_AGG_P20: q = 0.2; break;
+        case M3GPU_AGG_P25: q = 0.25; break;
+        case M3GPU_AGG_P30: q = 0.3; break;
+        case M3GPU_AGG_P40: q = 0.4; break;
+        case M3GPU_AGG_P60: q = 0.6; break;
+        case M3GPU_AGG_P70: q = 0.7; break;
+        case M3GPU_AGG_P75: q = 0.75; break;
+        case M3GPU_AGG_P80: q = 0.8; break;
+        case M3GPU_AGG_P90: q = 0.9; break;
+        case M3GPU_AGG_P95: q = 0.95; break;
+        case M3GPU_AGG_P99: q = 0.99; break;
+        case M3GPU_AGG_P999: q = 0.999; break;
+        case M3GPU_AGG_P9999: q = 0.9999; break;
+        default: break;
+        }
+        if (q >= 0)
+            for (int k = 0; k < nq; k++)
+                if (plan.qs[k] == q) { plan.qidx[i] = (int8_t)k; break; }
+    }
+
+    hipStream_t s = (hipStream_t)hip_stream;
+    hipLaunchKernelGGL(m3::k_rollup_batch, dim3(grid_for(nseries)),
+                       dim3(BLOCK_THREADS), 0, s,
+                       d_blobs, d_offsets, d_lens, nseries, int_optimized,
+                       default_unit, metric_type, window_ns, nbuckets, plan,
+                       d_out, d_out_window_ts, d_out_errs);
+    HIP_TRY(hipGetLastError());
+    return M3GPU_OK;
+}
+
+/* ---------- host-pointer convenience forms (the cgo surface) ---------- */
+
+int m3gpu_decode_batch(
+    const uint8_t* blobs, uint64_t blobs_len,
+    const uint64_t* offsets, const uint32_t* lens,
+    uint32_t nseries, int int_optimized, uint8_t default_unit,
+    int64_t* out_ts, double* out_vals, uint32_t* out_counts,
+    int32_t* out_errs, uint32_t stride) {
+    uint8_t* d_blobs = nullptr;
+    uint64_t* d_offsets = nullptr;
+    uint32_t* d_lens = nullptr;
+    int64_t* d_ts = nullptr;
+    double* d_vals = nullptr;
+    uint32_t* d_counts = nullptr;
+    int32_t* d_errs = nullptr;
+    uint64_t npts = (uint64_t)nseries * stride;
+    int rc = M3GPU_OK;
+    HIP_TRY(hipMalloc(&d_blobs, blobs_len));
+    HIP_TRY(hipMalloc(&d_offsets, (nseries + 1) * sizeof(uint64_t)));
+    HIP_TRY(hipMalloc(&d_lens, nseries * sizeof(uint32_t)));
+    HIP_TRY(hipMalloc(&d_ts, npts * sizeof(int64_t)));
+    HIP_TRY(hipMalloc(&d_vals, npts * sizeof(double)));
+    HIP_TRY(hipMalloc(&d_counts, nseries * sizeof(uint32_t)));
+    HIP_TRY(hipMalloc(&d_errs, nseries * sizeof(int32_t)));
+    HIP_TRY(hipMemcpy(d_blobs, blobs, blobs_len, hipMemcpyHostToDevice));
+    HIP_TRY(hipMemcpy(d_offsets, offsets, (nseries + 1) * sizeof(uint64_t), hipMemcpyHostToDevice));
+    HIP_TRY(hipMemcpy(d_lens, lens, nseries * sizeof(uint32_t), hipMemcpyHostToDevice));
+    rc = m3gpu_decode_batch_dev(d_blobs, d_offsets, d_lens, nseries,
+                                int_optimized, default_unit, d_ts, d_vals,
+                                d_counts, d_errs, stride, nullptr);
+    if (rc == M3GPU_OK) {
+        HIP_TRY(hipMemcpy(out_ts, d_ts, npts * sizeof(int64_t), hipMemcpyDeviceToHost));
+        HIP_TRY(hipMemcpy(out_vals, d_vals, npts * sizeof(double), hipMemcpyDeviceToHost));
+        HIP_TRY(hipMemcpy(out_counts, d_counts, nseries * sizeof(uint32_t), hipMemcpyDeviceToHost));
+        HIP_TRY(hipMemcpy(out_errs, d_errs, nseries * sizeof(int32_t), hipMemcpyDeviceToHost));
+    }
+    hipFree(d_blobs); hipFree(d_offsets); hipFree(d_lens);
+    hipFree(d_ts); hipFree(d_vals); hipFree(d_counts); hipFree(d_errs);
+    return rc;
+}
+
+int m3gpu_encode_batch(
+    const int64_t* ts, const double* vals, const uint32_t* counts,
+    uint32_t nseries, uint32_t stride, int int_optimized, uint8_t unit,
+    uint8_t* out_bytes, uint32_t out_stride, uint32_t* out_lens,
+    int32_t* out_errs) {
+    int64_t* d_ts = nullptr;
+    double* d_vals = nullptr;
+    uint32_t* d_counts = nullptr;
+    uint8_t* d_out = nullptr;
+    uint32_t* d_lens = nullptr;
+    int32_t* d_errs = nullptr;
+    uint64_t npts = (uint64_t)nseries * stride;
+    uint64_t outb = (uint64_t)nseries * out_stride;
+    int rc;
+    HIP_TRY(hipMalloc(&d_ts, npts * sizeof(int64_t)));
+    HIP_TRY(hipMalloc(&d_vals, npts * sizeof(double)));
+    HIP_TRY(hipMalloc(&d_counts, nseries * sizeof(uint32_t)));
+    HIP_TRY(hipMalloc(&d_out, outb));
+    HIP_TRY(hipMalloc(&d_lens, nseries * sizeof(uint32_t)));
+    HIP_TRY(hipMalloc(&d_errs, nseries * sizeof(int32_t)));
+    HIP_TRY(hipMemcpy(d_ts, ts, npts * sizeof(int64_t), hipMemcpyHostToDevice));
+    HIP_TRY(hipMemcpy(d_vals, vals, npts * sizeof(double), hipMemcpyHostToDevice));
+    HIP_TRY(hipMemcpy(d_counts, counts, nseries * sizeof(uint32_t), hipMemcpyHostToDevice));
+    HIP_TRY(hipMemset(d_out, 0, outb));
+    rc = m3gpu_encode_batch_dev(d_ts, d_vals, d_counts, nseries, stride,
+                                int_optimized, unit, d_out, out_stride,
+                                d_lens, d_errs, nullptr);
+    if (rc == M3GPU_OK) {
+        HIP_TRY(hipMemcpy(out_bytes, d_out, outb, hipMemcpyDeviceToHost));
+        HIP_TRY(hipMemcpy(out_lens, d_lens, nseries * sizeof(uint32_t), hipMemcpyDeviceToHost));
+        HIP_TRY(hipMemcpy(out_errs, d_errs, nseries * sizeof(int32_t), hipMemcpyDeviceToHost));
+    }
+    hipFree(d_ts); hipFree(d_vals); hipFree(d_counts);
+    hipFree(d_out); hipFree(d_lens); hipFree(d_errs);
+    return rc;
+}
+
+int m3gpu_rollup_batch(
+    const uint8_t* blobs, uint64_t blobs_len,
+    const uint64_t* offsets, const uint32_t* lens,
+    uint32_t nseries, int int_optimized, uint8_t default_unit,
+    int metric_type, int64_t window_ns, uint32_t nbuckets,
+    const int32_t* agg_types, int naggs,
+    double* out, int64_t* out_window_ts, int32_t* out_errs) {
+    uint8_t* d_blobs = nullptr;
+    uint64_t* d_offsets = nullptr;
+    uint32_t* d_lens = nullptr;
+    double* d_out = nullptr;
+    int64_t* d_wts = nullptr;
+    int32_t* d_errs = nullptr;
+    uint64_t nout = (uint64_t)nseries * nbuckets * naggs;
+    int rc;
+    HIP_TRY(hipMalloc(&d_blobs, blobs_len));
+    HIP_TRY(hipMalloc(&d_offsets, (nseries + 1) * sizeof(uint64_t)));
+    HIP_TRY(hipMalloc(&d_lens, nseries * sizeof(uint32_t)));
+    HIP_TRY(hipMalloc(&d_out, nout * sizeof(double)));
+    HIP_TRY(hipMalloc(&d_wts, (uint64_t)nseries * nbuckets * sizeof(int64_t)));
+    HIP_TRY(hipMalloc(&d_errs, nseries * sizeof(int32_t)));
+    HIP_TRY(hipMemcpy(d_blobs, blobs, blobs_len, hipMemcpyHostToDevice));
+    HIP_TRY(hipMemcpy(d_offsets, offsets, (nseries + 1) * sizeof(uint64_t), hipMemcpyHostToDevice));
+    HIP_TRY(hipMemcpy(d_lens, lens, nseries * sizeof(uint32_t), hipMemcpyHostToDevice));
+    rc = m3gpu_rollup_batch_dev(d_blobs, d_offsets, d_lens, nseries,
+                                int_optimized, default_unit, metric_type,
+                                window_ns, nbuckets, agg_types, naggs,
+                                d_out, d_wts, d_errs, nullptr);
+    if (rc == M3GPU_OK) {
+        HIP_TRY(hipMemcpy(out, d_out, nout * sizeof(double), hipMemcpyDeviceToHost));
+        if (out_window_ts)
+            HIP_TRY(hipMemcpy(out_window_ts, d_wts, (uint64_t)nseries * nbuckets * sizeof(int64_t), hipMemcpyDeviceToHost));
+        HIP_TRY(hipMemcpy(out_errs, d_errs, nseries * sizeof(int32_t), hipMemcpyDeviceToHost));
+    }
+    hipFree(d_blobs); hipFree(d_offsets); hipFree(d_lens);
+    hipFree(d_out); hipFree(d_wts); hipFree(d_errs);
+    return rc;
+}
+
+} /* extern "C" */

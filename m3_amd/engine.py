@@ -1,0 +1,248 @@
+"""ctypes binding of libm3gpu.so (include/m3gpu.h) + torch-tensor surface.
+
+Device entry points (`*_dev`) take torch CUDA tensors (ROCm) and enqueue on
+the current torch stream; host entry points take numpy arrays and block.
+"""
+import ctypes
+import os
+
+import numpy as np
+
+c_i64 = ctypes.c_int64
+c_u64 = ctypes.c_uint64
+c_i32 = ctypes.c_int32
+c_u32 = ctypes.c_uint32
+c_u8 = ctypes.c_uint8
+c_f64 = ctypes.c_double
+c_int = ctypes.c_int
+c_vp = ctypes.c_void_p
+P = ctypes.POINTER
+
+_DIR = os.path.dirname(os.path.abspath(__file__))
+_LIB_PATH = os.path.join(_DIR, "csrc", "libm3gpu.so")
+
+METRIC_COUNTER, METRIC_GAUGE, METRIC_TIMER = 0, 1, 2
+
+M3GPU_AGG = dict(last=1, min=2, max=3, mean=4, median=5, count=6, sum=7,
+                 sumsq=8, stdev=9, p10=10, p20=11, p30=12, p40=13, p50=14,
+                 p60=15, p70=16, p80=17, p90=18, p95=19, p99=20, p999=21,
+                 p9999=22, p25=23, p75=24)
+
+SERIES_ERRORS = {0: "ok", 1: "eof", 2: "dod_overflow", 3: "no_scheme",
+                 4: "invalid_mult", 5: "annotation", 6: "capacity",
+                 7: "unsorted", 8: "bucket_overflow"}
+
+
+class M3GpuError(RuntimeError):
+    pass
+
+
+_lib = None
+
+
+def engine_available():
+    return os.path.exists(_LIB_PATH)
+
+
+def lib():
+    """Load libm3gpu.so. Raises loudly when the HIP engine is missing —
+    the product path has no CPU fallback."""
+    global _lib
+    if _lib is None:
+        if not engine_available():
+            raise M3GpuError(
+                f"libm3gpu.so not found at {_LIB_PATH}; build it with "
+                f"`make -C m3_amd/csrc` (hipcc --offload-arch=gfx950). "
+                f"The m3_amd product path has no CPU fallback.")
+        L = ctypes.CDLL(_LIB_PATH)
+        L.m3gpu_init.restype = c_int
+        L.m3gpu_init.argtypes = [c_int]
+        L.m3gpu_last_error.restype = ctypes.c_char_p
+        L.m3gpu_decode_batch_dev.restype = c_int
+        L.m3gpu_decode_batch_dev.argtypes = [c_vp, c_vp, c_vp, c_u32, c_int, c_u8,
+                                             c_vp, c_vp, c_vp, c_vp, c_u32, c_vp]
+        L.m3gpu_decode_batch.restype = c_int
+        L.m3gpu_decode_batch.argtypes = [P(c_u8), c_u64, P(c_u64), P(c_u32), c_u32,
+                                         c_int, c_u8, P(c_i64), P(c_f64), P(c_u32),
+                                         P(c_i32), c_u32]
+        L.m3gpu_encode_batch_dev.restype = c_int
+        L.m3gpu_encode_batch_dev.argtypes = [c_vp, c_vp, c_vp, c_u32, c_u32, c_int,
+                                             c_u8, c_vp, c_u32, c_vp, c_vp, c_vp]
+        L.m3gpu_encode_batch.restype = c_int
+        L.m3gpu_encode_batch.argtypes = [P(c_i64), P(c_f64), P(c_u32), c_u32, c_u32,
+                                         c_int, c_u8, P(c_u8), c_u32, P(c_u32), P(c_i32)]
+        L.m3gpu_compact_dev.restype = c_int
+        L.m3gpu_compact_dev.argtypes = [c_vp, c_u32, c_vp, c_vp, c_u32, c_vp, c_vp]
+        L.m3gpu_rollup_batch_dev.restype = c_int
+        L.m3gpu_rollup_batch_dev.argtypes = [c_vp, c_vp, c_vp, c_u32, c_int, c_u8,
+                                             c_int, c_i64, c_u32, P(c_i32), c_int,
+                                             c_vp, c_vp, c_vp, c_vp]
+        L.m3gpu_rollup_batch.restype = c_int
+        L.m3gpu_rollup_batch.argtypes = [P(c_u8), c_u64, P(c_u64), P(c_u32), c_u32,
+                                         c_int, c_u8, c_int, c_i64, c_u32, P(c_i32),
+                                         c_int, P(c_f64), P(c_i64), P(c_i32)]
+        _lib = L
+    return _lib
+
+
+def _check(rc, what):
+    if rc != 0:
+        raise M3GpuError(f"{what} failed ({rc}): {lib().m3gpu_last_error().decode()}")
+
+
+def _raise_series_errors(errs, what):
+    errs = np.asarray(errs)
+    bad = np.nonzero(errs)[0]
+    if len(bad):
+        i = int(bad[0])
+        raise M3GpuError(
+            f"{what}: {len(bad)} series failed; first: series {i} -> "
+            f"{SERIES_ERRORS.get(int(errs[i]), errs[i])}")
+
+
+def pack_streams(streams):
+    """Pack a list of encoded streams into (blob, offsets, lens) with the
+    8-byte-aligned zero-padded layout the C ABI requires."""
+    n = len(streams)
+    lens = np.fromiter((len(s) for s in streams), dtype=np.uint32, count=n)
+    padded = (lens.astype(np.uint64) + 7) & ~np.uint64(7)
+    offsets = np.zeros(n + 1, dtype=np.uint64)
+    np.cumsum(padded, out=offsets[1:])
+    blob = np.zeros(int(offsets[-1]), dtype=np.uint8)
+    for i, s in enumerate(streams):
+        o = int(offsets[i])
+        blob[o:o + len(s)] = np.frombuffer(bytes(s), dtype=np.uint8)
+    return blob, offsets, lens
+
+
+def _np(a, dt):
+    return np.ascontiguousarray(a, dtype=dt)
+
+
+def _pp(a, ct):
+    return a.ctypes.data_as(P(ct))
+
+
+# ------------------------- host-pointer surface -------------------------
+
+def decode_batch(blob, offsets, lens, stride, int_optimized=True,
+                 default_unit=1, check_errors=True):
+    blob = _np(blob, np.uint8)
+    offsets = _np(offsets, np.uint64)
+    lens = _np(lens, np.uint32)
+    nseries = len(lens)
+    out_ts = np.empty((nseries, stride), dtype=np.int64)
+    out_vals = np.empty((nseries, stride), dtype=np.float64)
+    out_counts = np.empty(nseries, dtype=np.uint32)
+    out_errs = np.empty(nseries, dtype=np.int32)
+    rc = lib().m3gpu_decode_batch(
+        _pp(blob, c_u8), len(blob), _pp(offsets, c_u64), _pp(lens, c_u32),
+        nseries, 1 if int_optimized else 0, default_unit,
+        _pp(out_ts, c_i64), _pp(out_vals, c_f64), _pp(out_counts, c_u32),
+        _pp(out_errs, c_i32), stride)
+    _check(rc, "m3gpu_decode_batch")
+    if check_errors:
+        _raise_series_errors(out_errs, "decode")
+    return out_ts, out_vals, out_counts, out_errs
+
+
+def encode_batch(ts, vals, counts, int_optimized=True, unit=1,
+                 out_stride=None, check_errors=True):
+    ts = _np(ts, np.int64)
+    vals = _np(vals, np.float64)
+    counts = _np(counts, np.uint32)
+    nseries, stride = ts.shape
+    if out_stride is None:
+        out_stride = (24 * stride + 32 + 7) & ~7
+    out_bytes = np.zeros((nseries, out_stride), dtype=np.uint8)
+    out_lens = np.empty(nseries, dtype=np.uint32)
+    out_errs = np.empty(nseries, dtype=np.int32)
+    rc = lib().m3gpu_encode_batch(
+        _pp(ts, c_i64), _pp(vals, c_f64), _pp(counts, c_u32), nseries, stride,
+        1 if int_optimized else 0, unit, _pp(out_bytes, c_u8), out_stride,
+        _pp(out_lens, c_u32), _pp(out_errs, c_i32))
+    _check(rc, "m3gpu_encode_batch")
+    if check_errors:
+        _raise_series_errors(out_errs, "encode")
+    return out_bytes, out_lens, out_errs
+
+
+def rollup_batch(blob, offsets, lens, metric_type, window_ns, nbuckets,
+                 agg_types, int_optimized=True, default_unit=1,
+                 check_errors=True):
+    blob = _np(blob, np.uint8)
+    offsets = _np(offsets, np.uint64)
+    lens = _np(lens, np.uint32)
+    nseries = len(lens)
+    aggs = np.asarray([M3GPU_AGG[a] if isinstance(a, str) else a
+                       for a in agg_types], dtype=np.int32)
+    out = np.empty((nseries, nbuckets, len(aggs)), dtype=np.float64)
+    wts = np.empty((nseries, nbuckets), dtype=np.int64)
+    errs = np.empty(nseries, dtype=np.int32)
+    rc = lib().m3gpu_rollup_batch(
+        _pp(blob, c_u8), len(blob), _pp(offsets, c_u64), _pp(lens, c_u32),
+        nseries, 1 if int_optimized else 0, default_unit, metric_type,
+        window_ns, nbuckets, _pp(aggs, c_i32), len(aggs),
+        _pp(out, c_f64), _pp(wts, c_i64), _pp(errs, c_i32))
+    _check(rc, "m3gpu_rollup_batch")
+    if check_errors:
+        _raise_series_errors(errs, "rollup")
+    return out, wts, errs
+
+
+# ------------------------- torch device surface -------------------------
+
+def _dev_ptr(t):
+    return ctypes.c_void_p(t.data_ptr())
+
+
+def _torch_stream():
+    import torch
+    return ctypes.c_void_p(torch.cuda.current_stream().cuda_stream)
+
+
+def decode_batch_dev(d_blob, d_offsets, d_lens, out_ts, out_vals, out_counts,
+                     out_errs, int_optimized=True, default_unit=1):
+    """All args are torch CUDA tensors; enqueues on the current torch stream.
+    out_ts: int64 [nseries, stride]; out_vals: float64 [nseries, stride]."""
+    nseries = d_lens.numel()
+    stride = out_ts.shape[1]
+    rc = lib().m3gpu_decode_batch_dev(
+        _dev_ptr(d_blob), _dev_ptr(d_offsets), _dev_ptr(d_lens), nseries,
+        1 if int_optimized else 0, default_unit, _dev_ptr(out_ts),
+        _dev_ptr(out_vals), _dev_ptr(out_counts), _dev_ptr(out_errs), stride,
+        _torch_stream())
+    _check(rc, "m3gpu_decode_batch_dev")
+
+
+def encode_batch_dev(d_ts, d_vals, d_counts, out_bytes, out_lens, out_errs,
+                     int_optimized=True, unit=1):
+    nseries, stride = d_ts.shape
+    out_stride = out_bytes.shape[1]
+    rc = lib().m3gpu_encode_batch_dev(
+        _dev_ptr(d_ts), _dev_ptr(d_vals), _dev_ptr(d_counts), nseries, stride,
+        1 if int_optimized else 0, unit, _dev_ptr(out_bytes), out_stride,
+        _dev_ptr(out_lens), _dev_ptr(out_errs), _torch_stream())
+    _check(rc, "m3gpu_encode_batch_dev")
+
+
+def compact_dev(d_src, src_stride, d_lens, d_dst_offsets, d_dst):
+    nseries = d_lens.numel()
+    rc = lib().m3gpu_compact_dev(
+        _dev_ptr(d_src), src_stride, _dev_ptr(d_lens), _dev_ptr(d_dst_offsets),
+        nseries, _dev_ptr(d_dst), _torch_stream())
+    _check(rc, "m3gpu_compact_dev")
+
+
+def rollup_batch_dev(d_blob, d_offsets, d_lens, metric_type, window_ns,
+                     nbuckets, agg_types, out, out_window_ts, out_errs,
+                     int_optimized=True, default_unit=1):
+    nseries = d_lens.numel()
+    aggs = np.asarray([M3GPU_AGG[a] if isinstance(a, str) else a
+                       for a in agg_types], dtype=np.int32)
+    rc = lib().m3gpu_rollup_batch_dev(
+        _dev_ptr(d_blob), _dev_ptr(d_offsets), _dev_ptr(d_lens), nseries,
+        1 if int_optimized else 0, default_unit, metric_type, window_ns,
+        nbuckets, _pp(aggs, c_i32), len(aggs), _dev_ptr(out),
+        _dev_ptr(out_window_ts), _dev_ptr(out_errs), _torch_stream())
+    _check(rc, "m3gpu_rollup_batch_dev")

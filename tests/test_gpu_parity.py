@@ -1,0 +1,234 @@
+"""GPU parity: the HIP kernels vs the oracle (C restatement pinned to the
+reference's golden vectors). Bit-exact for decode/encode; exact (bit-equal
+accumulation order) for the fused rollup. Runs on an MI355X via gpurun."""
+import base64
+import json
+import os
+
+import numpy as np
+import pytest
+
+import oracle
+
+pytestmark = pytest.mark.gpu
+
+GOLDEN = os.path.join(os.path.dirname(__file__), "golden")
+
+
+@pytest.fixture(scope="module")
+def torch():
+    import torch as t
+    if not t.cuda.is_available():
+        pytest.skip("no GPU")
+    t.cuda.set_device("cuda:0")
+    return t
+
+
+@pytest.fixture(scope="module")
+def engine():
+    from m3_amd import engine as e
+    return e
+
+
+def _to_dev(torch, arr, dtype):
+    return torch.from_numpy(np.ascontiguousarray(arr).view(dtype)).to("cuda:0")
+
+
+def gpu_decode(torch, engine, streams, npts_cap, int_optimized=True):
+    from m3_amd.engine import pack_streams
+    blob, offsets, lens = pack_streams(streams)
+    n = len(streams)
+    d_blob = torch.from_numpy(blob).to("cuda:0")
+    d_off = torch.from_numpy(offsets.astype(np.int64)).to("cuda:0")
+    d_lens = torch.from_numpy(lens.astype(np.int32)).to("cuda:0")
+    out_ts = torch.empty((n, npts_cap), dtype=torch.int64, device="cuda:0")
+    out_vals = torch.empty((n, npts_cap), dtype=torch.float64, device="cuda:0")
+    out_counts = torch.empty(n, dtype=torch.int32, device="cuda:0")
+    out_errs = torch.empty(n, dtype=torch.int32, device="cuda:0")
+    engine.decode_batch_dev(d_blob, d_off, d_lens, out_ts, out_vals,
+                            out_counts, out_errs, int_optimized=int_optimized)
+    torch.cuda.synchronize()
+    return (out_ts.cpu().numpy(), out_vals.cpu().numpy(),
+            out_counts.cpu().numpy(), out_errs.cpu().numpy())
+
+
+def test_decode_production_streams(torch, engine):
+    """The 10 embedded production streams (encoder_benchmark_test.go:36-47)
+    + the regression stream decode bit-exactly vs the oracle."""
+    with open(os.path.join(GOLDEN, "production_streams.json")) as f:
+        ps = json.load(f)
+    streams = [base64.b64decode(b) for b in ps["samples"]] + \
+              [base64.b64decode(ps["regression"])]
+    g_ts, g_vals, g_counts, g_errs = gpu_decode(torch, engine, streams, 800)
+    assert np.all(g_errs == 0)
+    for i, s in enumerate(streams):
+        dec = oracle.decode_series(s, int_optimized=True)
+        n = len(dec["ts"])
+        assert g_counts[i] == n, i
+        assert np.array_equal(g_ts[i, :n], dec["ts"]), i
+        assert np.array_equal(g_vals[i, :n].view(np.uint64),
+                              np.asarray(dec["vals"]).view(np.uint64)), i
+
+
+@pytest.mark.parametrize("intopt", [True, False])
+def test_decode_random_roundtrip_vs_oracle(torch, engine, intopt):
+    """Oracle-encoded random distributions (all four §8d kinds + unit changes
+    + annotations) decode bit-exactly on the GPU."""
+    from m3_amd.workload import gen_chunk
+    rng = np.random.default_rng(1234)
+    nseries, npts = 512, 300
+    ts, vals = gen_chunk(0, nseries, npts)
+    # perturb cadence for irregular timestamps on half the series
+    jitter = rng.integers(0, 5, (nseries // 2, npts)) * 10**9
+    ts[: nseries // 2] += np.cumsum(jitter, axis=1)
+    streams = []
+    for i in range(nseries):
+        streams.append(oracle.encode_series(ts[i], vals[i], start_ns=int(ts[i, 0]),
+                                            int_optimized=intopt))
+    g_ts, g_vals, g_counts, g_errs = gpu_decode(torch, engine, streams, npts,
+                                                int_optimized=intopt)
+    assert np.all(g_errs == 0)
+    assert np.all(g_counts == npts)
+    assert np.array_equal(g_ts, ts)
+    assert np.array_equal(g_vals.view(np.uint64), vals.view(np.uint64))
+
+
+def test_decode_with_markers(torch, engine):
+    """Streams with time-unit changes and annotations (markers mid-stream)."""
+    rng = np.random.default_rng(7)
+    n, npts = 64, 200
+    streams, exp_ts, exp_vals = [], [], []
+    for i in range(n):
+        ts = 1427162462 * 10**9 + np.cumsum(rng.integers(1, 900, npts)) * 10**9
+        vals = np.round(rng.random(npts) * 1e5, 3)
+        units = np.full(npts, 1, np.uint8)
+        units[0] = 2
+        units[npts // 2] = 3
+        anns = [b"foo" if j < 3 else (b"" if j != 50 else b"mid-annotation")
+                for j in range(npts)]
+        streams.append(oracle.encode_series(ts, vals, units=units, annotations=anns,
+                                            start_ns=int(ts[0]) - 1))
+        exp_ts.append(ts)
+        exp_vals.append(vals)
+    g_ts, g_vals, g_counts, g_errs = gpu_decode(torch, engine, streams, npts)
+    assert np.all(g_errs == 0)
+    assert np.all(g_counts == npts)
+    assert np.array_equal(g_ts, np.stack(exp_ts))
+    assert np.array_equal(np.stack(exp_vals), g_vals)
+
+
+@pytest.mark.parametrize("intopt", [True, False])
+def test_encode_bit_exact_vs_oracle(torch, engine, intopt):
+    """GPU encoder output is byte-identical to the oracle encoder (which is
+    byte-identical to the reference on its golden vectors)."""
+    from m3_amd.workload import gen_chunk
+    nseries, npts = 512, 300
+    ts, vals = gen_chunk(0, nseries, npts)
+    counts = np.full(nseries, npts, np.uint32)
+    d_ts = torch.from_numpy(ts).to("cuda:0")
+    d_vals = torch.from_numpy(vals).to("cuda:0")
+    d_counts = torch.from_numpy(counts.astype(np.int32)).to("cuda:0")
+    out_stride = (24 * npts + 32 + 7) & ~7
+    d_out = torch.zeros((nseries, out_stride), dtype=torch.uint8, device="cuda:0")
+    d_lens = torch.empty(nseries, dtype=torch.int32, device="cuda:0")
+    d_errs = torch.empty(nseries, dtype=torch.int32, device="cuda:0")
+    engine.encode_batch_dev(d_ts, d_vals, d_counts, d_out, d_lens, d_errs,
+                            int_optimized=intopt)
+    torch.cuda.synchronize()
+    errs = d_errs.cpu().numpy()
+    assert np.all(errs == 0)
+    lens = d_lens.cpu().numpy()
+    rows = d_out.cpu().numpy()
+    o_rows, o_lens = oracle.encode_batch(ts, vals, counts, int_optimized=intopt)
+    assert np.array_equal(lens, o_lens)
+    for i in range(nseries):
+        assert bytes(rows[i, :lens[i]]) == bytes(o_rows[i, :o_lens[i]]), i
+
+
+@pytest.mark.parametrize("metric,aggs", [
+    ("counter", ["sum", "min", "max", "count", "mean", "sumsq", "stdev"]),
+    ("gauge", ["last", "min", "max", "mean", "count", "sum", "sumsq", "stdev"]),
+    ("timer", ["sum", "sumsq", "mean", "min", "max", "count", "stdev",
+               "median", "p50", "p95", "p99"]),
+])
+def test_rollup_vs_oracle(torch, engine, metric, aggs):
+    """Fused decode->rollup vs oracle full-CKMS rollup of the decoded data.
+    Timer aggs are the m3 default timer set (types_options.go:125-143)."""
+    from m3_amd.engine import pack_streams
+    rng = np.random.default_rng(17)
+    nseries, npts = 256, 240
+    start = (1427162462 * 10**9 // (60 * 10**9)) * 60 * 10**9
+    ts = start + np.arange(npts, dtype=np.int64) * 10 * 10**9
+    ts = np.broadcast_to(ts, (nseries, npts)).copy()
+    # irregular cadence for some series (gaps -> empty buckets)
+    ts[:32] += np.cumsum(rng.integers(0, 30, (32, npts)), axis=1) * 10**9 * 60
+    if metric == "counter":
+        vals = rng.integers(-10**6, 10**6, (nseries, npts)).astype(np.float64)
+    else:
+        vals = np.round(rng.random((nseries, npts)) * 1e4, 4)
+        if metric == "gauge":
+            vals[0, ::7] = np.nan
+    counts = np.full(nseries, npts, np.uint32)
+    streams = [oracle.encode_series(ts[i], vals[i], start_ns=int(ts[i, 0]))
+               for i in range(nseries)]
+    window = 60 * 10**9
+    nbuckets = 130 if metric != "counter" else 41
+    # oracle rollup over the decoded points
+    mt = dict(counter=oracle.METRIC_COUNTER, gauge=oracle.METRIC_GAUGE,
+              timer=oracle.METRIC_TIMER)[metric]
+    o_out, o_wts = oracle.rollup_batch(ts, vals, counts, mt, window, nbuckets, aggs)
+    # gpu fused rollup straight from the streams
+    blob, offsets, lens = pack_streams(streams)
+    d_blob = torch.from_numpy(blob).to("cuda:0")
+    d_off = torch.from_numpy(offsets.astype(np.int64)).to("cuda:0")
+    d_lens = torch.from_numpy(lens.astype(np.int32)).to("cuda:0")
+    out = torch.empty((nseries, nbuckets, len(aggs)), dtype=torch.float64,
+                      device="cuda:0")
+    wts = torch.empty((nseries, nbuckets), dtype=torch.int64, device="cuda:0")
+    errs = torch.empty(nseries, dtype=torch.int32, device="cuda:0")
+    engine.rollup_batch_dev(d_blob, d_off, d_lens, mt, window, nbuckets, aggs,
+                            out, wts, errs)
+    torch.cuda.synchronize()
+    assert np.all(errs.cpu().numpy() == 0)
+    assert np.array_equal(wts.cpu().numpy(), o_wts)
+    g = out.cpu().numpy()
+    eq = np.array_equal(g.view(np.uint64), o_out.view(np.uint64))
+    if not eq:
+        bad = np.argwhere(g.view(np.uint64) != o_out.view(np.uint64))
+        raise AssertionError(f"{metric}: {len(bad)} mismatches, first {bad[0]}: "
+                             f"gpu={g[tuple(bad[0])]} oracle={o_out[tuple(bad[0])]}")
+
+
+def test_decode_error_reporting(torch, engine):
+    """Truncated streams must flag per-series errors, not crash or fall back."""
+    good = oracle.encode_series(
+        [1427162462 * 10**9 + i * 10**9 for i in range(50)],
+        [float(i) for i in range(50)], start_ns=1427162462 * 10**9)
+    streams = [good, good[: len(good) // 2], b"\x00" * 8, good]
+    g_ts, g_vals, g_counts, g_errs = gpu_decode_no_check(torch, engine, streams, 64)
+    assert g_errs[0] == 0 and g_errs[3] == 0
+    assert g_counts[0] == 50 and g_counts[3] == 50
+    assert g_errs[1] != 0  # truncated mid-stream
+
+
+def gpu_decode_no_check(torch, engine, streams, npts_cap):
+    from m3_amd.engine import pack_streams
+    blob, offsets, lens = pack_streams(streams)
+    n = len(streams)
+    d_blob = torch.from_numpy(blob).to("cuda:0")
+    d_off = torch.from_numpy(offsets.astype(np.int64)).to("cuda:0")
+    d_lens = torch.from_numpy(lens.astype(np.int32)).to("cuda:0")
+    out_ts = torch.zeros((n, npts_cap), dtype=torch.int64, device="cuda:0")
+    out_vals = torch.zeros((n, npts_cap), dtype=torch.float64, device="cuda:0")
+    out_counts = torch.empty(n, dtype=torch.int32, device="cuda:0")
+    out_errs = torch.empty(n, dtype=torch.int32, device="cuda:0")
+    engine.decode_batch_dev(d_blob, d_off, d_lens, out_ts, out_vals,
+                            out_counts, out_errs)
+    torch.cuda.synchronize()
+    return (out_ts.cpu().numpy(), out_vals.cpu().numpy(),
+            out_counts.cpu().numpy(), out_errs.cpu().numpy())
+
+
+def test_smoke_entry(torch, engine):
+    import __graft_entry__
+    __graft_entry__.smoke()
