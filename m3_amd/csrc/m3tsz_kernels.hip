@@ -92,58 +92,61 @@ __device__ __forceinline__ double go_modf(double v, double* ip) {
  * contract: stream starts 8B-aligned, buffer zero-padded to 8B. */
 
 struct BitReader {
+    /* 128-bit register window (cur:rem valid bits, then nxt:nxt_bits). The
+     * NEXT word is prefetched as soon as it is consumed, so its HBM/L2
+     * latency overlaps parsing of up to 64 buffered bits, and peek_bits
+     * never issues a load. Byte-stream semantics identical to
+     * istream.go:73-115 over reader64.go:40-80. */
     const uint64_t* words; /* aligned start of this stream */
     int64_t len;           /* true byte length */
-    int64_t index;         /* byte index of next word (multiple of 8) */
-    uint64_t current;      /* left-aligned buffered bits */
-    uint32_t remaining;    /* valid bits in current */
+    int64_t index;         /* byte index of next word to load (multiple of 8) */
+    uint64_t cur;          /* left-aligned buffered bits */
+    uint32_t rem;          /* valid bits in cur */
+    uint64_t nxt;          /* prefetched following word (left-aligned) */
+    uint32_t nxt_bits;     /* valid bits in nxt */
 
+    __device__ __forceinline__ void refill() {
+        if (index < len) {
+            nxt = __builtin_bswap64(words[index >> 3]);
+            int64_t avail = len - index;
+            nxt_bits = avail >= 8 ? 64 : (uint32_t)(8 * avail);
+            index += 8;
+        } else {
+            nxt = 0;
+            nxt_bits = 0;
+        }
+    }
     __device__ void init(const uint8_t* base, uint64_t off, uint32_t l) {
         words = (const uint64_t*)(base + off);
         len = l;
         index = 0;
-        current = 0;
-        remaining = 0;
-    }
-    /* reader64.Read64: word + bit count; zero-padded tail comes for free */
-    __device__ __forceinline__ int read64(uint64_t* w, uint32_t* nbits) {
-        if (index >= len) return M3GPU_SERIES_EOF;
-        uint64_t v = __builtin_bswap64(words[index >> 3]);
-        int64_t avail = len - index;
-        if (avail >= 8) { *nbits = 64; }
-        else { *nbits = (uint32_t)(8 * avail); }
-        index += 8;
-        *w = v;
-        return 0;
+        cur = 0;
+        rem = 0;
+        refill();
     }
     __device__ __forceinline__ int read_bits(uint32_t n, uint64_t* out) {
-        uint64_t res = n ? (current >> (64 - n)) : 0;
-        if (n <= remaining) {
-            current = (n >= 64) ? 0 : (current << n);
-            remaining -= n;
-            *out = res;
+        if (n <= rem) {
+            *out = n ? (cur >> (64 - n)) : 0;
+            cur = (n >= 64) ? 0 : (cur << n);
+            rem -= n;
             return 0;
         }
-        uint32_t needed = n - remaining;
-        uint64_t w; uint32_t nb;
-        int err = read64(&w, &nb);
-        if (err) return err;
-        if (nb < needed) return M3GPU_SERIES_EOF;
-        current = (needed >= 64) ? 0 : (w << needed);
-        remaining = nb - needed;
-        *out = res | (w >> (64 - needed));
+        uint32_t need = n - rem;
+        if (nxt_bits < need) return M3GPU_SERIES_EOF;
+        uint64_t res = rem ? (cur >> (64 - rem)) : 0;
+        res = ((need >= 64) ? 0 : (res << need)) | (nxt >> (64 - need));
+        cur = (need >= 64) ? 0 : (nxt << need);
+        rem = nxt_bits - need;
+        refill();
+        *out = res;
         return 0;
     }
     __device__ __forceinline__ int peek_bits(uint32_t n, uint64_t* out) {
-        if (n <= remaining) { *out = n ? (current >> (64 - n)) : 0; return 0; }
-        uint64_t res = n ? (current >> (64 - n)) : 0;
-        uint32_t needed = n - remaining;
-        if (index >= len) return M3GPU_SERIES_EOF;
-        uint64_t w = __builtin_bswap64(words[index >> 3]);
-        int64_t avail = len - index;
-        uint32_t nb = avail >= 8 ? 64 : (uint32_t)(8 * avail);
-        if (nb < needed) return M3GPU_SERIES_EOF;
-        *out = res | (w >> (64 - needed));
+        if (n <= rem) { *out = n ? (cur >> (64 - n)) : 0; return 0; }
+        uint32_t need = n - rem;
+        if (nxt_bits < need) return M3GPU_SERIES_EOF;
+        uint64_t res = rem ? (cur >> (64 - rem)) : 0;
+        *out = ((need >= 64) ? 0 : (res << need)) | (nxt >> (64 - need));
         return 0;
     }
 };
