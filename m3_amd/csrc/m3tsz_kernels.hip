@@ -141,6 +141,18 @@ struct BitReader {
         *out = res;
         return 0;
     }
+    /* consume n bits already validated by a successful peek_bits(n) */
+    __device__ __forceinline__ void consume(uint32_t n) {
+        if (n <= rem) {
+            cur = (n >= 64) ? 0 : (cur << n);
+            rem -= n;
+            return;
+        }
+        uint32_t need = n - rem;
+        cur = (need >= 64) ? 0 : (nxt << need);
+        rem = nxt_bits - need;
+        refill();
+    }
     __device__ __forceinline__ int peek_bits(uint32_t n, uint64_t* out) {
         if (n <= rem) { *out = n ? (cur >> (64 - n)) : 0; return 0; }
         uint32_t need = n - rem;
@@ -272,38 +284,59 @@ struct Decoder {
     }
 
     /* readMarkerOrDeltaOfDelta (:237-248) with tryReadMarker (:174-235)
-     * unrolled into a loop (annotation/timeunit markers chain). */
+     * unrolled into a loop (annotation/timeunit markers chain), and the
+     * 11-bit marker peek FUSED with the DoD bucket decode: one peek
+     * classifies {dod==0 | marker | bucket} (GPU-relevant consequence (ii)
+     * of the grammar). Bit-identical to the sequential reference reads;
+     * falls back to the bit-by-bit path near end-of-stream (short peek)
+     * and for unit-change/no-scheme states. */
     __device__ int read_marker_or_dod(int64_t* out) {
         for (;;) {
             uint64_t ov;
-            if (r.peek_bits(MARKER_BITS, &ov) != 0) break; /* peek error => not a marker */
-            if ((ov >> 2) != MARKER_OPCODE) break;
-            uint64_t marker = ov & 0x3;
-            uint64_t discard;
-            int err;
-            if (marker == MARKER_EOS) {
-                err = r.read_bits(MARKER_BITS, &discard);
-                if (err) return err;
-                done = true;
+            if (tu_changed || !have_scheme) return read_dod(out);
+            if (r.peek_bits(MARKER_BITS, &ov) != 0) return read_dod(out);
+            if ((ov >> 10) == 0) { /* zero bucket: the 1-cadence fast path */
+                r.consume(1);
                 *out = 0;
                 return 0;
-            } else if (marker == MARKER_ANNOTATION) {
-                err = r.read_bits(MARKER_BITS, &discard);
-                if (err) return err;
-                err = skip_annotation();
-                if (err) return err;
-                continue;
-            } else if (marker == MARKER_TIMEUNIT) {
-                err = r.read_bits(MARKER_BITS, &discard);
-                if (err) return err;
-                err = read_time_unit();
-                if (err) return err;
-                continue;
-            } else {
-                break; /* unknown marker value: parse as dod (:232-234) */
             }
+            if ((ov >> 2) == MARKER_OPCODE) {
+                uint64_t marker = ov & 0x3;
+                int err;
+                if (marker == MARKER_EOS) {
+                    r.consume(MARKER_BITS);
+                    done = true;
+                    *out = 0;
+                    return 0;
+                } else if (marker == MARKER_ANNOTATION) {
+                    r.consume(MARKER_BITS);
+                    err = skip_annotation();
+                    if (err) return err;
+                    continue;
+                } else if (marker == MARKER_TIMEUNIT) {
+                    r.consume(MARKER_BITS);
+                    err = read_time_unit();
+                    if (err) return err;
+                    continue;
+                }
+                /* unknown marker value: falls through and parses as a
+                 * bucket (:232-234) — the 10... prefix selects bucket 0,
+                 * exactly as the sequential reads would. */
+            }
+            /* bucket select: count leading ones of the top 4 bits */
+            uint32_t top4 = (uint32_t)(ov >> (MARKER_BITS - 4)) & 0xF;
+            uint32_t L = __builtin_clz(~(top4 << 28)); /* in [1,4] */
+            uint32_t ob = (L <= 3) ? L + 1 : 4;        /* opcode bits */
+            uint32_t vb = (L == 1) ? 7 : (L == 2) ? 9 : (L == 3) ? 12
+                          : (uint32_t)scheme_default_bits(scheme_unit);
+            r.consume(ob);
+            uint64_t db;
+            int err = r.read_bits(vb, &db);
+            if (err) return err;
+            if (!unit_valid(time_unit)) { *out = 0; return 0; } /* swallowed */
+            *out = sign_extend(db, vb) * UNIT_NS_D[time_unit];
+            return 0;
         }
-        return read_dod(out);
     }
 
     /* timestamp_iterator.go:137-161 + initialTimeUnit */
@@ -510,7 +543,10 @@ k_decode_batch(const uint8_t* __restrict__ blobs,
                int64_t* __restrict__ out_ts, double* __restrict__ out_vals,
                uint32_t* __restrict__ out_counts, int32_t* __restrict__ out_errs,
                uint32_t stride) {
-    const uint32_t wave = threadIdx.x / WAVE;
+    /* readfirstlane makes the series index provably wave-uniform: the
+     * whole parser then compiles to scalar (SGPR) code with scalar
+     * branches instead of exec-mask divergence sequences. */
+    const uint32_t wave = __builtin_amdgcn_readfirstlane(threadIdx.x / WAVE);
     const uint32_t lane = threadIdx.x % WAVE;
     const uint32_t series = blockIdx.x * WAVES_PER_BLOCK + wave;
     if (series >= nseries) return;
@@ -911,7 +947,10 @@ k_encode_batch(const int64_t* __restrict__ ts, const double* __restrict__ vals,
                uint32_t stride, int int_optimized, uint8_t unit,
                uint8_t* __restrict__ out_bytes, uint32_t out_stride,
                uint32_t* __restrict__ out_lens, int32_t* __restrict__ out_errs) {
-    const uint32_t wave = threadIdx.x / WAVE;
+    /* readfirstlane makes the series index provably wave-uniform: the
+     * whole parser then compiles to scalar (SGPR) code with scalar
+     * branches instead of exec-mask divergence sequences. */
+    const uint32_t wave = __builtin_amdgcn_readfirstlane(threadIdx.x / WAVE);
     const uint32_t lane = threadIdx.x % WAVE;
     const uint32_t series = blockIdx.x * WAVES_PER_BLOCK + wave;
     if (series >= nseries) return;
@@ -998,7 +1037,10 @@ k_rollup_batch(const uint8_t* __restrict__ blobs,
                RollupPlan plan,
                double* __restrict__ out, int64_t* __restrict__ out_window_ts,
                int32_t* __restrict__ out_errs) {
-    const uint32_t wave = threadIdx.x / WAVE;
+    /* readfirstlane makes the series index provably wave-uniform: the
+     * whole parser then compiles to scalar (SGPR) code with scalar
+     * branches instead of exec-mask divergence sequences. */
+    const uint32_t wave = __builtin_amdgcn_readfirstlane(threadIdx.x / WAVE);
     const uint32_t lane = threadIdx.x % WAVE;
     const uint32_t series = blockIdx.x * WAVES_PER_BLOCK + wave;
     /* per-wave bucket value staging for quantiles (timer) */
@@ -1175,7 +1217,10 @@ k_compact(const uint8_t* __restrict__ src, uint32_t src_stride,
           const uint32_t* __restrict__ lens,
           const uint64_t* __restrict__ dst_offsets, uint32_t nseries,
           uint8_t* __restrict__ dst) {
-    const uint32_t wave = threadIdx.x / WAVE;
+    /* readfirstlane makes the series index provably wave-uniform: the
+     * whole parser then compiles to scalar (SGPR) code with scalar
+     * branches instead of exec-mask divergence sequences. */
+    const uint32_t wave = __builtin_amdgcn_readfirstlane(threadIdx.x / WAVE);
     const uint32_t lane = threadIdx.x % WAVE;
     const uint32_t series = blockIdx.x * WAVES_PER_BLOCK + wave;
     if (series >= nseries) return;
