@@ -92,73 +92,82 @@ __device__ __forceinline__ double go_modf(double v, double* ip) {
  * contract: stream starts 8B-aligned, buffer zero-padded to 8B. */
 
 struct BitReader {
-    /* 128-bit register window (cur:rem valid bits, then nxt:nxt_bits). The
-     * NEXT word is prefetched as soon as it is consumed, so its HBM/L2
-     * latency overlaps parsing of up to 64 buffered bits, and peek_bits
-     * never issues a load. Byte-stream semantics identical to
-     * istream.go:73-115 over reader64.go:40-80. */
-    const uint64_t* words; /* aligned start of this stream */
-    int64_t len;           /* true byte length */
-    int64_t index;         /* byte index of next word to load (multiple of 8) */
-    uint64_t cur;          /* left-aligned buffered bits */
-    uint32_t rem;          /* valid bits in cur */
-    uint64_t nxt;          /* prefetched following word (left-aligned) */
-    uint32_t nxt_bits;     /* valid bits in nxt */
+    /* Branchless 128-bit register window (hi:lo, left-aligned `loaded` real
+     * bits) over the byte stream, with a two-word software prefetch pipeline:
+     * pw is the word about to be inserted, pf the word loaded ~128 bits
+     * before first use, so refill latency overlaps parsing. Byte-stream
+     * semantics identical to istream.go:73-115 over reader64.go:40-80
+     * (EOF iff the request exceeds the stream's remaining bits; the blob's
+     * 8B zero padding reproduces the zero-filled partial tail word).
+     * Invariant: after refill, loaded > 64 unless the stream is exhausted,
+     * so peek/read of n <= 64 bits always comes straight from hi. */
+    const uint64_t* words;
+    int64_t len;        /* true byte length */
+    int64_t index;      /* byte index of the next word to prefetch */
+    uint64_t hi, lo;    /* window */
+    uint32_t loaded;    /* real bits in window */
+    int64_t bits_left;  /* real stream bits not yet consumed */
+    uint64_t pw, pf;    /* pipeline words (left-aligned) */
+    uint32_t pw_bits, pf_bits;
 
-    __device__ __forceinline__ void refill() {
+    __device__ __forceinline__ void prefetch() {
         if (index < len) {
-            nxt = __builtin_bswap64(words[index >> 3]);
+            pf = __builtin_bswap64(words[index >> 3]);
             int64_t avail = len - index;
-            nxt_bits = avail >= 8 ? 64 : (uint32_t)(8 * avail);
+            pf_bits = avail >= 8 ? 64 : (uint32_t)(8 * avail);
             index += 8;
         } else {
-            nxt = 0;
-            nxt_bits = 0;
+            pf = 0;
+            pf_bits = 0;
         }
+    }
+    __device__ __forceinline__ void refill() {
+        if (pw_bits) {
+            if (loaded == 0) {
+                hi |= pw;
+            } else if (loaded < 64) {
+                hi |= pw >> loaded;
+                lo |= pw << (64 - loaded);
+            } else {
+                lo |= pw;
+            }
+            loaded += pw_bits;
+        }
+        pw = pf;
+        pw_bits = pf_bits;
+        prefetch();
     }
     __device__ void init(const uint8_t* base, uint64_t off, uint32_t l) {
         words = (const uint64_t*)(base + off);
         len = l;
         index = 0;
-        cur = 0;
-        rem = 0;
-        refill();
+        hi = 0; lo = 0;
+        loaded = 0;
+        bits_left = (int64_t)l * 8;
+        pw = 0; pw_bits = 0;
+        prefetch();          /* word 0 -> pf */
+        refill();            /* pf -> pw, word 1 -> pf */
+        refill();            /* insert word 0, word 2 -> pf */
+        refill();            /* insert word 1: loaded > 64 (if available) */
+    }
+    __device__ __forceinline__ void advance(uint32_t n) {
+        hi = (n >= 64) ? lo : ((hi << n) | (n ? (lo >> (64 - n)) : 0));
+        lo = (n >= 64) ? 0 : (lo << n);
+        loaded -= n;
+        bits_left -= n;
+        if (loaded <= 64) refill();
     }
     __device__ __forceinline__ int read_bits(uint32_t n, uint64_t* out) {
-        if (n <= rem) {
-            *out = n ? (cur >> (64 - n)) : 0;
-            cur = (n >= 64) ? 0 : (cur << n);
-            rem -= n;
-            return 0;
-        }
-        uint32_t need = n - rem;
-        if (nxt_bits < need) return M3GPU_SERIES_EOF;
-        uint64_t res = rem ? (cur >> (64 - rem)) : 0;
-        res = ((need >= 64) ? 0 : (res << need)) | (nxt >> (64 - need));
-        cur = (need >= 64) ? 0 : (nxt << need);
-        rem = nxt_bits - need;
-        refill();
-        *out = res;
+        if ((int64_t)n > bits_left) return M3GPU_SERIES_EOF;
+        *out = n ? (hi >> (64 - n)) : 0;
+        advance(n);
         return 0;
     }
     /* consume n bits already validated by a successful peek_bits(n) */
-    __device__ __forceinline__ void consume(uint32_t n) {
-        if (n <= rem) {
-            cur = (n >= 64) ? 0 : (cur << n);
-            rem -= n;
-            return;
-        }
-        uint32_t need = n - rem;
-        cur = (need >= 64) ? 0 : (nxt << need);
-        rem = nxt_bits - need;
-        refill();
-    }
+    __device__ __forceinline__ void consume(uint32_t n) { advance(n); }
     __device__ __forceinline__ int peek_bits(uint32_t n, uint64_t* out) {
-        if (n <= rem) { *out = n ? (cur >> (64 - n)) : 0; return 0; }
-        uint32_t need = n - rem;
-        if (nxt_bits < need) return M3GPU_SERIES_EOF;
-        uint64_t res = rem ? (cur >> (64 - rem)) : 0;
-        *out = ((need >= 64) ? 0 : (res << need)) | (nxt >> (64 - need));
+        if ((int64_t)n > bits_left) return M3GPU_SERIES_EOF;
+        *out = n ? (hi >> (64 - n)) : 0;
         return 0;
     }
 };
@@ -629,59 +638,49 @@ k_decode_batch(const uint8_t* __restrict__ blobs,
  * ostream.go:133-221; lane (w & 63) stages word w, coalesced 512B flushes. */
 
 struct BitWriter {
+    /* Per-lane big-endian bit emitter producing the byte-identical stream
+     * of ostream.go:133-221. Each lane owns one output row; completed 64-bit
+     * words store directly (8B per store; rows padded to 8B). */
     uint64_t acc;       /* left-aligned bit accumulator */
     uint32_t used;      /* bits used in acc */
     uint32_t nwords;    /* full words emitted */
-    uint64_t staged;    /* this lane's staged word */
     uint64_t* out;      /* aligned output row */
     uint32_t cap_words;
-    uint32_t lane;
     int err;
 
-    __device__ void init(uint8_t* row, uint32_t cap_bytes, uint32_t l) {
-        acc = 0; used = 0; nwords = 0; staged = 0;
+    __device__ void init(uint8_t* row, uint32_t cap_bytes) {
+        acc = 0; used = 0; nwords = 0;
         out = (uint64_t*)row;
         cap_words = cap_bytes / 8;
-        lane = l;
         err = 0;
     }
     __device__ __forceinline__ void emit_word(uint64_t w) {
-        if ((nwords & 63) == lane) staged = __builtin_bswap64(w);
-        nwords++;
-        if ((nwords & 63) == 0) {
-            uint32_t base = nwords - 64;
-            if (nwords > cap_words) { err = M3GPU_SERIES_CAPACITY; return; }
-            out[base + lane] = staged;
-        }
+        if (nwords >= cap_words) { err = M3GPU_SERIES_CAPACITY; return; }
+        out[nwords++] = __builtin_bswap64(w);
     }
     __device__ __forceinline__ void write_bits(uint64_t v, uint32_t n) {
         if (n == 0 || err) return;
         v = (n >= 64) ? v : (v & ((1ULL << n) - 1)); /* low n bits */
         uint32_t space = 64 - used;
         if (n <= space) {
-            acc |= (space - n >= 64) ? 0 : (v << (space - n));
+            acc |= v << (space - n);
             used += n;
             if (used == 64) { emit_word(acc); acc = 0; used = 0; }
         } else {
-            uint32_t hi = space;          /* bits that fit */
             uint32_t lo = n - space;      /* remainder */
-            if (hi) acc |= v >> lo;
+            acc |= v >> lo;
             emit_word(acc);
             acc = (lo >= 64) ? 0 : (v << (64 - lo));
             used = lo;
         }
     }
     __device__ __forceinline__ void write_bit(uint32_t b) { write_bits(b, 1); }
-    /* Flush staged words + partial tail. Returns total byte length. */
+    /* Flush the partial tail word. Returns total byte length. */
     __device__ uint32_t finish() {
-        uint32_t rem_words = nwords & 63;
-        uint32_t base = nwords - rem_words;
-        if (nwords > cap_words) { err = M3GPU_SERIES_CAPACITY; return 0; }
-        if (lane < rem_words) out[base + lane] = staged;
         uint32_t nbytes = nwords * 8;
         if (used > 0) {
-            if (nwords + 1 > cap_words) { err = M3GPU_SERIES_CAPACITY; return 0; }
-            if (lane == 0) out[nwords] = __builtin_bswap64(acc); /* zero-padded */
+            if (nwords >= cap_words) { err = M3GPU_SERIES_CAPACITY; return 0; }
+            out[nwords] = __builtin_bswap64(acc); /* zero-padded */
             nbytes += (used + 7) / 8;
         }
         return nbytes;
@@ -730,9 +729,9 @@ struct Encoder {
     bool has_written_first, is_float, int_optimized;
     uint32_t num_encoded;
 
-    __device__ void init(uint8_t* row, uint32_t cap_bytes, uint32_t lane,
+    __device__ void init(uint8_t* row, uint32_t cap_bytes,
                          int64_t start_ns, bool intopt, uint8_t default_unit) {
-        w.init(row, cap_bytes, lane);
+        w.init(row, cap_bytes);
         prev_time = start_ns;
         prev_time_delta = 0;
         prev_xor = 0; prev_float_bits = 0;
@@ -982,12 +981,9 @@ k_encode_batch(const int64_t* __restrict__ ts, const double* __restrict__ vals,
                uint32_t stride, int int_optimized, uint8_t unit,
                uint8_t* __restrict__ out_bytes, uint32_t out_stride,
                uint32_t* __restrict__ out_lens, int32_t* __restrict__ out_errs) {
-    /* readfirstlane makes the series index provably wave-uniform: the
-     * whole parser then compiles to scalar (SGPR) code with scalar
-     * branches instead of exec-mask divergence sequences. */
-    const uint32_t wave = __builtin_amdgcn_readfirstlane(threadIdx.x / WAVE);
-    const uint32_t lane = threadIdx.x % WAVE;
-    const uint32_t series = blockIdx.x * WAVES_PER_BLOCK + wave;
+    /* ONE SERIES PER LANE (like decode): 64 independent encoders per wave,
+     * per-lane bit emitters storing completed words to their own rows. */
+    const uint32_t series = blockIdx.x * BLOCK_THREADS + threadIdx.x;
     if (series >= nseries) return;
 
     const int64_t* row_ts = ts + (uint64_t)series * stride;
@@ -995,7 +991,7 @@ k_encode_batch(const int64_t* __restrict__ ts, const double* __restrict__ vals,
     uint32_t n = counts[series];
 
     Encoder e;
-    e.init(out_bytes + (uint64_t)series * out_stride, out_stride, lane,
+    e.init(out_bytes + (uint64_t)series * out_stride, out_stride,
            n ? row_ts[0] : 0, int_optimized != 0, unit);
 
     int err = 0;
@@ -1008,10 +1004,8 @@ k_encode_batch(const int64_t* __restrict__ ts, const double* __restrict__ vals,
         len = e.finalize();
         err = e.w.err;
     }
-    if (lane == 0) {
-        out_lens[series] = err ? 0 : len;
-        out_errs[series] = err;
-    }
+    out_lens[series] = err ? 0 : len;
+    out_errs[series] = err;
 }
 
 /* ===================== fused decode -> rollup kernel ===================== */
