@@ -53,6 +53,8 @@ def parse_args():
     p.add_argument("--cpu-sample-series", type=int, default=0,
                    help="series in the cpu_baseline sample (0 = auto-size to ~15s)")
     p.add_argument("--skip-cpu-baseline", action="store_true")
+    p.add_argument("--no-sort", action="store_true",
+                   help="disable length-sorted wave scheduling")
     return p.parse_args()
 
 
@@ -91,10 +93,15 @@ def main():
     out_vals = torch.empty((n_local, args.npts), dtype=torch.float64, device=device)
     out_counts = torch.empty(n_local, dtype=torch.int32, device=device)
     out_errs = torch.empty(n_local, dtype=torch.int32, device=device)
+    # length-sorted wave scheduling: every wavefront decodes 64 similar-cost
+    # streams (stream length is known before decode)
+    d_perm = None
+    if not args.no_sort:
+        d_perm = torch.argsort(d_lens).to(torch.int32)
 
     def step():
         engine.decode_batch_dev(d_blob, d_offsets, d_lens, out_ts, out_vals,
-                                out_counts, out_errs)
+                                out_counts, out_errs, d_perm=d_perm)
 
     # correctness gate outside the timed region: every series decodes fully
     step()

@@ -82,6 +82,16 @@ int m3gpu_decode_batch_dev(
     int64_t* d_out_ts, double* d_out_vals, uint32_t* d_out_counts,
     int32_t* d_out_errs, uint32_t stride, void* hip_stream);
 
+/* Scheduling variant: d_perm (device, nseries int32) permutes which stream
+ * each lane decodes (outputs still land at the stream's own row). Passing a
+ * length-sorted order gives every wavefront 64 similar-cost streams. */
+int m3gpu_decode_batch_dev_perm(
+    const uint8_t* d_blobs, const uint64_t* d_offsets, const uint32_t* d_lens,
+    const int32_t* d_perm,
+    uint32_t nseries, int int_optimized, uint8_t default_unit,
+    int64_t* d_out_ts, double* d_out_vals, uint32_t* d_out_counts,
+    int32_t* d_out_errs, uint32_t stride, void* hip_stream);
+
 int m3gpu_decode_batch(
     const uint8_t* blobs, uint64_t blobs_len,
     const uint64_t* offsets, const uint32_t* lens,

@@ -563,6 +563,7 @@ __global__ void __launch_bounds__(BLOCK_THREADS)
 k_decode_batch(const uint8_t* __restrict__ blobs,
                const uint64_t* __restrict__ offsets,
                const uint32_t* __restrict__ lens,
+               const int32_t* __restrict__ perm,
                uint32_t nseries, int int_optimized, uint8_t default_unit,
                int64_t* __restrict__ out_ts, double* __restrict__ out_vals,
                uint32_t* __restrict__ out_counts, int32_t* __restrict__ out_errs,
@@ -570,17 +571,19 @@ k_decode_batch(const uint8_t* __restrict__ blobs,
     const uint32_t lane = threadIdx.x & (WAVE - 1);
     const uint32_t wave = __builtin_amdgcn_readfirstlane(threadIdx.x >> 6);
     const uint32_t s_base = blockIdx.x * BLOCK_THREADS + wave * WAVE;
-    const uint32_t series = s_base + lane;
+    const uint32_t slot = s_base + lane;
+    /* optional scheduling permutation (e.g. length-sorted): waves then get
+     * 64 similar-cost streams, removing intra-wave and intra-CU skew.
+     * Purely a schedule: outputs still land in series order. */
+    const uint32_t series = (perm && slot < nseries) ? (uint32_t)perm[slot] : slot;
 
     __shared__ int64_t ts_tile_all[WAVES_PER_BLOCK][WAVE][DEC_TILE + 1];
     __shared__ double val_tile_all[WAVES_PER_BLOCK][WAVE][DEC_TILE + 1];
     int64_t (*ts_tile)[DEC_TILE + 1] = ts_tile_all[wave];
     double (*val_tile)[DEC_TILE + 1] = val_tile_all[wave];
 
-    int64_t* row0_ts = out_ts + (uint64_t)s_base * stride;
-    double* row0_val = out_vals + (uint64_t)s_base * stride;
 
-    const bool in_range = series < nseries;
+    const bool in_range = slot < nseries;
     Decoder d;
     if (in_range)
         d.init(blobs, offsets[series], lens[series], int_optimized != 0, default_unit);
@@ -598,10 +601,11 @@ k_decode_batch(const uint8_t* __restrict__ blobs,
         for (uint32_t j = 0; j < WAVE / DEC_TILE; j++) {
             uint32_t r = r0 + j * (WAVE / DEC_TILE);
             uint32_t c = (uint32_t)__shfl((int)cnt, (int)r);
+            uint64_t row = (uint64_t)__shfl((int)series, (int)r);
             uint32_t pt = base_pt + p;
             if (pt < c) {
-                row0_ts[(uint64_t)r * stride + pt] = ts_tile[r][p];
-                row0_val[(uint64_t)r * stride + pt] = val_tile[r][p];
+                out_ts[row * stride + pt] = ts_tile[r][p];
+                out_vals[row * stride + pt] = val_tile[r][p];
             }
         }
         __builtin_amdgcn_wave_barrier();
@@ -1290,8 +1294,13 @@ int m3gpu_init(int device) {
 
 void m3gpu_shutdown(void) { g_inited = false; }
 
+/* wave-per-series kernels (rollup, compact): 4 series per block */
 static inline uint32_t grid_for(uint32_t nseries) {
     return (nseries + WAVES_PER_BLOCK - 1) / WAVES_PER_BLOCK;
+}
+/* lane-per-series kernels (decode, encode): 256 series per block */
+static inline uint32_t grid_lane(uint32_t nseries) {
+    return (nseries + BLOCK_THREADS - 1) / BLOCK_THREADS;
 }
 
 int m3gpu_decode_batch_dev(
@@ -1299,13 +1308,25 @@ int m3gpu_decode_batch_dev(
     uint32_t nseries, int int_optimized, uint8_t default_unit,
     int64_t* d_out_ts, double* d_out_vals, uint32_t* d_out_counts,
     int32_t* d_out_errs, uint32_t stride, void* hip_stream) {
+    return m3gpu_decode_batch_dev_perm(d_blobs, d_offsets, d_lens, NULL,
+                                       nseries, int_optimized, default_unit,
+                                       d_out_ts, d_out_vals, d_out_counts,
+                                       d_out_errs, stride, hip_stream);
+}
+
+int m3gpu_decode_batch_dev_perm(
+    const uint8_t* d_blobs, const uint64_t* d_offsets, const uint32_t* d_lens,
+    const int32_t* d_perm,
+    uint32_t nseries, int int_optimized, uint8_t default_unit,
+    int64_t* d_out_ts, double* d_out_vals, uint32_t* d_out_counts,
+    int32_t* d_out_errs, uint32_t stride, void* hip_stream) {
     if (!nseries) return M3GPU_OK;
     hipStream_t s = (hipStream_t)hip_stream;
-    hipLaunchKernelGGL(m3::k_decode_batch, dim3(grid_for(nseries)),
+    hipLaunchKernelGGL(m3::k_decode_batch, dim3(grid_lane(nseries)),
                        dim3(BLOCK_THREADS), 0, s,
-                       d_blobs, d_offsets, d_lens, nseries, int_optimized,
-                       default_unit, d_out_ts, d_out_vals, d_out_counts,
-                       d_out_errs, stride);
+                       d_blobs, d_offsets, d_lens, d_perm, nseries,
+                       int_optimized, default_unit, d_out_ts, d_out_vals,
+                       d_out_counts, d_out_errs, stride);
     HIP_TRY(hipGetLastError());
     return M3GPU_OK;
 }
@@ -1321,7 +1342,7 @@ int m3gpu_encode_batch_dev(
         return M3GPU_ERR_BADARG;
     }
     hipStream_t s = (hipStream_t)hip_stream;
-    hipLaunchKernelGGL(m3::k_encode_batch, dim3(grid_for(nseries)),
+    hipLaunchKernelGGL(m3::k_encode_batch, dim3(grid_lane(nseries)),
                        dim3(BLOCK_THREADS), 0, s,
                        d_ts, d_vals, d_counts, nseries, stride, int_optimized,
                        unit, d_out_bytes, out_stride, d_out_lens, d_out_errs);

@@ -61,6 +61,10 @@ def lib():
         L.m3gpu_decode_batch_dev.restype = c_int
         L.m3gpu_decode_batch_dev.argtypes = [c_vp, c_vp, c_vp, c_u32, c_int, c_u8,
                                              c_vp, c_vp, c_vp, c_vp, c_u32, c_vp]
+        L.m3gpu_decode_batch_dev_perm.restype = c_int
+        L.m3gpu_decode_batch_dev_perm.argtypes = [c_vp, c_vp, c_vp, c_vp, c_u32,
+                                                  c_int, c_u8, c_vp, c_vp, c_vp,
+                                                  c_vp, c_u32, c_vp]
         L.m3gpu_decode_batch.restype = c_int
         L.m3gpu_decode_batch.argtypes = [P(c_u8), c_u64, P(c_u64), P(c_u32), c_u32,
                                          c_int, c_u8, P(c_i64), P(c_f64), P(c_u32),
@@ -202,13 +206,16 @@ def _torch_stream():
 
 
 def decode_batch_dev(d_blob, d_offsets, d_lens, out_ts, out_vals, out_counts,
-                     out_errs, int_optimized=True, default_unit=1):
+                     out_errs, int_optimized=True, default_unit=1, d_perm=None):
     """All args are torch CUDA tensors; enqueues on the current torch stream.
-    out_ts: int64 [nseries, stride]; out_vals: float64 [nseries, stride]."""
+    out_ts: int64 [nseries, stride]; out_vals: float64 [nseries, stride].
+    d_perm (int32, optional): scheduling permutation — e.g.
+    torch.argsort(d_lens) so each wavefront decodes similar-cost streams."""
     nseries = d_lens.numel()
     stride = out_ts.shape[1]
-    rc = lib().m3gpu_decode_batch_dev(
-        _dev_ptr(d_blob), _dev_ptr(d_offsets), _dev_ptr(d_lens), nseries,
+    rc = lib().m3gpu_decode_batch_dev_perm(
+        _dev_ptr(d_blob), _dev_ptr(d_offsets), _dev_ptr(d_lens),
+        _dev_ptr(d_perm) if d_perm is not None else None, nseries,
         1 if int_optimized else 0, default_unit, _dev_ptr(out_ts),
         _dev_ptr(out_vals), _dev_ptr(out_counts), _dev_ptr(out_errs), stride,
         _torch_stream())

@@ -25,9 +25,11 @@ def gen_chunk(series0, nseries, npts, seed_base=42):
     # lane) sees a single distribution — the host controls series placement,
     # exactly as the reference's shard assignment does, and grouping similar
     # series keeps the 64 lane parsers branch-coherent.
+    import os
+    kind_shift = int(os.environ.get("M3_KIND_SHIFT", "6"))
     idx = series0 + np.arange(nseries)
     for kind in range(4):
-        rows = np.nonzero((idx >> 6) % 4 == kind)[0]
+        rows = np.nonzero((idx >> kind_shift) % 4 == kind)[0]
         if not len(rows):
             continue
         rng = np.random.default_rng(seed_base + series0 * 7 + kind)

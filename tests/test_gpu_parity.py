@@ -233,3 +233,34 @@ def gpu_decode_no_check(torch, engine, streams, npts_cap):
 def test_smoke_entry(torch, engine):
     import __graft_entry__
     __graft_entry__.smoke()
+
+
+def test_decode_perm_schedule_matches(torch, engine):
+    """Length-sorted scheduling permutation is schedule-only: outputs are
+    identical to the unpermuted decode."""
+    from m3_amd.engine import pack_streams
+    from m3_amd.workload import gen_chunk
+    nseries, npts = 300, 200
+    ts, vals = gen_chunk(0, nseries, npts)
+    streams = [oracle.encode_series(ts[i], vals[i], start_ns=int(ts[i, 0]))
+               for i in range(nseries)]
+    blob, offsets, lens = pack_streams(streams)
+    d_blob = torch.from_numpy(blob).to("cuda:0")
+    d_off = torch.from_numpy(offsets.astype(np.int64)).to("cuda:0")
+    d_lens = torch.from_numpy(lens.astype(np.int32)).to("cuda:0")
+    outs = []
+    for perm in (None, torch.argsort(d_lens).to(torch.int32)):
+        out_ts = torch.zeros((nseries, npts), dtype=torch.int64, device="cuda:0")
+        out_vals = torch.zeros((nseries, npts), dtype=torch.float64, device="cuda:0")
+        out_counts = torch.empty(nseries, dtype=torch.int32, device="cuda:0")
+        out_errs = torch.empty(nseries, dtype=torch.int32, device="cuda:0")
+        engine.decode_batch_dev(d_blob, d_off, d_lens, out_ts, out_vals,
+                                out_counts, out_errs, d_perm=perm)
+        torch.cuda.synchronize()
+        assert np.all(out_errs.cpu().numpy() == 0)
+        outs.append((out_ts.cpu().numpy(), out_vals.cpu().numpy(),
+                     out_counts.cpu().numpy()))
+    assert np.array_equal(outs[0][0], outs[1][0])
+    assert np.array_equal(outs[0][1].view(np.uint64), outs[1][1].view(np.uint64))
+    assert np.array_equal(outs[0][2], outs[1][2])
+    assert np.array_equal(outs[0][0], ts)
