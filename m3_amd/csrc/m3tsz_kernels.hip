@@ -595,7 +595,6 @@ k_decode_batch(const uint8_t* __restrict__ blobs,
 
     auto flush = [&](uint32_t base_pt) {
         __builtin_amdgcn_wave_barrier();
-        __threadfence_block();
         const uint32_t p = lane & (DEC_TILE - 1);
         const uint32_t r0 = lane >> 3;
         for (uint32_t j = 0; j < WAVE / DEC_TILE; j++) {

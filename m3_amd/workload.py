@@ -27,9 +27,13 @@ def gen_chunk(series0, nseries, npts, seed_base=42):
     # series keeps the 64 lane parsers branch-coherent.
     import os
     kind_shift = int(os.environ.get("M3_KIND_SHIFT", "6"))
+    kind_only = os.environ.get("M3_KIND_ONLY")
     idx = series0 + np.arange(nseries)
     for kind in range(4):
-        rows = np.nonzero((idx >> kind_shift) % 4 == kind)[0]
+        if kind_only is not None:
+            rows = np.arange(nseries) if kind == int(kind_only) else np.array([], int)
+        else:
+            rows = np.nonzero((idx >> kind_shift) % 4 == kind)[0]
         if not len(rows):
             continue
         rng = np.random.default_rng(seed_base + series0 * 7 + kind)
