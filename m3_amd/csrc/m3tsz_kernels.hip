@@ -302,6 +302,17 @@ struct Decoder {
      * falls back to the bit-by-bit path near end-of-stream (short peek)
      * and for unit-change/no-scheme states. */
     __device__ int read_marker_or_dod(int64_t* out) {
+        /* dominant path: a regular cadence emits dod == 0 = a single 0 bit,
+         * which can never be a marker (markers start with 1) — classify it
+         * from a 1-bit peek before the full 11-bit marker peek. */
+        if (!tu_changed && have_scheme) {
+            uint64_t b1;
+            if (r.peek_bits(1, &b1) == 0 && b1 == 0) {
+                r.consume(1);
+                *out = 0;
+                return 0;
+            }
+        }
         for (;;) {
             uint64_t ov;
             /* markers are checked BEFORE any scheme/unit-change gating
@@ -399,6 +410,44 @@ struct Decoder {
         if (err) return err;
         prev_float_bits = vb;
         prev_xor = vb;
+        return 0;
+    }
+    /* Fused XOR-field read: one 64-bit peek classifies the control bits
+     * and extracts contained payloads; falls back to the stepwise reads
+     * near end-of-stream (identical error semantics). */
+    __device__ int read_next_float_fused() {
+        uint64_t w;
+        if (r.peek_bits(64, &w) != 0) return read_next_float();
+        if ((w >> 63) == 0) { r.consume(1); prev_xor = 0; return 0; }
+        if ((w >> 62) == 0x2) { /* contained */
+            uint32_t lead = prev_xor ? __builtin_clzll(prev_xor) : 64;
+            uint32_t trail = prev_xor ? __builtin_ctzll(prev_xor) : 0;
+            uint32_t nmean = 64 - lead - trail;
+            if (nmean <= 62) {
+                uint64_t mb = nmean ? ((w << 2) >> (64 - nmean)) : 0;
+                r.consume(2 + nmean);
+                prev_xor = mb << trail;
+                prev_float_bits ^= prev_xor;
+                return 0;
+            }
+            r.consume(2);
+            uint64_t mb;
+            int err = r.read_bits(nmean, &mb);
+            if (err) return err;
+            prev_xor = mb << trail;
+            prev_float_bits ^= prev_xor;
+            return 0;
+        }
+        /* uncontained: 11 + 6b lead + 6b (nmean-1) + payload */
+        uint64_t lead = (w >> 56) & 0x3f;
+        uint64_t nmean = ((w >> 50) & 0x3f) + 1;
+        r.consume(14);
+        uint64_t mb;
+        int err = r.read_bits((uint32_t)nmean, &mb);
+        if (err) return err;
+        uint64_t trail = 64 - lead - nmean;
+        prev_xor = mb << trail;
+        prev_float_bits ^= prev_xor;
         return 0;
     }
     __device__ int read_next_float() {
@@ -500,7 +549,21 @@ struct Decoder {
         return read_int_val_diff();
     }
     __device__ int read_next_value() {
-        if (!int_optimized) return read_next_float();
+        if (!int_optimized) return read_next_float_fused();
+        /* fused common case: '1' + sign + sig-bit diff from one peek */
+        uint64_t w;
+        if (!is_float && sig <= 62 && r.peek_bits(64, &w) == 0) {
+            if (w >> 63) { /* opcodeNoUpdate -> int diff */
+                uint64_t bits = (w << 1) >> (63 - sig); /* sign + payload */
+                r.consume(2 + sig);
+                double sgn = -1.0;
+                if ((bits >> sig) == 1) { sgn = 1.0; bits ^= (1ULL << sig); }
+                int_val += sgn * (double)bits;
+                return 0;
+            }
+            if ((w >> 62) == 0x1) { r.consume(2); return 0; } /* repeat */
+            /* '00': mode/sig/mult update (rare) — stepwise below */
+        }
         uint64_t b;
         int err = r.read_bits(1, &b);
         if (err) return err;
@@ -523,7 +586,7 @@ struct Decoder {
             is_float = false;
             return 0;
         }
-        if (is_float) return read_next_float();
+        if (is_float) return read_next_float_fused();
         return read_int_val_diff();
     }
 
@@ -557,7 +620,7 @@ struct Decoder {
  *   flush step j: lane l stores row (l>>3)+8j, point (l&7) — 8 consecutive
  *   8B addresses per row = full 64B line utilization for ts[] and val[]. */
 
-#define DEC_TILE 8
+#define DEC_TILE 4
 
 __global__ void __launch_bounds__(BLOCK_THREADS)
 k_decode_batch(const uint8_t* __restrict__ blobs,
@@ -596,15 +659,25 @@ k_decode_batch(const uint8_t* __restrict__ blobs,
     auto flush = [&](uint32_t base_pt) {
         __builtin_amdgcn_wave_barrier();
         const uint32_t p = lane & (DEC_TILE - 1);
-        const uint32_t r0 = lane >> 3;
-        for (uint32_t j = 0; j < WAVE / DEC_TILE; j++) {
-            uint32_t r = r0 + j * (WAVE / DEC_TILE);
-            uint32_t c = (uint32_t)__shfl((int)cnt, (int)r);
-            uint64_t row = (uint64_t)__shfl((int)series, (int)r);
-            uint32_t pt = base_pt + p;
-            if (pt < c) {
+        const uint32_t r0 = lane / DEC_TILE;
+        const uint32_t pt = base_pt + p;
+        if (__all(cnt >= base_pt + DEC_TILE)) {
+            /* all 64 rows full: unconditional stores, no per-row counts */
+            for (uint32_t j = 0; j < DEC_TILE; j++) {
+                uint32_t r = r0 + j * (WAVE / DEC_TILE);
+                uint64_t row = (uint64_t)__shfl((int)series, (int)r);
                 out_ts[row * stride + pt] = ts_tile[r][p];
                 out_vals[row * stride + pt] = val_tile[r][p];
+            }
+        } else {
+            for (uint32_t j = 0; j < DEC_TILE; j++) {
+                uint32_t r = r0 + j * (WAVE / DEC_TILE);
+                uint32_t c = (uint32_t)__shfl((int)cnt, (int)r);
+                uint64_t row = (uint64_t)__shfl((int)series, (int)r);
+                if (pt < c) {
+                    out_ts[row * stride + pt] = ts_tile[r][p];
+                    out_vals[row * stride + pt] = val_tile[r][p];
+                }
             }
         }
         __builtin_amdgcn_wave_barrier();
