@@ -92,22 +92,23 @@ __device__ __forceinline__ double go_modf(double v, double* ip) {
  * contract: stream starts 8B-aligned, buffer zero-padded to 8B. */
 
 struct BitReader {
-    /* Register bit window with a cheap common path (bits served from `cur`)
-     * and a two-deep word pipeline: `nxt` is the word the reference reader
-     * would fetch next (so EOF semantics match istream.go:73-115 over
-     * reader64.go:40-80 exactly), `pf` is prefetched a further word ahead —
-     * its load is issued ~128 bits before first use, hiding refill latency
-     * under parsing. The blob's zero padding reproduces reader64's
-     * zero-filled partial tail word. */
+    /* Branchless 128-bit register window (hi:lo, left-aligned `loaded` real
+     * bits) over the byte stream, with a two-word software prefetch pipeline:
+     * pw is the word about to be inserted, pf the word loaded ~128 bits
+     * before first use, so refill latency overlaps parsing. Byte-stream
+     * semantics identical to istream.go:73-115 over reader64.go:40-80
+     * (EOF iff the request exceeds the stream's remaining bits; the blob's
+     * 8B zero padding reproduces the zero-filled partial tail word).
+     * Invariant: after refill, loaded > 64 unless the stream is exhausted,
+     * so peek/read of n <= 64 bits always comes straight from hi. */
     const uint64_t* words;
     int64_t len;        /* true byte length */
     int64_t index;      /* byte index of the next word to prefetch */
-    uint64_t cur;       /* left-aligned buffered bits */
-    uint32_t rem;       /* valid bits in cur */
-    uint64_t nxt;       /* next word (left-aligned) */
-    uint32_t nxt_bits;
-    uint64_t pf;        /* prefetched word after nxt */
-    uint32_t pf_bits;
+    uint64_t hi, lo;    /* window */
+    uint32_t loaded;    /* real bits in window */
+    int64_t bits_left;  /* real stream bits not yet consumed */
+    uint64_t pw, pf;    /* pipeline words (left-aligned) */
+    uint32_t pw_bits, pf_bits;
 
     __device__ __forceinline__ void prefetch() {
         if (index < len) {
@@ -120,56 +121,53 @@ struct BitReader {
             pf_bits = 0;
         }
     }
+    __device__ __forceinline__ void refill() {
+        if (pw_bits) {
+            if (loaded == 0) {
+                hi |= pw;
+            } else if (loaded < 64) {
+                hi |= pw >> loaded;
+                lo |= pw << (64 - loaded);
+            } else {
+                lo |= pw;
+            }
+            loaded += pw_bits;
+        }
+        pw = pf;
+        pw_bits = pf_bits;
+        prefetch();
+    }
     __device__ void init(const uint8_t* base, uint64_t off, uint32_t l) {
         words = (const uint64_t*)(base + off);
         len = l;
         index = 0;
-        cur = 0;
-        rem = 0;
-        prefetch();
-        nxt = pf;
-        nxt_bits = pf_bits;
-        prefetch();
+        hi = 0; lo = 0;
+        loaded = 0;
+        bits_left = (int64_t)l * 8;
+        pw = 0; pw_bits = 0;
+        prefetch();          /* word 0 -> pf */
+        refill();            /* pf -> pw, word 1 -> pf */
+        refill();            /* insert word 0, word 2 -> pf */
+        refill();            /* insert word 1: loaded > 64 (if available) */
+    }
+    __device__ __forceinline__ void advance(uint32_t n) {
+        hi = (n >= 64) ? lo : ((hi << n) | (n ? (lo >> (64 - n)) : 0));
+        lo = (n >= 64) ? 0 : (lo << n);
+        loaded -= n;
+        bits_left -= n;
+        if (loaded <= 64) refill();
     }
     __device__ __forceinline__ int read_bits(uint32_t n, uint64_t* out) {
-        if (n <= rem) {
-            *out = n ? (cur >> (64 - n)) : 0;
-            cur = (n >= 64) ? 0 : (cur << n);
-            rem -= n;
-            return 0;
-        }
-        uint32_t need = n - rem;
-        if (nxt_bits < need) return M3GPU_SERIES_EOF;
-        uint64_t res = rem ? (cur >> (64 - rem)) : 0;
-        res = ((need >= 64) ? 0 : (res << need)) | (nxt >> (64 - need));
-        cur = (need >= 64) ? 0 : (nxt << need);
-        rem = nxt_bits - need;
-        nxt = pf;
-        nxt_bits = pf_bits;
-        prefetch();
-        *out = res;
+        if ((int64_t)n > bits_left) return M3GPU_SERIES_EOF;
+        *out = n ? (hi >> (64 - n)) : 0;
+        advance(n);
         return 0;
     }
     /* consume n bits already validated by a successful peek_bits(n) */
-    __device__ __forceinline__ void consume(uint32_t n) {
-        if (n <= rem) {
-            cur = (n >= 64) ? 0 : (cur << n);
-            rem -= n;
-            return;
-        }
-        uint32_t need = n - rem;
-        cur = (need >= 64) ? 0 : (nxt << need);
-        rem = nxt_bits - need;
-        nxt = pf;
-        nxt_bits = pf_bits;
-        prefetch();
-    }
+    __device__ __forceinline__ void consume(uint32_t n) { advance(n); }
     __device__ __forceinline__ int peek_bits(uint32_t n, uint64_t* out) {
-        if (n <= rem) { *out = n ? (cur >> (64 - n)) : 0; return 0; }
-        uint32_t need = n - rem;
-        if (nxt_bits < need) return M3GPU_SERIES_EOF;
-        uint64_t res = rem ? (cur >> (64 - rem)) : 0;
-        *out = ((need >= 64) ? 0 : (res << need)) | (nxt >> (64 - need));
+        if ((int64_t)n > bits_left) return M3GPU_SERIES_EOF;
+        *out = n ? (hi >> (64 - n)) : 0;
         return 0;
     }
 };
@@ -620,7 +618,7 @@ struct Decoder {
  *   flush step j: lane l stores row (l>>3)+8j, point (l&7) — 8 consecutive
  *   8B addresses per row = full 64B line utilization for ts[] and val[]. */
 
-#define DEC_TILE 4
+#define DEC_TILE 8
 
 __global__ void __launch_bounds__(BLOCK_THREADS)
 k_decode_batch(const uint8_t* __restrict__ blobs,
