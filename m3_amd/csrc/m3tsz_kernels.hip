@@ -137,7 +137,7 @@ struct BitReader {
         pw_bits = pf_bits;
         prefetch();
     }
-    __device__ void init(const uint8_t* base, uint64_t off, uint32_t l) {
+    __device__ __forceinline__ void init(const uint8_t* base, uint64_t off, uint32_t l) {
         words = (const uint64_t*)(base + off);
         len = l;
         index = 0;
@@ -186,7 +186,7 @@ struct Decoder {
     bool have_scheme, tu_changed, done, is_float;
     bool int_optimized;
 
-    __device__ void init(const uint8_t* base, uint64_t off, uint32_t len,
+    __device__ __forceinline__ void init(const uint8_t* base, uint64_t off, uint32_t len,
                          bool intopt, uint8_t dunit) {
         r.init(base, off, len);
         prev_time = 0; prev_time_delta = 0;
@@ -198,7 +198,7 @@ struct Decoder {
     }
 
     /* timestamp_iterator.go:115-135 */
-    __device__ int read_time_unit() {
+    __device__ __forceinline__ int read_time_unit() {
         uint64_t tu_bits;
         int err = r.read_bits(8, &tu_bits);
         if (err) return err;
@@ -212,7 +212,7 @@ struct Decoder {
     }
 
     /* binary.ReadVarint via ReadByte */
-    __device__ int read_varint(int64_t* out) {
+    __device__ __forceinline__ int read_varint(int64_t* out) {
         uint64_t ux = 0;
         int shift = 0;
         for (int i = 0; i < 10; i++) {
@@ -233,7 +233,7 @@ struct Decoder {
 
     /* timestamp_iterator.go:327-356 (annotation bytes are skipped — the bulk
      * decode surface does not return annotations) */
-    __device__ int skip_annotation() {
+    __device__ __forceinline__ int skip_annotation() {
         int64_t alen;
         int err = read_varint(&alen);
         if (err) return err;
@@ -248,7 +248,7 @@ struct Decoder {
     }
 
     /* timestamp_iterator.go:307-325 */
-    __device__ int read_full_timestamp(int64_t* dod) {
+    __device__ __forceinline__ int read_full_timestamp(int64_t* dod) {
         if (!scheme_default_bits(time_unit)) return M3GPU_SERIES_NO_SCHEME;
         have_scheme = true;
         scheme_unit = time_unit;
@@ -260,7 +260,7 @@ struct Decoder {
     }
 
     /* timestamp_iterator.go:250-305 */
-    __device__ int read_dod(int64_t* out) {
+    __device__ __forceinline__ int read_dod(int64_t* out) {
         if (tu_changed) return read_full_timestamp(out);
         if (!have_scheme) return M3GPU_SERIES_NO_SCHEME;
         uint64_t cb;
@@ -299,7 +299,7 @@ struct Decoder {
      * of the grammar). Bit-identical to the sequential reference reads;
      * falls back to the bit-by-bit path near end-of-stream (short peek)
      * and for unit-change/no-scheme states. */
-    __device__ int read_marker_or_dod(int64_t* out) {
+    __device__ __forceinline__ int read_marker_or_dod(int64_t* out) {
         /* dominant path: a regular cadence emits dod == 0 = a single 0 bit,
          * which can never be a marker (markers start with 1) — classify it
          * from a 1-bit peek before the full 11-bit marker peek. */
@@ -363,7 +363,7 @@ struct Decoder {
     }
 
     /* timestamp_iterator.go:137-161 + initialTimeUnit */
-    __device__ int read_first_timestamp() {
+    __device__ __forceinline__ int read_first_timestamp() {
         uint64_t nt_bits;
         int err = r.read_bits(64, &nt_bits);
         if (err) return err;
@@ -382,7 +382,7 @@ struct Decoder {
     }
 
     /* timestamp_iterator.go:80-113 */
-    __device__ int read_timestamp(bool* first) {
+    __device__ __forceinline__ int read_timestamp(bool* first) {
         *first = false;
         int err;
         if (prev_time != 0) {
@@ -402,7 +402,7 @@ struct Decoder {
     }
 
     /* float_encoder_iterator.go:105-165 */
-    __device__ int read_full_float() {
+    __device__ __forceinline__ int read_full_float() {
         uint64_t vb;
         int err = r.read_bits(64, &vb);
         if (err) return err;
@@ -413,7 +413,7 @@ struct Decoder {
     /* Fused XOR-field read: one 64-bit peek classifies the control bits
      * and extracts contained payloads; falls back to the stepwise reads
      * near end-of-stream (identical error semantics). */
-    __device__ int read_next_float_fused() {
+    __device__ __forceinline__ int read_next_float_fused() {
         uint64_t w;
         if (r.peek_bits(64, &w) != 0) return read_next_float();
         if ((w >> 63) == 0) { r.consume(1); prev_xor = 0; return 0; }
@@ -448,7 +448,7 @@ struct Decoder {
         prev_float_bits ^= prev_xor;
         return 0;
     }
-    __device__ int read_next_float() {
+    __device__ __forceinline__ int read_next_float() {
         uint64_t cb;
         int err = r.read_bits(1, &cb);
         if (err) return err;
@@ -483,7 +483,7 @@ struct Decoder {
     }
 
     /* iterator.go:178-219 */
-    __device__ int read_int_sig_mult() {
+    __device__ __forceinline__ int read_int_sig_mult() {
         uint64_t b;
         int err = r.read_bits(1, &b);
         if (err) return err;
@@ -509,7 +509,7 @@ struct Decoder {
         }
         return 0;
     }
-    __device__ int read_int_val_diff() {
+    __device__ __forceinline__ int read_int_val_diff() {
         if (sig == 64) { /* readIntValDiffSlow */
             uint64_t sb;
             int err = r.read_bits(1, &sb);
@@ -531,7 +531,7 @@ struct Decoder {
     }
 
     /* iterator.go:108-176 */
-    __device__ int read_first_value() {
+    __device__ __forceinline__ int read_first_value() {
         if (!int_optimized) return read_full_float();
         uint64_t b;
         int err = r.read_bits(1, &b);
@@ -546,7 +546,7 @@ struct Decoder {
         if (err) return err;
         return read_int_val_diff();
     }
-    __device__ int read_next_value() {
+    __device__ __forceinline__ int read_next_value() {
         if (!int_optimized) return read_next_float_fused();
         /* fused common case: '1' + sign + sig-bit diff from one peek */
         uint64_t w;
@@ -589,7 +589,7 @@ struct Decoder {
     }
 
     /* One point. Returns 1 = value in (*t,*v), 0 = done, -err on error. */
-    __device__ int next(int64_t* t, double* v) {
+    __device__ __forceinline__ int next(int64_t* t, double* v) {
         if (done) return 0;
         bool first;
         int err = read_timestamp(&first);
@@ -724,7 +724,7 @@ struct BitWriter {
     uint32_t cap_words;
     int err;
 
-    __device__ void init(uint8_t* row, uint32_t cap_bytes) {
+    __device__ __forceinline__ void init(uint8_t* row, uint32_t cap_bytes) {
         acc = 0; used = 0; nwords = 0;
         out = (uint64_t*)row;
         cap_words = cap_bytes / 8;
@@ -752,7 +752,7 @@ struct BitWriter {
     }
     __device__ __forceinline__ void write_bit(uint32_t b) { write_bits(b, 1); }
     /* Flush the partial tail word. Returns total byte length. */
-    __device__ uint32_t finish() {
+    __device__ __forceinline__ uint32_t finish() {
         uint32_t nbytes = nwords * 8;
         if (used > 0) {
             if (nwords >= cap_words) { err = M3GPU_SERIES_CAPACITY; return 0; }
@@ -764,7 +764,7 @@ struct BitWriter {
 };
 
 /* m3tsz.go:78-119 convertToIntFloat — bit-sensitive: -ffp-contract=off */
-__device__ int convert_to_int_float(double v, uint8_t cur_max_mult,
+__device__ __forceinline__ int convert_to_int_float(double v, uint8_t cur_max_mult,
                                     double* out_val, uint8_t* out_mult, bool* out_is_float) {
     const double mult_tab[7] = {1.0, 10.0, 100.0, 1000.0, 10000.0, 100000.0, 1000000.0};
     const double MAXINT = 9223372036854775808.0;
@@ -805,7 +805,7 @@ struct Encoder {
     bool has_written_first, is_float, int_optimized;
     uint32_t num_encoded;
 
-    __device__ void init(uint8_t* row, uint32_t cap_bytes,
+    __device__ __forceinline__ void init(uint8_t* row, uint32_t cap_bytes,
                          int64_t start_ns, bool intopt, uint8_t default_unit) {
         w.init(row, cap_bytes);
         prev_time = start_ns;
@@ -822,13 +822,13 @@ struct Encoder {
         num_encoded = 0;
     }
 
-    __device__ void write_marker(uint32_t marker) {
+    __device__ __forceinline__ void write_marker(uint32_t marker) {
         w.write_bits(MARKER_OPCODE, 9);
         w.write_bits(marker, 2);
     }
 
     /* timestamp_encoder.go:205-246 */
-    __device__ int write_dod_unchanged(int64_t prev_delta, int64_t cur_delta, uint8_t unit) {
+    __device__ __forceinline__ int write_dod_unchanged(int64_t prev_delta, int64_t cur_delta, uint8_t unit) {
         if (!unit_valid(unit)) return M3GPU_SERIES_NO_SCHEME;
         int64_t u = UNIT_NS_D[unit];
         int64_t dod = (cur_delta - prev_delta) / u;
@@ -856,7 +856,7 @@ struct Encoder {
     }
 
     /* timestamp_encoder.go:72-129 (fixed unit, no annotations) */
-    __device__ int write_time(int64_t cur_time, uint8_t unit) {
+    __device__ __forceinline__ int write_time(int64_t cur_time, uint8_t unit) {
         if (!has_written_first) {
             w.write_bits((uint64_t)prev_time, 64);
             has_written_first = true;
@@ -881,12 +881,12 @@ struct Encoder {
     }
 
     /* float_encoder_iterator.go:69-103 */
-    __device__ void write_full_float(uint64_t val) {
+    __device__ __forceinline__ void write_full_float(uint64_t val) {
         prev_float_bits = val;
         prev_xor = val;
         w.write_bits(val, 64);
     }
-    __device__ void write_xor(uint64_t cur_xor) {
+    __device__ __forceinline__ void write_xor(uint64_t cur_xor) {
         if (cur_xor == 0) { w.write_bits(0, 1); return; }
         uint32_t pl = prev_xor ? __builtin_clzll(prev_xor) : 64;
         uint32_t pt = prev_xor ? __builtin_ctzll(prev_xor) : 0;
@@ -903,7 +903,7 @@ struct Encoder {
         w.write_bits(nmean - 1, 6);
         w.write_bits(cur_xor >> ct, nmean);
     }
-    __device__ void write_next_float(uint64_t val) {
+    __device__ __forceinline__ void write_next_float(uint64_t val) {
         uint64_t x = prev_float_bits ^ val;
         write_xor(x);
         prev_xor = x;
@@ -911,11 +911,11 @@ struct Encoder {
     }
 
     /* int_sig_bits_tracker.go:35-91 */
-    __device__ void tracker_write_int_val_diff(uint64_t val_bits, bool neg) {
+    __device__ __forceinline__ void tracker_write_int_val_diff(uint64_t val_bits, bool neg) {
         w.write_bit(neg ? 1 : 0);
         w.write_bits(val_bits, num_sig_state);
     }
-    __device__ void tracker_write_int_sig(uint8_t s) {
+    __device__ __forceinline__ void tracker_write_int_sig(uint8_t s) {
         if (num_sig_state != s) {
             w.write_bit(1);
             if (s == 0) w.write_bit(0);
@@ -925,7 +925,7 @@ struct Encoder {
         }
         num_sig_state = s;
     }
-    __device__ uint8_t tracker_track_new_sig(uint8_t nsig) {
+    __device__ __forceinline__ uint8_t tracker_track_new_sig(uint8_t nsig) {
         uint8_t new_sig = num_sig_state;
         if (nsig > num_sig_state) {
             new_sig = nsig;
@@ -944,7 +944,7 @@ struct Encoder {
     }
 
     /* encoder.go:233-250 */
-    __device__ void write_int_sig_mult(uint8_t s, uint8_t m, bool float_changed) {
+    __device__ __forceinline__ void write_int_sig_mult(uint8_t s, uint8_t m, bool float_changed) {
         tracker_write_int_sig(s);
         if (m > max_mult) {
             w.write_bit(1);
@@ -959,7 +959,7 @@ struct Encoder {
     }
 
     /* encoder.go:112-146 */
-    __device__ int write_first_value(double v) {
+    __device__ __forceinline__ int write_first_value(double v) {
         if (!int_optimized) { write_full_float(f2bits(v)); return 0; }
         double val; uint8_t m; bool isf;
         int err = convert_to_int_float(v, 0, &val, &m, &isf);
@@ -983,7 +983,7 @@ struct Encoder {
     }
 
     /* encoder.go:174-231 */
-    __device__ void write_float_val(uint64_t val, uint8_t m) {
+    __device__ __forceinline__ void write_float_val(uint64_t val, uint8_t m) {
         if (!is_float) {
             w.write_bit(0); w.write_bit(0); w.write_bit(1);
             write_full_float(val);
@@ -995,7 +995,7 @@ struct Encoder {
         w.write_bit(1);
         write_next_float(val);
     }
-    __device__ void write_int_val(double val, uint8_t m, bool isf, double val_diff) {
+    __device__ __forceinline__ void write_int_val(double val, uint8_t m, bool isf, double val_diff) {
         if (val_diff == 0 && isf == is_float && m == max_mult) {
             w.write_bit(0); w.write_bit(1);
             return;
@@ -1019,7 +1019,7 @@ struct Encoder {
     }
 
     /* encoder.go:148-172 */
-    __device__ int write_next_value(double v) {
+    __device__ __forceinline__ int write_next_value(double v) {
         if (!int_optimized) { write_next_float(f2bits(v)); return 0; }
         double val; uint8_t m; bool isf;
         int err = convert_to_int_float(v, max_mult, &val, &m, &isf);
@@ -1034,7 +1034,7 @@ struct Encoder {
         return 0;
     }
 
-    __device__ int encode(int64_t t, double v, uint8_t unit) {
+    __device__ __forceinline__ int encode(int64_t t, double v, uint8_t unit) {
         int err = write_time(t, unit);
         if (err) return err;
         err = num_encoded == 0 ? write_first_value(v) : write_next_value(v);
@@ -1044,7 +1044,7 @@ struct Encoder {
 
     /* Finalize: appending the EOS marker to the live bitstream produces
      * exactly head[:len-1] + Tail(lastByte, pos) (scheme.go:198-212). */
-    __device__ uint32_t finalize() {
+    __device__ __forceinline__ uint32_t finalize() {
         if (num_encoded == 0 && w.nwords == 0 && w.used == 0) return 0;
         write_marker(MARKER_EOS);
         return w.finish();
@@ -1116,7 +1116,7 @@ struct BucketState {
     int64_t last_at;
     int64_t count;
 
-    __device__ void reset() {
+    __device__ __forceinline__ void reset() {
         isum = 0; isumsq = 0;
         imin = INT64_MAX; imax = INT64_MIN;   /* counter.go:44-47 */
         fsum = 0; fsumsq = 0;
@@ -1127,7 +1127,7 @@ struct BucketState {
     }
 };
 
-__device__ double m3_stdev(int64_t count, double sum_sq, double sum) { /* common.go:29-36 */
+__device__ __forceinline__ double m3_stdev(int64_t count, double sum_sq, double sum) { /* common.go:29-36 */
     int64_t div = count * (count - 1);
     if (div == 0) return 0.0;
     return sqrt(((double)count * sum_sq - sum * sum) / (double)div);
