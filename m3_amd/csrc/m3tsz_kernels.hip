@@ -92,23 +92,22 @@ __device__ __forceinline__ double go_modf(double v, double* ip) {
  * contract: stream starts 8B-aligned, buffer zero-padded to 8B. */
 
 struct BitReader {
-    /* Branchless 128-bit register window (hi:lo, left-aligned `loaded` real
-     * bits) over the byte stream, with a two-word software prefetch pipeline:
-     * pw is the word about to be inserted, pf the word loaded ~128 bits
-     * before first use, so refill latency overlaps parsing. Byte-stream
-     * semantics identical to istream.go:73-115 over reader64.go:40-80
-     * (EOF iff the request exceeds the stream's remaining bits; the blob's
-     * 8B zero padding reproduces the zero-filled partial tail word).
-     * Invariant: after refill, loaded > 64 unless the stream is exhausted,
-     * so peek/read of n <= 64 bits always comes straight from hi. */
+    /* Register bit window with a cheap common path (bits served from `cur`)
+     * and a two-deep word pipeline: `nxt` is the word the reference reader
+     * would fetch next (so EOF semantics match istream.go:73-115 over
+     * reader64.go:40-80 exactly), `pf` is prefetched a further word ahead —
+     * its load is issued ~128 bits before first use, hiding refill latency
+     * under parsing. The blob's zero padding reproduces reader64's
+     * zero-filled partial tail word. */
     const uint64_t* words;
     int64_t len;        /* true byte length */
     int64_t index;      /* byte index of the next word to prefetch */
-    uint64_t hi, lo;    /* window */
-    uint32_t loaded;    /* real bits in window */
-    int64_t bits_left;  /* real stream bits not yet consumed */
-    uint64_t pw, pf;    /* pipeline words (left-aligned) */
-    uint32_t pw_bits, pf_bits;
+    uint64_t cur;       /* left-aligned buffered bits */
+    uint32_t rem;       /* valid bits in cur */
+    uint64_t nxt;       /* next word (left-aligned) */
+    uint32_t nxt_bits;
+    uint64_t pf;        /* prefetched word after nxt */
+    uint32_t pf_bits;
 
     __device__ __forceinline__ void prefetch() {
         if (index < len) {
@@ -121,53 +120,56 @@ struct BitReader {
             pf_bits = 0;
         }
     }
-    __device__ __forceinline__ void refill() {
-        if (pw_bits) {
-            if (loaded == 0) {
-                hi |= pw;
-            } else if (loaded < 64) {
-                hi |= pw >> loaded;
-                lo |= pw << (64 - loaded);
-            } else {
-                lo |= pw;
-            }
-            loaded += pw_bits;
-        }
-        pw = pf;
-        pw_bits = pf_bits;
-        prefetch();
-    }
     __device__ __forceinline__ void init(const uint8_t* base, uint64_t off, uint32_t l) {
         words = (const uint64_t*)(base + off);
         len = l;
         index = 0;
-        hi = 0; lo = 0;
-        loaded = 0;
-        bits_left = (int64_t)l * 8;
-        pw = 0; pw_bits = 0;
-        prefetch();          /* word 0 -> pf */
-        refill();            /* pf -> pw, word 1 -> pf */
-        refill();            /* insert word 0, word 2 -> pf */
-        refill();            /* insert word 1: loaded > 64 (if available) */
-    }
-    __device__ __forceinline__ void advance(uint32_t n) {
-        hi = (n >= 64) ? lo : ((hi << n) | (n ? (lo >> (64 - n)) : 0));
-        lo = (n >= 64) ? 0 : (lo << n);
-        loaded -= n;
-        bits_left -= n;
-        if (loaded <= 64) refill();
+        cur = 0;
+        rem = 0;
+        prefetch();
+        nxt = pf;
+        nxt_bits = pf_bits;
+        prefetch();
     }
     __device__ __forceinline__ int read_bits(uint32_t n, uint64_t* out) {
-        if ((int64_t)n > bits_left) return M3GPU_SERIES_EOF;
-        *out = n ? (hi >> (64 - n)) : 0;
-        advance(n);
+        if (n <= rem) {
+            *out = n ? (cur >> (64 - n)) : 0;
+            cur = (n >= 64) ? 0 : (cur << n);
+            rem -= n;
+            return 0;
+        }
+        uint32_t need = n - rem;
+        if (nxt_bits < need) return M3GPU_SERIES_EOF;
+        uint64_t res = rem ? (cur >> (64 - rem)) : 0;
+        res = ((need >= 64) ? 0 : (res << need)) | (nxt >> (64 - need));
+        cur = (need >= 64) ? 0 : (nxt << need);
+        rem = nxt_bits - need;
+        nxt = pf;
+        nxt_bits = pf_bits;
+        prefetch();
+        *out = res;
         return 0;
     }
     /* consume n bits already validated by a successful peek_bits(n) */
-    __device__ __forceinline__ void consume(uint32_t n) { advance(n); }
+    __device__ __forceinline__ void consume(uint32_t n) {
+        if (n <= rem) {
+            cur = (n >= 64) ? 0 : (cur << n);
+            rem -= n;
+            return;
+        }
+        uint32_t need = n - rem;
+        cur = (need >= 64) ? 0 : (nxt << need);
+        rem = nxt_bits - need;
+        nxt = pf;
+        nxt_bits = pf_bits;
+        prefetch();
+    }
     __device__ __forceinline__ int peek_bits(uint32_t n, uint64_t* out) {
-        if ((int64_t)n > bits_left) return M3GPU_SERIES_EOF;
-        *out = n ? (hi >> (64 - n)) : 0;
+        if (n <= rem) { *out = n ? (cur >> (64 - n)) : 0; return 0; }
+        uint32_t need = n - rem;
+        if (nxt_bits < need) return M3GPU_SERIES_EOF;
+        uint64_t res = rem ? (cur >> (64 - rem)) : 0;
+        *out = ((need >= 64) ? 0 : (res << need)) | (nxt >> (64 - need));
         return 0;
     }
 };
@@ -300,9 +302,8 @@ struct Decoder {
      * falls back to the bit-by-bit path near end-of-stream (short peek)
      * and for unit-change/no-scheme states. */
     __device__ __forceinline__ int read_marker_or_dod(int64_t* out) {
-        /* dominant path: a regular cadence emits dod == 0 = a single 0 bit,
-         * which can never be a marker (markers start with 1) — classify it
-         * from a 1-bit peek before the full 11-bit marker peek. */
+        /* dominant path: regular cadence emits dod == 0 = a single 0 bit,
+         * which can never be a marker (markers start with 1). */
         if (!tu_changed && have_scheme) {
             uint64_t b1;
             if (r.peek_bits(1, &b1) == 0 && b1 == 0) {
@@ -410,44 +411,6 @@ struct Decoder {
         prev_xor = vb;
         return 0;
     }
-    /* Fused XOR-field read: one 64-bit peek classifies the control bits
-     * and extracts contained payloads; falls back to the stepwise reads
-     * near end-of-stream (identical error semantics). */
-    __device__ __forceinline__ int read_next_float_fused() {
-        uint64_t w;
-        if (r.peek_bits(64, &w) != 0) return read_next_float();
-        if ((w >> 63) == 0) { r.consume(1); prev_xor = 0; return 0; }
-        if ((w >> 62) == 0x2) { /* contained */
-            uint32_t lead = prev_xor ? __builtin_clzll(prev_xor) : 64;
-            uint32_t trail = prev_xor ? __builtin_ctzll(prev_xor) : 0;
-            uint32_t nmean = 64 - lead - trail;
-            if (nmean <= 62) {
-                uint64_t mb = nmean ? ((w << 2) >> (64 - nmean)) : 0;
-                r.consume(2 + nmean);
-                prev_xor = mb << trail;
-                prev_float_bits ^= prev_xor;
-                return 0;
-            }
-            r.consume(2);
-            uint64_t mb;
-            int err = r.read_bits(nmean, &mb);
-            if (err) return err;
-            prev_xor = mb << trail;
-            prev_float_bits ^= prev_xor;
-            return 0;
-        }
-        /* uncontained: 11 + 6b lead + 6b (nmean-1) + payload */
-        uint64_t lead = (w >> 56) & 0x3f;
-        uint64_t nmean = ((w >> 50) & 0x3f) + 1;
-        r.consume(14);
-        uint64_t mb;
-        int err = r.read_bits((uint32_t)nmean, &mb);
-        if (err) return err;
-        uint64_t trail = 64 - lead - nmean;
-        prev_xor = mb << trail;
-        prev_float_bits ^= prev_xor;
-        return 0;
-    }
     __device__ __forceinline__ int read_next_float() {
         uint64_t cb;
         int err = r.read_bits(1, &cb);
@@ -547,21 +510,7 @@ struct Decoder {
         return read_int_val_diff();
     }
     __device__ __forceinline__ int read_next_value() {
-        if (!int_optimized) return read_next_float_fused();
-        /* fused common case: '1' + sign + sig-bit diff from one peek */
-        uint64_t w;
-        if (!is_float && sig <= 62 && r.peek_bits(64, &w) == 0) {
-            if (w >> 63) { /* opcodeNoUpdate -> int diff */
-                uint64_t bits = (w << 1) >> (63 - sig); /* sign + payload */
-                r.consume(2 + sig);
-                double sgn = -1.0;
-                if ((bits >> sig) == 1) { sgn = 1.0; bits ^= (1ULL << sig); }
-                int_val += sgn * (double)bits;
-                return 0;
-            }
-            if ((w >> 62) == 0x1) { r.consume(2); return 0; } /* repeat */
-            /* '00': mode/sig/mult update (rare) — stepwise below */
-        }
+        if (!int_optimized) return read_next_float();
         uint64_t b;
         int err = r.read_bits(1, &b);
         if (err) return err;
@@ -584,7 +533,7 @@ struct Decoder {
             is_float = false;
             return 0;
         }
-        if (is_float) return read_next_float_fused();
+        if (is_float) return read_next_float();
         return read_int_val_diff();
     }
 
