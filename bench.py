@@ -55,6 +55,10 @@ def parse_args():
     p.add_argument("--skip-cpu-baseline", action="store_true")
     p.add_argument("--no-sort", action="store_true",
                    help="disable length-sorted wave scheduling")
+    p.add_argument("--mode", choices=["decode", "encode", "rollup"],
+                   default="decode",
+                   help="decode = headline metric (BASELINE configs[1]); "
+                        "encode/rollup = BASELINE configs[2]/[3]")
     return p.parse_args()
 
 
@@ -89,6 +93,13 @@ def main():
               f"in {time.time()-t0:.1f}s", flush=True)
 
     npts_total_local = n_local * args.npts
+    if args.mode == "encode":
+        bench_encode(args, torch, engine, workload, device, n_local, lo)
+        return
+    if args.mode == "rollup":
+        bench_rollup(args, torch, engine, d_blob, d_offsets, d_lens,
+                     enc_bytes, device, n_local)
+        return
     out_ts = torch.empty((n_local, args.npts), dtype=torch.int64, device=device)
     out_vals = torch.empty((n_local, args.npts), dtype=torch.float64, device=device)
     out_counts = torch.empty(n_local, dtype=torch.int32, device=device)
@@ -238,3 +249,104 @@ def run_cpu_baseline(args, d_blob, d_offsets, d_lens, n_local):
 
 if __name__ == "__main__":
     main()
+
+
+def bench_encode(args, torch, engine, workload, device, n_local, lo):
+    """BASELINE configs[2]: 1M-series batched M3TSZ encode, 1x MI355X.
+    Input: decoded SoA resident in HBM; step = encode into strided rows."""
+    import numpy as np
+    npts = args.npts
+    d_ts = torch.empty((n_local, npts), dtype=torch.int64, device=device)
+    d_vals = torch.empty((n_local, npts), dtype=torch.float64, device=device)
+    chunk = args.chunk
+    for s0 in range(0, n_local, chunk):
+        n = min(chunk, n_local - s0)
+        ts, vals = workload.gen_chunk(lo + s0, n, npts)
+        d_ts[s0:s0 + n] = torch.from_numpy(ts).to(device)
+        d_vals[s0:s0 + n] = torch.from_numpy(vals).to(device)
+    d_counts = torch.full((n_local,), npts, dtype=torch.int32, device=device)
+    out_stride = (24 * npts + 32 + 7) & ~7
+    d_out = torch.zeros((n_local, out_stride), dtype=torch.uint8, device=device)
+    d_lens = torch.empty(n_local, dtype=torch.int32, device=device)
+    d_errs = torch.empty(n_local, dtype=torch.int32, device=device)
+
+    def step():
+        engine.encode_batch_dev(d_ts, d_vals, d_counts, d_out, d_lens, d_errs)
+
+    step()
+    torch.cuda.synchronize()
+    assert int(d_errs.abs().sum().item()) == 0
+    enc_bytes = int(d_lens.to(torch.int64).sum().item())
+    for _ in range(args.warmup):
+        step()
+    torch.cuda.synchronize()
+    t0 = time.time()
+    for _ in range(args.steps):
+        step()
+    torch.cuda.synchronize()
+    dt = time.time() - t0
+    npts_total = n_local * npts
+    dps = npts_total * args.steps / dt
+    algo = enc_bytes + 16 * npts_total
+    print(json.dumps({
+        "metric": "datapoints/sec M3TSZ encode (1M series x 1440 pts)",
+        "value": dps, "unit": "datapoints/sec", "n_gpus": 1,
+        "steps": args.steps, "warmup": args.warmup,
+        "ms_per_step": dt / args.steps * 1e3, "higher_is_better": True,
+        "scaling": "weak", "vs_baseline": None, "dtype": "f64",
+        "data": "synthetic",
+        "config": {"workload": "1M series batched M3TSZ encode, 1x MI355X",
+                   "nseries": n_local, "npts": npts,
+                   "encoded_bytes_per_pt": enc_bytes / npts_total},
+        "roofline": {"bound": "hbm", "achieved": algo / (dt / args.steps) / 1e9,
+                     "peak": HBM_PEAK_GBS, "unit": "GB/s",
+                     "frac": algo / (dt / args.steps) / 1e9 / HBM_PEAK_GBS,
+                     "traffic": None},
+    }), flush=True)
+
+
+def bench_rollup(args, torch, engine, d_blob, d_offsets, d_lens, enc_bytes,
+                 device, n_local):
+    """BASELINE configs[3]: fused decode->downsample, 10s -> 1m
+    sum/min/max/p99 with m3aggregator semantics."""
+    window = 60 * 10**9
+    nbuckets = args.npts * 10 // 60
+    aggs = ["sum", "min", "max", "p99"]
+    out = torch.empty((n_local, nbuckets, len(aggs)), dtype=torch.float64,
+                      device=device)
+    wts = torch.empty((n_local, nbuckets), dtype=torch.int64, device=device)
+    errs = torch.empty(n_local, dtype=torch.int32, device=device)
+
+    def step():
+        engine.rollup_batch_dev(d_blob, d_offsets, d_lens, engine.METRIC_TIMER,
+                                window, nbuckets, aggs, out, wts, errs)
+
+    step()
+    torch.cuda.synchronize()
+    assert int(errs.abs().sum().item()) == 0
+    for _ in range(args.warmup):
+        step()
+    torch.cuda.synchronize()
+    t0 = time.time()
+    for _ in range(args.steps):
+        step()
+    torch.cuda.synchronize()
+    dt = time.time() - t0
+    npts_total = n_local * args.npts
+    dps = npts_total * args.steps / dt
+    algo = enc_bytes + n_local * nbuckets * (len(aggs) * 8 + 8)
+    print(json.dumps({
+        "metric": "datapoints/sec fused M3TSZ decode->1m rollup (sum/min/max/p99)",
+        "value": dps, "unit": "datapoints/sec", "n_gpus": 1,
+        "steps": args.steps, "warmup": args.warmup,
+        "ms_per_step": dt / args.steps * 1e3, "higher_is_better": True,
+        "scaling": "weak", "vs_baseline": None, "dtype": "f64",
+        "data": "synthetic",
+        "config": {"workload": "Fused decode->downsample 10s->1m rollup, 1x MI355X",
+                   "nseries": n_local, "npts": args.npts, "nbuckets": nbuckets,
+                   "aggs": aggs},
+        "roofline": {"bound": "hbm", "achieved": algo / (dt / args.steps) / 1e9,
+                     "peak": HBM_PEAK_GBS, "unit": "GB/s",
+                     "frac": algo / (dt / args.steps) / 1e9 / HBM_PEAK_GBS,
+                     "traffic": None},
+    }), flush=True)
