@@ -247,8 +247,6 @@ def run_cpu_baseline(args, d_blob, d_offsets, d_lens, n_local):
     }
 
 
-if __name__ == "__main__":
-    main()
 
 
 def bench_encode(args, torch, engine, workload, device, n_local, lo):
@@ -350,3 +348,7 @@ def bench_rollup(args, torch, engine, d_blob, d_offsets, d_lens, enc_bytes,
                      "frac": algo / (dt / args.steps) / 1e9 / HBM_PEAK_GBS,
                      "traffic": None},
     }), flush=True)
+
+
+if __name__ == "__main__":
+    main()
