@@ -1086,6 +1086,7 @@ __global__ void __launch_bounds__(BLOCK_THREADS)
 k_rollup_batch(const uint8_t* __restrict__ blobs,
                const uint64_t* __restrict__ offsets,
                const uint32_t* __restrict__ lens,
+               const int32_t* __restrict__ select, /* optional series subset */
                uint32_t nseries, int int_optimized, uint8_t default_unit,
                int metric_type, int64_t window_ns, uint32_t nbuckets,
                RollupPlan plan,
@@ -1096,13 +1097,15 @@ k_rollup_batch(const uint8_t* __restrict__ blobs,
      * branches instead of exec-mask divergence sequences. */
     const uint32_t wave = __builtin_amdgcn_readfirstlane(threadIdx.x / WAVE);
     const uint32_t lane = threadIdx.x % WAVE;
-    const uint32_t series = blockIdx.x * WAVES_PER_BLOCK + wave;
+    const uint32_t slot = blockIdx.x * WAVES_PER_BLOCK + wave;
+    const uint32_t series = __builtin_amdgcn_readfirstlane(
+        (select && slot < nseries) ? (uint32_t)select[slot] : slot);
     /* per-wave bucket value staging for quantiles (timer) */
     __shared__ double qvals_all[WAVES_PER_BLOCK][QCAP];
     __shared__ double sorted_all[WAVES_PER_BLOCK][QCAP];
     double* qvals = qvals_all[wave];
     double* sorted = sorted_all[wave];
-    if (series >= nseries) return;
+    if (slot >= nseries) return;
 
     Decoder d;
     d.init(blobs, offsets[series], lens[series], int_optimized != 0, default_unit);
@@ -1282,6 +1285,189 @@ k_compact(const uint8_t* __restrict__ src, uint32_t src_stride,
     uint64_t* d = (uint64_t*)(dst + dst_offsets[series]);
     uint32_t nwords = (lens[series] + 7) / 8;
     for (uint32_t w = lane; w < nwords; w += WAVE) d[w] = s[w];
+}
+
+
+/* -------- series-per-lane fused rollup --------
+ * Same semantics as k_rollup_batch but one series per LANE (64 parsers per
+ * wave, like k_decode_batch). Timer quantile values stage in a per-lane LDS
+ * row kept sorted by insertion (bucket sizes <= RQCAP_LANE); larger buckets
+ * flag M3GPU_SERIES_BUCKET_OVERFLOW and the host retries those series on the
+ * wave-per-series kernel (64-deep staging). */
+#define RQCAP_LANE 16
+
+template <bool WITH_Q>
+__global__ void __launch_bounds__(BLOCK_THREADS)
+k_rollup_lane(const uint8_t* __restrict__ blobs,
+              const uint64_t* __restrict__ offsets,
+              const uint32_t* __restrict__ lens,
+              uint32_t nseries, int int_optimized, uint8_t default_unit,
+              int metric_type, int64_t window_ns, uint32_t nbuckets,
+              RollupPlan plan,
+              double* __restrict__ out, int64_t* __restrict__ out_window_ts,
+              int32_t* __restrict__ out_errs) {
+    const uint32_t lane = threadIdx.x & (WAVE - 1);
+    const uint32_t wave = __builtin_amdgcn_readfirstlane(threadIdx.x >> 6);
+    const uint32_t series = blockIdx.x * BLOCK_THREADS + wave * WAVE + lane;
+
+    __shared__ double qbuf_all[WITH_Q ? WAVES_PER_BLOCK : 1]
+                              [WITH_Q ? WAVE : 1][WITH_Q ? RQCAP_LANE : 1];
+    double* qrow = WITH_Q ? qbuf_all[wave][lane] : nullptr;
+
+    const bool in_range = series < nseries;
+    Decoder d;
+    if (in_range)
+        d.init(blobs, offsets[series], lens[series], int_optimized != 0, default_unit);
+
+    double* out_row = out + (uint64_t)series * nbuckets * plan.naggs;
+    int64_t* wts_row = out_window_ts ? out_window_ts + (uint64_t)series * nbuckets : nullptr;
+
+    BucketState bs;
+    bs.reset();
+    int64_t base = 0;
+    int64_t cur_bucket = -1;
+    uint32_t nq = 0;
+    int err = 0;
+    bool running = in_range;
+
+    auto emit_bucket = [&](int64_t b) {
+        if (b < 0 || b >= (int64_t)nbuckets) return;
+        if (wts_row) wts_row[b] = base + (b + 1) * window_ns;
+        for (int k = 0; k < plan.naggs; k++) {
+            int32_t t = plan.agg_types[k];
+            int8_t qi = plan.qidx[k];
+            double r = 0;
+            if (metric_type == M3GPU_METRIC_COUNTER) { /* counter.go:112-131 */
+                switch (t) {
+                case M3GPU_AGG_MIN: r = (double)bs.imin; break;
+                case M3GPU_AGG_MAX: r = (double)bs.imax; break;
+                case M3GPU_AGG_MEAN: r = bs.count ? (double)bs.isum / (double)bs.count : 0; break;
+                case M3GPU_AGG_COUNT: r = (double)bs.count; break;
+                case M3GPU_AGG_SUM: r = (double)bs.isum; break;
+                case M3GPU_AGG_SUMSQ: r = (double)bs.isumsq; break;
+                case M3GPU_AGG_STDEV: r = m3_stdev(bs.count, (double)bs.isumsq, (double)bs.isum); break;
+                default: break;
+                }
+            } else if (metric_type == M3GPU_METRIC_GAUGE) { /* gauge.go:144-165 */
+                switch (t) {
+                case M3GPU_AGG_LAST: r = bs.last; break;
+                case M3GPU_AGG_MIN: r = bs.fmin; break;
+                case M3GPU_AGG_MAX: r = bs.fmax; break;
+                case M3GPU_AGG_MEAN: r = bs.count ? bs.fsum / (double)bs.count : 0.0; break;
+                case M3GPU_AGG_COUNT: r = (double)bs.count; break;
+                case M3GPU_AGG_SUM: r = bs.fsum; break;
+                case M3GPU_AGG_SUMSQ: r = bs.fsumsq; break;
+                case M3GPU_AGG_STDEV: r = m3_stdev(bs.count, bs.fsumsq, bs.fsum); break;
+                default: break;
+                }
+            } else { /* timer.go:131-153; qrow holds the sorted bucket values */
+                if (WITH_Q && qi >= 0) {
+                    if (nq == 0) r = 0.0; /* empty stream Quantile -> 0 */
+                    else if (nq <= 3) { /* quantilesFromBuf :210-229 */
+                        uint32_t idx = (uint32_t)(plan.qs[qi] * (double)nq);
+                        if (idx >= nq) idx = nq - 1;
+                        r = qrow[idx];
+                    } else { /* calcQuantiles walk closed form */
+                        int kk = 0;
+                        for (int i = 0; i <= qi; i++) {
+                            int rank = (int)ceil(plan.qs[i] * (double)nq);
+                            kk = (i == 0) ? rank : ((rank > kk + 1) ? rank : kk + 1);
+                        }
+                        if (kk > (int)nq) kk = nq;
+                        r = qrow[kk - 1];
+                    }
+                } else {
+                    switch (t) {
+                    case M3GPU_AGG_MIN: r = (WITH_Q && nq) ? qrow[0] : 0.0; break;
+                    case M3GPU_AGG_MAX: r = (WITH_Q && nq) ? qrow[nq - 1] : 0.0; break;
+                    case M3GPU_AGG_MEAN: r = bs.count ? bs.fsum / (double)bs.count : 0.0; break;
+                    case M3GPU_AGG_COUNT: r = (double)bs.count; break;
+                    case M3GPU_AGG_SUM: r = bs.fsum; break;
+                    case M3GPU_AGG_SUMSQ: r = bs.fsumsq; break;
+                    case M3GPU_AGG_STDEV: r = m3_stdev(bs.count, bs.fsumsq, bs.fsum); break;
+                    default: break;
+                    }
+                }
+            }
+            out_row[(uint64_t)b * plan.naggs + k] = r;
+        }
+    };
+
+    while (__any(running)) {
+        if (running) {
+            int64_t t;
+            double v;
+            int rstat = d.next(&t, &v);
+            if (rstat < 0) {
+                err = -rstat;
+                running = false;
+            } else {
+                bool have = rstat == 1;
+                int64_t b = -1;
+                bool ok = true;
+                if (have) {
+                    if (cur_bucket < 0) base = (t / window_ns) * window_ns; /* Truncate */
+                    if (t < base) { err = M3GPU_SERIES_UNSORTED; running = false; ok = false; }
+                    else {
+                        b = (t - base) / window_ns;
+                        if (b >= (int64_t)nbuckets) { err = M3GPU_SERIES_CAPACITY; running = false; ok = false; }
+                        else if (b < cur_bucket) { err = M3GPU_SERIES_UNSORTED; running = false; ok = false; }
+                    }
+                }
+                if (ok && (!have || b != cur_bucket)) {
+                    if (cur_bucket >= 0) emit_bucket(cur_bucket);
+                    int64_t stop = have ? b : (int64_t)nbuckets;
+                    for (int64_t eb = (cur_bucket < 0 ? 0 : cur_bucket + 1); eb < stop; eb++) {
+                        bs.reset();
+                        nq = 0;
+                        emit_bucket(eb);
+                    }
+                    if (!have) { running = false; ok = false; }
+                    else {
+                        bs.reset();
+                        nq = 0;
+                        cur_bucket = b;
+                    }
+                }
+                if (ok) {
+                    if (metric_type == M3GPU_METRIC_COUNTER) { /* counter.go:52-78 */
+                        int64_t iv = go_f2i(v);
+                        bs.isum += iv;
+                        bs.count++;
+                        if (bs.imax < iv) bs.imax = iv;
+                        if (bs.imin > iv) bs.imin = iv;
+                        bs.isumsq += iv * iv;
+                    } else if (metric_type == M3GPU_METRIC_GAUGE) { /* gauge.go:73-103 */
+                        if (bs.last_at == 0 || t > bs.last_at) { bs.last_at = t; bs.last = v; }
+                        bs.count++;
+                        if (!isnan(v)) {
+                            bs.fsum += v;
+                            if (isnan(bs.fmax) || bs.fmax < v) bs.fmax = v;
+                            if (isnan(bs.fmin) || bs.fmin > v) bs.fmin = v;
+                            bs.fsumsq += v * v;
+                        }
+                    } else { /* timer.go:56-75 */
+                        bs.count++;
+                        bs.fsum += v;
+                        bs.fsumsq += v * v;
+                        if (WITH_Q && plan.nq > 0) {
+                            if (nq >= RQCAP_LANE) {
+                                err = M3GPU_SERIES_BUCKET_OVERFLOW;
+                                running = false;
+                            } else {
+                                /* insertion into this lane's sorted LDS row */
+                                uint32_t j = nq;
+                                while (j > 0 && qrow[j - 1] > v) { qrow[j] = qrow[j - 1]; j--; }
+                                qrow[j] = v;
+                                nq++;
+                            }
+                        }
+                    }
+                }
+            }
+        }
+    }
+    if (in_range) out_errs[series] = err;
 }
 
 } // namespace m3
@@ -1465,12 +1651,58 @@ int m3gpu_rollup_batch_dev(
     }
 
     hipStream_t s = (hipStream_t)hip_stream;
-    hipLaunchKernelGGL(m3::k_rollup_batch, dim3(grid_for(nseries)),
-                       dim3(BLOCK_THREADS), 0, s,
-                       d_blobs, d_offsets, d_lens, nseries, int_optimized,
-                       default_unit, metric_type, window_ns, nbuckets, plan,
-                       d_out, d_out_window_ts, d_out_errs);
+    bool with_q = plan.nq > 0 && metric_type == M3GPU_METRIC_TIMER;
+    if (with_q)
+        hipLaunchKernelGGL(m3::k_rollup_lane<true>, dim3(grid_lane(nseries)),
+                           dim3(BLOCK_THREADS), 0, s,
+                           d_blobs, d_offsets, d_lens, nseries, int_optimized,
+                           default_unit, metric_type, window_ns, nbuckets, plan,
+                           d_out, d_out_window_ts, d_out_errs);
+    else
+        hipLaunchKernelGGL(m3::k_rollup_lane<false>, dim3(grid_lane(nseries)),
+                           dim3(BLOCK_THREADS), 0, s,
+                           d_blobs, d_offsets, d_lens, nseries, int_optimized,
+                           default_unit, metric_type, window_ns, nbuckets, plan,
+                           d_out, d_out_window_ts, d_out_errs);
     HIP_TRY(hipGetLastError());
+    if (with_q) {
+        /* buckets deeper than RQCAP_LANE overflow the per-lane staging:
+         * retry exactly those series on the wave-per-series kernel (64-deep
+         * staging). Requires a stream sync to inspect the error flags. */
+        HIP_TRY(hipStreamSynchronize(s));
+        int32_t* h_errs = (int32_t*)malloc(nseries * sizeof(int32_t));
+        if (!h_errs) { snprintf(g_err, sizeof(g_err), "oom"); return M3GPU_ERR_HIP; }
+        hipError_t ce = hipMemcpy(h_errs, d_out_errs, nseries * sizeof(int32_t),
+                                  hipMemcpyDeviceToHost);
+        if (ce != hipSuccess) { free(h_errs); return set_hip_err("errs copy", ce); }
+        uint32_t nretry = 0;
+        for (uint32_t i = 0; i < nseries; i++)
+            if (h_errs[i] == M3GPU_SERIES_BUCKET_OVERFLOW) nretry++;
+        if (nretry) {
+            int32_t* h_sel = (int32_t*)malloc(nretry * sizeof(int32_t));
+            uint32_t k = 0;
+            for (uint32_t i = 0; i < nseries; i++)
+                if (h_errs[i] == M3GPU_SERIES_BUCKET_OVERFLOW) h_sel[k++] = (int32_t)i;
+            int32_t* d_sel = nullptr;
+            hipError_t e2 = hipMalloc(&d_sel, nretry * sizeof(int32_t));
+            if (e2 == hipSuccess)
+                e2 = hipMemcpy(d_sel, h_sel, nretry * sizeof(int32_t), hipMemcpyHostToDevice);
+            if (e2 == hipSuccess) {
+                hipLaunchKernelGGL(m3::k_rollup_batch, dim3(grid_for(nretry)),
+                                   dim3(BLOCK_THREADS), 0, s,
+                                   d_blobs, d_offsets, d_lens, d_sel, nretry,
+                                   int_optimized, default_unit, metric_type,
+                                   window_ns, nbuckets, plan,
+                                   d_out, d_out_window_ts, d_out_errs);
+                e2 = hipGetLastError();
+                if (e2 == hipSuccess) e2 = hipStreamSynchronize(s);
+            }
+            hipFree(d_sel);
+            free(h_sel);
+            if (e2 != hipSuccess) { free(h_errs); return set_hip_err("rollup retry", e2); }
+        }
+        free(h_errs);
+    }
     return M3GPU_OK;
 }
 

@@ -264,3 +264,40 @@ def test_decode_perm_schedule_matches(torch, engine):
     assert np.array_equal(outs[0][1].view(np.uint64), outs[1][1].view(np.uint64))
     assert np.array_equal(outs[0][2], outs[1][2])
     assert np.array_equal(outs[0][0], ts)
+
+
+def test_rollup_deep_buckets_retry(torch, engine):
+    """Timer buckets deeper than the per-lane staging (16) must transparently
+    retry on the wave-per-series kernel — results still exact vs oracle."""
+    from m3_amd.engine import pack_streams
+    rng = np.random.default_rng(29)
+    nseries, npts = 96, 300
+    start = (1427162462 * 10**9 // (60 * 10**9)) * 60 * 10**9
+    ts = start + np.arange(npts, dtype=np.int64) * 10**9  # 1s cadence: 60/bucket
+    ts = np.broadcast_to(ts, (nseries, npts)).copy()
+    ts[:16] = start + np.arange(npts, dtype=np.int64) * 10 * 10**9  # some shallow
+    vals = np.round(rng.random((nseries, npts)) * 1e4, 4)
+    counts = np.full(nseries, npts, np.uint32)
+    aggs = ["sum", "min", "max", "median", "p95", "p99", "count"]
+    window = 60 * 10**9
+    base = (ts[:, 0] // window) * window
+    nbuckets = int(((ts[:, -1] - base) // window).max()) + 1
+    o_out, o_wts = oracle.rollup_batch(ts, vals, counts, oracle.METRIC_TIMER,
+                                       window, nbuckets, aggs)
+    streams = [oracle.encode_series(ts[i], vals[i], start_ns=int(ts[i, 0]))
+               for i in range(nseries)]
+    blob, offsets, lens = pack_streams(streams)
+    d_blob = torch.from_numpy(blob).to("cuda:0")
+    d_off = torch.from_numpy(offsets.astype(np.int64)).to("cuda:0")
+    d_lens = torch.from_numpy(lens.astype(np.int32)).to("cuda:0")
+    out = torch.empty((nseries, nbuckets, len(aggs)), dtype=torch.float64,
+                      device="cuda:0")
+    wts = torch.empty((nseries, nbuckets), dtype=torch.int64, device="cuda:0")
+    errs = torch.empty(nseries, dtype=torch.int32, device="cuda:0")
+    engine.rollup_batch_dev(d_blob, d_off, d_lens, engine.METRIC_TIMER,
+                            window, nbuckets, aggs, out, wts, errs)
+    torch.cuda.synchronize()
+    assert np.all(errs.cpu().numpy() == 0)  # retried series cleared
+    assert np.array_equal(wts.cpu().numpy(), o_wts)
+    g = out.cpu().numpy()
+    assert np.array_equal(g.view(np.uint64), o_out.view(np.uint64))
