@@ -1006,31 +1006,69 @@ k_encode_batch(const int64_t* __restrict__ ts, const double* __restrict__ vals,
                uint32_t stride, int int_optimized, uint8_t unit,
                uint8_t* __restrict__ out_bytes, uint32_t out_stride,
                uint32_t* __restrict__ out_lens, int32_t* __restrict__ out_errs) {
-    /* ONE SERIES PER LANE (like decode): 64 independent encoders per wave,
-     * per-lane bit emitters storing completed words to their own rows. */
-    const uint32_t series = blockIdx.x * BLOCK_THREADS + threadIdx.x;
-    if (series >= nseries) return;
+    /* ONE SERIES PER LANE (like decode). Input points stage through an LDS
+     * tile filled cooperatively — fill step j: lane l loads row (l>>3)+8j,
+     * point (l&7): 8 consecutive 8B addresses per row = full 64B-line
+     * utilization — instead of 64 independent 8B gathers per point. */
+    const uint32_t lane = threadIdx.x & (WAVE - 1);
+    const uint32_t wave = __builtin_amdgcn_readfirstlane(threadIdx.x >> 6);
+    const uint32_t s_base = blockIdx.x * BLOCK_THREADS + wave * WAVE;
+    const uint32_t series = s_base + lane;
 
-    const int64_t* row_ts = ts + (uint64_t)series * stride;
-    const double* row_vals = vals + (uint64_t)series * stride;
-    uint32_t n = counts[series];
+    __shared__ int64_t ts_tile_all[WAVES_PER_BLOCK][WAVE][DEC_TILE + 1];
+    __shared__ double val_tile_all[WAVES_PER_BLOCK][WAVE][DEC_TILE + 1];
+    int64_t (*ts_tile)[DEC_TILE + 1] = ts_tile_all[wave];
+    double (*val_tile)[DEC_TILE + 1] = val_tile_all[wave];
+
+    const bool in_range = series < nseries;
+    uint32_t n = in_range ? counts[series] : 0;
 
     Encoder e;
-    e.init(out_bytes + (uint64_t)series * out_stride, out_stride,
-           n ? row_ts[0] : 0, int_optimized != 0, unit);
+    if (in_range)
+        e.init(out_bytes + (uint64_t)series * out_stride, out_stride,
+               n ? ts[(uint64_t)series * stride] : 0, int_optimized != 0, unit);
 
+    auto fill_tile = [&](uint32_t base_pt) {
+        __builtin_amdgcn_wave_barrier();
+        const uint32_t p = lane & (DEC_TILE - 1);
+        const uint32_t r0 = lane / DEC_TILE;
+        for (uint32_t j = 0; j < DEC_TILE; j++) {
+            uint32_t r = r0 + j * (WAVE / DEC_TILE);
+            uint32_t c = (uint32_t)__shfl((int)n, (int)r);
+            uint32_t pt = base_pt + p;
+            if (pt < c) {
+                uint64_t row = (uint64_t)(s_base + r) * stride + pt;
+                ts_tile[r][p] = ts[row];
+                val_tile[r][p] = vals[row];
+            }
+        }
+        __builtin_amdgcn_wave_barrier();
+    };
+
+    bool running = in_range && n > 0;
     int err = 0;
-    for (uint32_t j = 0; j < n; j++) {
-        err = e.encode(row_ts[j], row_vals[j], unit);
-        if (err) break;
-    }
     uint32_t len = 0;
-    if (!err) {
-        len = e.finalize();
-        err = e.w.err;
+    uint32_t j = 0;
+    while (__any(running)) {
+        if ((j & (DEC_TILE - 1)) == 0) fill_tile(j);
+        if (running) {
+            int64_t t = ts_tile[lane][j & (DEC_TILE - 1)];
+            double v = val_tile[lane][j & (DEC_TILE - 1)];
+            err = e.encode(t, v, unit);
+            if (err) {
+                running = false;
+            } else if (j + 1 == n) {
+                len = e.finalize();
+                err = e.w.err;
+                running = false;
+            }
+        }
+        j++;
     }
-    out_lens[series] = err ? 0 : len;
-    out_errs[series] = err;
+    if (in_range) {
+        out_lens[series] = err ? 0 : len;
+        out_errs[series] = err;
+    }
 }
 
 /* ===================== fused decode -> rollup kernel ===================== */
