@@ -21,9 +21,9 @@
  *  - Encoded streams are the reference's on-disk M3TSZ block format
  *    (src/dbnode/encoding/m3tsz, incl. the EOS-marker tail), bit-exact.
  *  - Packed stream layout: blobs[offsets[i] .. offsets[i]+lens[i]) is series
- *    i's stream. offsets[i] MUST be 8-byte aligned (pad between streams with
- *    zero bytes; the total buffer padded to a multiple of 8). lens[] are the
- *    true stream lengths.
+ *    i's stream. offsets[i] MUST be 16-byte aligned (pad between streams with
+ *    zero bytes; the total buffer padded to a multiple of 16) so stream
+ *    refills are single aligned 16B loads. lens[] are true stream lengths.
  *  - Time units use the reference's xtime.Unit byte values
  *    (src/x/time/unit.go:30-42): 1=s, 2=ms, 3=us, 4=ns.
  */

@@ -109,7 +109,7 @@ def pack_streams(streams):
     8-byte-aligned zero-padded layout the C ABI requires."""
     n = len(streams)
     lens = np.fromiter((len(s) for s in streams), dtype=np.uint32, count=n)
-    padded = (lens.astype(np.uint64) + 7) & ~np.uint64(7)
+    padded = (lens.astype(np.uint64) + 15) & ~np.uint64(15)
     offsets = np.zeros(n + 1, dtype=np.uint64)
     np.cumsum(padded, out=offsets[1:])
     blob = np.zeros(int(offsets[-1]), dtype=np.uint8)
