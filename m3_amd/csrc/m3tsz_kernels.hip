@@ -93,64 +93,43 @@ __device__ __forceinline__ double go_modf(double v, double* ip) {
 
 struct BitReader {
     /* Register bit window with a cheap common path (bits served from `cur`)
-     * and a PAIRED prefetch pipeline: stream words are fetched two at a time
-     * with one 16-byte load (layout contract: stream offsets 16B-aligned,
-     * zero-padded), issued 2-3 words before first use — halving the
-     * line-granular fetch count and doubling latency cover vs single-word
-     * refills. Byte-stream semantics identical to istream.go:73-115 over
-     * reader64.go:40-80 (the zero padding reproduces the zero-filled
-     * partial tail word). */
+     * and a two-deep word pipeline: `nxt` is the word the reference reader
+     * would fetch next (so EOF semantics match istream.go:73-115 over
+     * reader64.go:40-80 exactly), `pf` is prefetched a further word ahead —
+     * its load is issued ~128 bits before first use, hiding refill latency
+     * under parsing. The blob's zero padding reproduces reader64's
+     * zero-filled partial tail word. */
     const uint64_t* words;
     int64_t len;        /* true byte length */
-    int64_t index;      /* byte index of the next PAIR to prefetch (16B mult) */
+    int64_t index;      /* byte index of the next word to prefetch */
     uint64_t cur;       /* left-aligned buffered bits */
     uint32_t rem;       /* valid bits in cur */
-    uint64_t nxt;       /* next word (the one the reference would read now) */
+    uint64_t nxt;       /* next word (left-aligned) */
     uint32_t nxt_bits;
-    uint64_t spare;     /* word after nxt (second of the current pair) */
-    uint32_t spare_bits;
-    bool have_spare;
-    uint64_t b0, b1;    /* prefetched pair */
-    uint32_t b0_bits, b1_bits;
+    uint64_t pf;        /* prefetched word after nxt */
+    uint32_t pf_bits;
 
-    __device__ __forceinline__ void prefetch_pair() {
+    __device__ __forceinline__ void prefetch() {
         if (index < len) {
-            ulonglong2 w = *(const ulonglong2*)(words + (index >> 3));
-            b0 = __builtin_bswap64(w.x);
-            b1 = __builtin_bswap64(w.y);
-            int64_t a0 = len - index;
-            b0_bits = a0 >= 8 ? 64 : (uint32_t)(8 * a0);
-            int64_t a1 = a0 - 8;
-            b1_bits = a1 >= 8 ? 64 : (a1 > 0 ? (uint32_t)(8 * a1) : 0);
-            index += 16;
+            pf = __builtin_bswap64(words[index >> 3]);
+            int64_t avail = len - index;
+            pf_bits = avail >= 8 ? 64 : (uint32_t)(8 * avail);
+            index += 8;
         } else {
-            b0 = 0; b1 = 0;
-            b0_bits = 0; b1_bits = 0;
+            pf = 0;
+            pf_bits = 0;
         }
     }
-    __device__ __forceinline__ void advance_word() {
-        if (have_spare) {
-            nxt = spare;
-            nxt_bits = spare_bits;
-            have_spare = false;
-        } else {
-            nxt = b0;
-            nxt_bits = b0_bits;
-            spare = b1;
-            spare_bits = b1_bits;
-            have_spare = true;
-            prefetch_pair();
-        }
-    }
-    __device__ void init(const uint8_t* base, uint64_t off, uint32_t l) {
+    __device__ __forceinline__ void init(const uint8_t* base, uint64_t off, uint32_t l) {
         words = (const uint64_t*)(base + off);
         len = l;
         index = 0;
         cur = 0;
         rem = 0;
-        have_spare = false;
-        prefetch_pair();
-        advance_word();   /* nxt = word0, spare = word1, next pair loading */
+        prefetch();
+        nxt = pf;
+        nxt_bits = pf_bits;
+        prefetch();
     }
     __device__ __forceinline__ int read_bits(uint32_t n, uint64_t* out) {
         if (n <= rem) {
@@ -165,7 +144,9 @@ struct BitReader {
         res = ((need >= 64) ? 0 : (res << need)) | (nxt >> (64 - need));
         cur = (need >= 64) ? 0 : (nxt << need);
         rem = nxt_bits - need;
-        advance_word();
+        nxt = pf;
+        nxt_bits = pf_bits;
+        prefetch();
         *out = res;
         return 0;
     }
@@ -179,7 +160,9 @@ struct BitReader {
         uint32_t need = n - rem;
         cur = (need >= 64) ? 0 : (nxt << need);
         rem = nxt_bits - need;
-        advance_word();
+        nxt = pf;
+        nxt_bits = pf_bits;
+        prefetch();
     }
     __device__ __forceinline__ int peek_bits(uint32_t n, uint64_t* out) {
         if (n <= rem) { *out = n ? (cur >> (64 - n)) : 0; return 0; }

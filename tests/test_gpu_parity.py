@@ -301,3 +301,30 @@ def test_rollup_deep_buckets_retry(torch, engine):
     assert np.array_equal(wts.cpu().numpy(), o_wts)
     g = out.cpu().numpy()
     assert np.array_equal(g.view(np.uint64), o_out.view(np.uint64))
+
+
+def test_full_pipeline_roundtrip_large(torch, engine):
+    """Full-size property test (no oracle in the loop): GPU encode of 32k
+    generated series -> compact -> GPU decode reproduces the inputs
+    bit-exactly. Exercises the whole product pipeline at depth 1440."""
+    from m3_amd import workload
+    nseries, npts = 32768, 1440
+    d_blob, d_offsets, d_lens, enc_bytes = workload.encode_on_device(
+        torch, nseries, npts, chunk=16384, device="cuda:0")
+    out_ts = torch.empty((nseries, npts), dtype=torch.int64, device="cuda:0")
+    out_vals = torch.empty((nseries, npts), dtype=torch.float64, device="cuda:0")
+    out_counts = torch.empty(nseries, dtype=torch.int32, device="cuda:0")
+    out_errs = torch.empty(nseries, dtype=torch.int32, device="cuda:0")
+    engine.decode_batch_dev(d_blob, d_offsets, d_lens, out_ts, out_vals,
+                            out_counts, out_errs,
+                            d_perm=torch.argsort(d_lens).to(torch.int32))
+    torch.cuda.synchronize()
+    assert int(out_errs.abs().sum().item()) == 0
+    assert bool((out_counts == npts).all().item())
+    g_ts = out_ts.cpu().numpy()
+    g_vals = out_vals.cpu().numpy()
+    for s0 in range(0, nseries, 16384):
+        ts, vals = workload.gen_chunk(s0, 16384, npts)
+        assert np.array_equal(g_ts[s0:s0 + 16384], ts)
+        assert np.array_equal(g_vals[s0:s0 + 16384].view(np.uint64),
+                              vals.view(np.uint64))
