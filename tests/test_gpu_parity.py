@@ -303,10 +303,12 @@ def test_rollup_deep_buckets_retry(torch, engine):
     assert np.array_equal(g.view(np.uint64), o_out.view(np.uint64))
 
 
-def test_full_pipeline_roundtrip_large(torch, engine):
-    """Full-size property test (no oracle in the loop): GPU encode of 32k
-    generated series -> compact -> GPU decode reproduces the inputs
-    bit-exactly. Exercises the whole product pipeline at depth 1440."""
+def test_full_pipeline_parity_large(torch, engine):
+    """Full-size parity: GPU encode of 32k series x 1440 pts -> compact ->
+    GPU decode, compared BIT-EXACTLY against the oracle decoding the same
+    blob on the CPU. (Comparing against the generated inputs instead would
+    be wrong: convertToIntFloat canonicalizes near-decimal floats — the
+    reference-documented lossy conversion, m3tsz.go:72-77.)"""
     from m3_amd import workload
     nseries, npts = 32768, 1440
     d_blob, d_offsets, d_lens, enc_bytes = workload.encode_on_device(
@@ -323,8 +325,10 @@ def test_full_pipeline_roundtrip_large(torch, engine):
     assert bool((out_counts == npts).all().item())
     g_ts = out_ts.cpu().numpy()
     g_vals = out_vals.cpu().numpy()
-    for s0 in range(0, nseries, 16384):
-        ts, vals = workload.gen_chunk(s0, 16384, npts)
-        assert np.array_equal(g_ts[s0:s0 + 16384], ts)
-        assert np.array_equal(g_vals[s0:s0 + 16384].view(np.uint64),
-                              vals.view(np.uint64))
+    assert np.all(np.diff(g_ts, axis=1) > 0)  # strictly monotonic everywhere
+    blob = d_blob.cpu().numpy()
+    offs = d_offsets.cpu().numpy().astype(np.uint64)
+    o_ts, o_vals, o_counts = oracle.decode_batch(blob, offs, stride=npts)
+    assert np.array_equal(o_counts, np.full(nseries, npts, np.uint32))
+    assert np.array_equal(g_ts, o_ts)
+    assert np.array_equal(g_vals.view(np.uint64), o_vals.view(np.uint64))
