@@ -332,3 +332,50 @@ def test_full_pipeline_parity_large(torch, engine):
     assert np.array_equal(o_counts, np.full(nseries, npts, np.uint32))
     assert np.array_equal(g_ts, o_ts)
     assert np.array_equal(g_vals.view(np.uint64), o_vals.view(np.uint64))
+
+
+def test_decode_edge_cases(torch, engine):
+    """Edge grammar paths on GPU: us/ns units (64-bit default DoD buckets),
+    int-overflow values (sig==64 slow path), huge time jumps, float/int mode
+    flapping, repeats — all bit-exact vs the oracle."""
+    rng = np.random.default_rng(99)
+    streams, exp = [], []
+    start = 1427162462 * 10**9
+    # us-unit and ns-unit streams (64-bit default bucket)
+    for unit in (3, 4):
+        ts = start + np.cumsum(rng.integers(1, 10**7, 100))  # ns steps
+        if unit == 3:
+            ts = (ts // 1000) * 1000  # us-aligned
+        vals = rng.random(100) * 1e6
+        units = np.full(100, unit, np.uint8)
+        streams.append(oracle.encode_series(ts, vals, units=units,
+                                            start_ns=start - 1))
+        exp.append((ts, None))
+    # overflow / giant-diff values (sig 64, float<->int flapping)
+    li, ln = float(2**63 - 2), float(-(2**63) + 1)
+    vals = np.array([li, 10, ln, 10, ln, li, -12, li, 14.5, li, ln,
+                     12.34858499392, li, 0.0, -0.0, 1e13, 1e13 - 2, 5.0])
+    ts = start + np.arange(len(vals)) * 10**9
+    streams.append(oracle.encode_series(ts, vals, start_ns=start))
+    exp.append((ts, None))
+    # repeat-heavy + mode flapping
+    vals = np.tile(np.array([7.0, 7.0, 7.0, 1.5e300, 1.5e300, 7.0]), 40)
+    ts = start + np.arange(len(vals)) * 10**9
+    streams.append(oracle.encode_series(ts, vals, start_ns=start))
+    exp.append((ts, None))
+    # huge forward/backward time jumps (32-bit s-unit default bucket limits)
+    ts = start + np.cumsum(rng.integers(-2000, 500000, 200)) * 10**9
+    ts = np.maximum.accumulate(ts - ts.min() + start)  # keep positive
+    vals = rng.random(200)
+    streams.append(oracle.encode_series(ts, vals, start_ns=int(ts[0])))
+    exp.append((ts, None))
+
+    g_ts, g_vals, g_counts, g_errs = gpu_decode(torch, engine, streams, 256)
+    assert np.all(g_errs == 0)
+    for i, s in enumerate(streams):
+        dec = oracle.decode_series(s, int_optimized=True)
+        n = len(dec["ts"])
+        assert g_counts[i] == n, i
+        assert np.array_equal(g_ts[i, :n], dec["ts"]), i
+        assert np.array_equal(g_vals[i, :n].view(np.uint64),
+                              np.asarray(dec["vals"]).view(np.uint64)), i
