@@ -567,7 +567,7 @@ struct Decoder {
  *   flush step j: lane l stores row (l>>3)+8j, point (l&7) — 8 consecutive
  *   8B addresses per row = full 64B line utilization for ts[] and val[]. */
 
-#define DEC_TILE 4
+#define DEC_TILE 8
 
 __global__ void __launch_bounds__(BLOCK_THREADS)
 k_decode_batch(const uint8_t* __restrict__ blobs,
