@@ -55,6 +55,10 @@ def parse_args():
     p.add_argument("--skip-cpu-baseline", action="store_true")
     p.add_argument("--no-sort", action="store_true",
                    help="disable length-sorted wave scheduling")
+    p.add_argument("--data", choices=["synthetic", "production"],
+                   default="synthetic",
+                   help="production = the 10 real M3TSZ streams embedded in "
+                        "the reference's benchmarks, replicated to nseries")
     p.add_argument("--mode", choices=["decode", "encode", "rollup"],
                    default="decode",
                    help="decode = headline metric (BASELINE configs[1]); "
@@ -82,9 +86,15 @@ def main():
     lo, hi = shard_range(args.nseries, world, rank)
     n_local = hi - lo
     t0 = time.time()
-    d_blob, d_offsets, d_lens, enc_bytes = workload.encode_on_device(
-        torch, n_local, args.npts, chunk=args.chunk, device=device,
-        rank_offset=lo, verbose=(rank == 0))
+    if args.data == "production":
+        d_blob, d_offsets, d_lens, enc_bytes, expected_counts = \
+            build_production_batch(torch, n_local, device)
+        args.npts = int(expected_counts.max())
+    else:
+        d_blob, d_offsets, d_lens, enc_bytes = workload.encode_on_device(
+            torch, n_local, args.npts, chunk=args.chunk, device=device,
+            rank_offset=lo, verbose=(rank == 0))
+        expected_counts = None
     torch.cuda.synchronize()
     if rank == 0:
         print(f"[bench] built {n_local} series x {args.npts} pts: "
@@ -117,9 +127,13 @@ def main():
     # correctness gate outside the timed region: every series decodes fully
     step()
     torch.cuda.synchronize()
-    if int(out_errs.abs().sum().item()) != 0 or \
-       int((out_counts != args.npts).sum().item()) != 0:
+    if expected_counts is None:
+        counts_ok = int((out_counts != args.npts).sum().item()) == 0
+    else:
+        counts_ok = bool((out_counts == expected_counts).all().item())
+    if int(out_errs.abs().sum().item()) != 0 or not counts_ok:
         raise SystemExit("[bench] decode errors in workload — aborting")
+    npts_decoded_local = int(out_counts.to(torch.int64).sum().item())
 
     for _ in range(args.warmup):
         step()
@@ -149,11 +163,11 @@ def main():
         elapsed = float(t.item())
 
     kernel_ms = float(np.mean([a.elapsed_time(b) for a, b in ev]))
-    npts_global = args.nseries * args.npts
+    npts_global = npts_decoded_local * world  # == nseries*npts for synthetic
     dps = npts_global * args.steps / elapsed  # datapoints/sec, whole job
 
     # roofline on the dominant kernel (decode): algorithmic bytes per launch
-    algo_bytes = enc_bytes + 16 * npts_total_local  # stream in + (ts,val) out
+    algo_bytes = enc_bytes + 16 * npts_decoded_local  # stream in + (ts,val) out
     achieved_gbs = algo_bytes / (kernel_ms / 1e3) / 1e9
     read_only_gbs = enc_bytes / (kernel_ms / 1e3) / 1e9
     traffic = None
@@ -187,10 +201,13 @@ def main():
             "dtype": "f64",
             "data": "synthetic",
             "config": {
-                "workload": "1M series x 1440 pts batched M3TSZ decode, HBM-bandwidth microbench",
+                "workload": "1M series x 1440 pts batched M3TSZ decode, HBM-bandwidth microbench"
+                            if args.data == "synthetic" else
+                            "1M replicated production streams batched M3TSZ decode",
                 "nseries": args.nseries,
                 "npts": args.npts,
-                "encoded_bytes_per_pt": enc_bytes / npts_total_local,
+                "data_detail": args.data,
+                "encoded_bytes_per_pt": enc_bytes / npts_decoded_local,
                 "parallelism": f"series-sharded x{world} (no collectives)",
             },
             "roofline": {
@@ -348,6 +365,34 @@ def bench_rollup(args, torch, engine, d_blob, d_offsets, d_lens, enc_bytes,
                      "frac": algo / (dt / args.steps) / 1e9 / HBM_PEAK_GBS,
                      "traffic": None},
     }), flush=True)
+
+
+def build_production_batch(torch, nseries, device):
+    """Replicate the 10 production-encoded streams from the reference's own
+    benchmarks (tests/golden/production_streams.json, 719-720 pts each,
+    0.41-2.97 B/pt) to nseries device-resident unique copies."""
+    import base64
+    import oracle  # stream prep + expected counts only (not the timed path)
+    with open(os.path.join(REPO, "tests", "golden", "production_streams.json")) as f:
+        ps = json.load(f)
+    streams = [base64.b64decode(b) for b in ps["samples"]]
+    counts1 = [len(oracle.decode_series(s)["ts"]) for s in streams]
+    from m3_amd.engine import pack_streams
+    blob1, offs1, lens1 = pack_streams(streams)
+    period = int(offs1[-1])
+    nrep = (nseries + len(streams) - 1) // len(streams)
+    d_small = torch.from_numpy(blob1).to(device)
+    d_blob = d_small.repeat(nrep)[: nrep * period]
+    offsets = (np.tile(offs1[:-1], nrep)
+               + np.repeat(np.arange(nrep, dtype=np.int64) * period, len(streams)))
+    offsets = offsets[:nseries]
+    offsets = np.concatenate([offsets, [nrep * period]]).astype(np.int64)
+    lens = np.tile(lens1, nrep)[:nseries]
+    counts = np.tile(np.asarray(counts1, np.int32), nrep)[:nseries]
+    d_offsets = torch.from_numpy(offsets).to(device)
+    d_lens = torch.from_numpy(lens.astype(np.int32)).to(device)
+    expected_counts = torch.from_numpy(counts).to(device)
+    return d_blob, d_offsets, d_lens, int(lens.astype(np.int64).sum()), expected_counts
 
 
 if __name__ == "__main__":
