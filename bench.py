@@ -199,7 +199,7 @@ def main():
             "scaling": "weak",
             "vs_baseline": None,
             "dtype": "f64",
-            "data": "synthetic",
+            "data": "synthetic" if args.data == "synthetic" else "replicated production streams",
             "config": {
                 "workload": "1M series x 1440 pts batched M3TSZ decode, HBM-bandwidth microbench"
                             if args.data == "synthetic" else
