@@ -537,9 +537,59 @@ struct Decoder {
         return read_int_val_diff();
     }
 
+    /* Fully fused fast path: ONE 64-bit peek decodes the timestamp field
+     * AND the common int-mode value field, then ONE consume advances the
+     * window. Grammar consequence (i) from SURVEY.md Appendix A: short ts
+     * fields (<=16 bits for buckets 0-2) plus a no-update int diff
+     * (2+sig bits) fit one 64-bit window. Falls through to the stepwise
+     * path (bit-identical) for markers, default buckets, float mode,
+     * wide sig, unit changes and near-EOS. Returns 1/0/-err like next().
+     * Returns -1000 to mean "take the general path" (nothing consumed). */
+    __device__ __forceinline__ int next_fused(int64_t* t, double* v) {
+        uint64_t w;
+        if (tu_changed || !have_scheme || !int_optimized || is_float ||
+            prev_time == 0 || sig > 45 || r.peek_bits(64, &w) != 0)
+            return -1000;
+        uint32_t c1;
+        int64_t dod;
+        if (!(w >> 63)) {
+            c1 = 1;
+            dod = 0;
+        } else {
+            if ((w >> 55) == MARKER_OPCODE) return -1000; /* marker */
+            uint32_t top4 = (uint32_t)(w >> 60);
+            uint32_t L = __builtin_clz(~(top4 << 28)); /* leading ones, 1..4 */
+            if (L >= 4) return -1000; /* default bucket: stepwise */
+            uint32_t vb = (L == 1) ? 7 : (L == 2) ? 9 : 12;
+            uint32_t ob = L + 1;
+            dod = sign_extend((w << ob) >> (64 - vb), vb) * UNIT_NS_D[time_unit];
+            c1 = ob + vb;
+        }
+        uint64_t w2 = w << c1;
+        if (w2 >> 63) { /* opcodeNoUpdate -> sign + sig-bit int diff */
+            if (c1 + 2 + sig > 64) return -1000;
+            uint64_t bits = (w2 << 1) >> (63 - sig);
+            r.consume(c1 + 2 + sig);
+            double sgn = -1.0;
+            if ((bits >> sig) == 1) { sgn = 1.0; bits ^= (1ULL << sig); }
+            int_val += sgn * (double)bits;
+        } else if ((w2 >> 62) == 0x1) { /* repeat */
+            r.consume(c1 + 2);
+        } else {
+            return -1000; /* '00': mode/sig/mult update -> stepwise */
+        }
+        prev_time_delta += dod;
+        prev_time += prev_time_delta;
+        *t = prev_time;
+        *v = (mult == 0) ? int_val : int_val / exp10_table(mult);
+        return 1;
+    }
+
     /* One point. Returns 1 = value in (*t,*v), 0 = done, -err on error. */
     __device__ __forceinline__ int next(int64_t* t, double* v) {
         if (done) return 0;
+        int f = next_fused(t, v);
+        if (f != -1000) return f;
         bool first;
         int err = read_timestamp(&first);
         if (err) return -err;
