@@ -545,10 +545,43 @@ struct Decoder {
      * path (bit-identical) for markers, default buckets, float mode,
      * wide sig, unit changes and near-EOS. Returns 1/0/-err like next().
      * Returns -1000 to mean "take the general path" (nothing consumed). */
+    /* XOR field fused from w2 (value bits left-aligned), pre = bits already
+     * classified before the XOR control (ts field + optional mode prefix).
+     * Returns 0 (consumed + state updated) or -1000 (take stepwise path). */
+    __device__ __forceinline__ int fused_xor(uint64_t w2, uint32_t pre) {
+        if (!(w2 >> 63)) { /* '0': same value */
+            r.consume(pre + 1);
+            prev_xor = 0;
+            return 0;
+        }
+        if ((w2 >> 62) == 0x2) { /* '10' contained */
+            uint32_t lead = prev_xor ? __builtin_clzll(prev_xor) : 64;
+            uint32_t trail = prev_xor ? __builtin_ctzll(prev_xor) : 0;
+            uint32_t nmean = 64 - lead - trail;
+            if (pre + 2 + nmean > 64) return -1000;
+            uint64_t mb = nmean ? ((w2 << 2) >> (64 - nmean)) : 0;
+            r.consume(pre + 2 + nmean);
+            prev_xor = mb << trail;
+            prev_float_bits ^= prev_xor;
+            return 0;
+        }
+        /* '11' + 6b lead + 6b (nmean-1), then payload (may exceed the peek) */
+        uint64_t lead = (w2 >> 56) & 0x3f;
+        uint64_t nmean = ((w2 >> 50) & 0x3f) + 1;
+        r.consume(pre + 14);
+        uint64_t mb;
+        int err = r.read_bits((uint32_t)nmean, &mb);
+        if (err) return -err; /* same EOF point as the stepwise reads */
+        uint64_t trail = 64 - lead - nmean;
+        prev_xor = mb << trail;
+        prev_float_bits ^= prev_xor;
+        return 0;
+    }
+
     __device__ __forceinline__ int next_fused(int64_t* t, double* v) {
         uint64_t w;
-        if (tu_changed || !have_scheme || !int_optimized || is_float ||
-            prev_time == 0 || sig > 45 || r.peek_bits(64, &w) != 0)
+        if (tu_changed || !have_scheme || prev_time == 0 ||
+            r.peek_bits(64, &w) != 0)
             return -1000;
         uint32_t c1;
         int64_t dod;
@@ -566,8 +599,23 @@ struct Decoder {
             c1 = ob + vb;
         }
         uint64_t w2 = w << c1;
-        if (w2 >> 63) { /* opcodeNoUpdate -> sign + sig-bit int diff */
-            if (c1 + 2 + sig > 64) return -1000;
+        if (!int_optimized) { /* pure float stream: XOR field directly */
+            int rx = fused_xor(w2, c1);
+            if (rx == -1000) return -1000;
+            if (rx) return rx; /* negative error, same point as stepwise */
+        } else if (is_float) {
+            /* float mode: '1' + XOR | '01' repeat | '00...' stepwise */
+            if (w2 >> 63) {
+                int rx = fused_xor(w2 << 1, c1 + 1);
+                if (rx == -1000) return -1000;
+                if (rx) return rx;
+            } else if ((w2 >> 62) == 0x1) {
+                r.consume(c1 + 2);
+            } else {
+                return -1000;
+            }
+        } else if (w2 >> 63) { /* int mode, opcodeNoUpdate: sign + sig diff */
+            if (sig > 45 || c1 + 2 + sig > 64) return -1000;
             uint64_t bits = (w2 << 1) >> (63 - sig);
             r.consume(c1 + 2 + sig);
             double sgn = -1.0;
@@ -581,7 +629,8 @@ struct Decoder {
         prev_time_delta += dod;
         prev_time += prev_time_delta;
         *t = prev_time;
-        *v = (mult == 0) ? int_val : int_val / exp10_table(mult);
+        if (!int_optimized || is_float) *v = bits2f(prev_float_bits);
+        else *v = (mult == 0) ? int_val : int_val / exp10_table(mult);
         return 1;
     }
 
