@@ -28,6 +28,7 @@ from .engine import (  # noqa: F401
     engine_available,
     lib,
     pack_streams,
+    aggregate_tiles_dev,
     rollup_batch,
     rollup_batch_dev,
 )

@@ -253,3 +253,41 @@ def rollup_batch_dev(d_blob, d_offsets, d_lens, metric_type, window_ns,
         nbuckets, _pp(aggs, c_i32), len(aggs), _dev_ptr(out),
         _dev_ptr(out_window_ts), _dev_ptr(out_errs), _torch_stream())
     _check(rc, "m3gpu_rollup_batch_dev")
+
+
+def aggregate_tiles_dev(torch, d_blob, d_offsets, d_lens, metric_type,
+                        window_ns, nbuckets, agg, int_optimized=True,
+                        default_unit=1, unit_out=1):
+    """AggregateTiles-shaped backend (what dbnode's large-tiles job,
+    storage/shard.go:2682, would drive): fused decode -> windowed rollup with
+    m3aggregator semantics -> re-encode one downsampled M3TSZ stream per
+    series. Each non-empty tile emits (window-END timestamp, agg value)
+    (list.go:541-543); empty tiles emit nothing. Returns (d_tile_bytes
+    [n, out_stride] uint8, d_tile_lens int32, d_counts int32).
+    All work stays on-device (rollup -> gap compaction -> encode)."""
+    nseries = d_lens.numel()
+    aggs = [agg, "count"]
+    out = torch.empty((nseries, nbuckets, 2), dtype=torch.float64, device=d_blob.device)
+    wts = torch.empty((nseries, nbuckets), dtype=torch.int64, device=d_blob.device)
+    errs = torch.empty(nseries, dtype=torch.int32, device=d_blob.device)
+    rollup_batch_dev(d_blob, d_offsets, d_lens, metric_type, window_ns,
+                     nbuckets, aggs, out, wts, errs,
+                     int_optimized=int_optimized, default_unit=default_unit)
+    # compact out the empty tiles (stable: valid buckets keep time order)
+    valid = out[:, :, 1] > 0
+    order = torch.argsort((~valid).to(torch.int8), dim=1, stable=True)
+    ts_c = torch.gather(wts, 1, order)
+    vals_c = torch.gather(out[:, :, 0].contiguous(), 1, order)
+    counts = valid.sum(1).to(torch.int32)
+    out_stride = (24 * nbuckets + 32 + 15) & ~15
+    d_tile_bytes = torch.zeros((nseries, out_stride), dtype=torch.uint8,
+                               device=d_blob.device)
+    d_tile_lens = torch.empty(nseries, dtype=torch.int32, device=d_blob.device)
+    d_tile_errs = torch.empty(nseries, dtype=torch.int32, device=d_blob.device)
+    encode_batch_dev(ts_c.contiguous(), vals_c.contiguous(), counts,
+                     d_tile_bytes, d_tile_lens, d_tile_errs,
+                     int_optimized=int_optimized, unit=unit_out)
+    torch.cuda.synchronize()
+    _raise_series_errors(errs.cpu().numpy(), "tiles rollup")
+    _raise_series_errors(d_tile_errs.cpu().numpy(), "tiles encode")
+    return d_tile_bytes, d_tile_lens, counts

@@ -30,38 +30,33 @@ def _ts_for(n, deltas):
     return START + np.cumsum(steps) * 10**9
 
 
-# |v| < 2^59: near +-2^63 the reference's own float64 diff arithmetic
-# (encoder.go:162 valDiff, iterator.go:175 accumulation) drifts by ULPs and
-# one round is NOT a fixpoint — see test_extreme_magnitude_drift.
-moderate_floats = st.floats(allow_nan=True, allow_infinity=True, width=64
-                            ).filter(lambda v: not (abs(v) > 2.0**59 and abs(v) < 1e300))
-
-
 @settings(max_examples=150, deadline=None)
-@given(vals=st.lists(moderate_floats, min_size=1, max_size=60),
+@given(vals=st.lists(any_floats, min_size=1, max_size=60),
        deltas=st.lists(st.integers(0, 10**15), min_size=60, max_size=60),
        intopt=st.booleans())
-def test_roundtrip_fixpoint(vals, deltas, intopt):
+def test_roundtrip_stabilizes(vals, deltas, intopt):
+    """Timestamps roundtrip exactly every round, and repeated
+    encode->decode stabilizes: the stream bytes become a fixpoint within a
+    few rounds. (Exactly ONE round is NOT a reference guarantee: the lossy
+    convertToIntFloat canonicalization, float64 diff arithmetic near +-2^63
+    — encoder.go:162 / iterator.go:175 — and the single-maxMult scale mixing
+    near 2^53 can each drift values once more before settling; see
+    test_extreme_magnitude_drift for a pinned example.)"""
     n = len(vals)
     ts = _ts_for(n, deltas)
-    enc1 = oracle.encode_series(ts, vals, start_ns=START - 1, int_optimized=intopt)
-    dec1 = oracle.decode_series(enc1, int_optimized=intopt)
-    assert np.array_equal(dec1["ts"], ts)
-    # canonical values: one more round is lossless and byte-stable
-    enc2 = oracle.encode_series(dec1["ts"], dec1["vals"], start_ns=START - 1,
-                                int_optimized=intopt)
-    dec2 = oracle.decode_series(enc2, int_optimized=intopt)
-    assert np.array_equal(dec2["ts"], ts)
-    a = np.asarray(dec1["vals"])
-    b = np.asarray(dec2["vals"])
-    eq = (a.view(np.uint64) == b.view(np.uint64)) | (np.isnan(a) & np.isnan(b))
-    if intopt:
-        # -0.0 folds into +0.0 via the repeat opcode (encoder.go:200-206)
-        eq |= (a == b)
-    assert eq.all(), (a[~eq], b[~eq])
-    enc3 = oracle.encode_series(dec2["ts"], dec2["vals"], start_ns=START - 1,
-                                int_optimized=intopt)
-    assert enc2 == enc3
+    cur = np.asarray(vals, dtype=np.float64)
+    prev_enc = None
+    for round_i in range(6):
+        enc = oracle.encode_series(ts, cur, start_ns=START - 1,
+                                   int_optimized=intopt)
+        if enc == prev_enc:
+            break
+        dec = oracle.decode_series(enc, int_optimized=intopt)
+        assert np.array_equal(dec["ts"], ts), round_i
+        cur = np.asarray(dec["vals"])
+        prev_enc = enc
+    else:
+        raise AssertionError(f"no fixpoint after 6 rounds: {list(cur)}")
 
 
 @settings(max_examples=200, deadline=None)
