@@ -59,7 +59,8 @@ def parse_args():
                    default="synthetic",
                    help="production = the 10 real M3TSZ streams embedded in "
                         "the reference's benchmarks, replicated to nseries")
-    p.add_argument("--mode", choices=["decode", "encode", "rollup"],
+    p.add_argument("--mode", choices=["decode", "encode", "rollup", "tiles",
+                                      "parse-only"],
                    default="decode",
                    help="decode = headline metric (BASELINE configs[1]); "
                         "encode/rollup = BASELINE configs[2]/[3]")
@@ -110,6 +111,10 @@ def main():
         bench_rollup(args, torch, engine, d_blob, d_offsets, d_lens,
                      enc_bytes, device, n_local)
         return
+    if args.mode == "tiles":
+        bench_tiles(args, torch, engine, d_blob, d_offsets, d_lens,
+                    enc_bytes, device, n_local)
+        return
     out_ts = torch.empty((n_local, args.npts), dtype=torch.int64, device=device)
     out_vals = torch.empty((n_local, args.npts), dtype=torch.float64, device=device)
     out_counts = torch.empty(n_local, dtype=torch.int32, device=device)
@@ -120,9 +125,22 @@ def main():
     if not args.no_sort:
         d_perm = torch.argsort(d_lens).to(torch.int32)
 
+    parse_only = args.mode == "parse-only"
+
     def step():
-        engine.decode_batch_dev(d_blob, d_offsets, d_lens, out_ts, out_vals,
-                                out_counts, out_errs, d_perm=d_perm)
+        if parse_only:
+            # diagnostic: decode with outputs discarded (null out pointers)
+            from m3_amd.engine import lib as _lib, _dev_ptr, _torch_stream, _check
+            rc = _lib().m3gpu_decode_batch_dev_perm(
+                _dev_ptr(d_blob), _dev_ptr(d_offsets), _dev_ptr(d_lens),
+                _dev_ptr(d_perm) if d_perm is not None else None,
+                d_lens.numel(), 1, 1, None, None,
+                _dev_ptr(out_counts), _dev_ptr(out_errs), args.npts,
+                _torch_stream())
+            _check(rc, "decode parse-only")
+        else:
+            engine.decode_batch_dev(d_blob, d_offsets, d_lens, out_ts, out_vals,
+                                    out_counts, out_errs, d_perm=d_perm)
 
     # correctness gate outside the timed region: every series decodes fully
     step()
@@ -393,6 +411,42 @@ def build_production_batch(torch, nseries, device):
     d_lens = torch.from_numpy(lens.astype(np.int32)).to(device)
     expected_counts = torch.from_numpy(counts).to(device)
     return d_blob, d_offsets, d_lens, int(lens.astype(np.int64).sum()), expected_counts
+
+
+def bench_tiles(args, torch, engine, d_blob, d_offsets, d_lens, enc_bytes,
+                device, n_local):
+    """AggregateTiles-shaped job: decode -> 10s->1m
+    rollup -> re-encode the downsampled tile streams, all on-device."""
+    window = 60 * 10**9
+    nbuckets = args.npts * 10 // 60
+
+    def step():
+        return engine.aggregate_tiles_dev(torch, d_blob, d_offsets, d_lens,
+                                          engine.METRIC_GAUGE, window,
+                                          nbuckets, "last")
+    tb, tl, tc = step()
+    out_bytes = int(tl.to(torch.int64).sum().item())
+    for _ in range(args.warmup):
+        step()
+    torch.cuda.synchronize()
+    t0 = time.time()
+    for _ in range(args.steps):
+        step()
+    torch.cuda.synchronize()
+    dt = time.time() - t0
+    npts_total = n_local * args.npts
+    dps = npts_total * args.steps / dt
+    print(json.dumps({
+        "metric": "datapoints/sec AggregateTiles (decode->1m last->re-encode)",
+        "value": dps, "unit": "datapoints/sec", "n_gpus": 1,
+        "steps": args.steps, "warmup": args.warmup,
+        "ms_per_step": dt / args.steps * 1e3, "higher_is_better": True,
+        "scaling": "weak", "vs_baseline": None, "dtype": "f64",
+        "data": "synthetic",
+        "config": {"workload": "AggregateTiles 10s->1m, 1x MI355X",
+                   "nseries": n_local, "npts": args.npts,
+                   "tile_bytes_out": out_bytes},
+    }), flush=True)
 
 
 if __name__ == "__main__":

@@ -704,7 +704,9 @@ k_decode_batch(const uint8_t* __restrict__ blobs,
     uint32_t cnt = 0;
     uint32_t k = 0;
 
+    const bool discard = out_ts == nullptr; /* parse-only diagnostic */
     auto flush = [&](uint32_t base_pt) {
+        if (discard) return;
         __builtin_amdgcn_wave_barrier();
         const uint32_t p = lane & (DEC_TILE - 1);
         const uint32_t r0 = lane / DEC_TILE;
@@ -743,8 +745,10 @@ k_decode_batch(const uint8_t* __restrict__ blobs,
                 err = M3GPU_SERIES_CAPACITY;
                 running = false;
             } else {
-                ts_tile[lane][k & (DEC_TILE - 1)] = t;
-                val_tile[lane][k & (DEC_TILE - 1)] = v;
+                if (!discard) {
+                    ts_tile[lane][k & (DEC_TILE - 1)] = t;
+                    val_tile[lane][k & (DEC_TILE - 1)] = v;
+                }
                 cnt++;
             }
         }
