@@ -1845,7 +1845,7 @@ int m3gpu_rollup_batch_dev(
                 e2 = hipGetLastError();
                 if (e2 == hipSuccess) e2 = hipStreamSynchronize(s);
             }
-            hipFree(d_sel);
+            (void)hipFree(d_sel);
             free(h_sel);
             if (e2 != hipSuccess) { free(h_errs); return set_hip_err("rollup retry", e2); }
         }
@@ -1890,8 +1890,8 @@ int m3gpu_decode_batch(
         HIP_TRY(hipMemcpy(out_counts, d_counts, nseries * sizeof(uint32_t), hipMemcpyDeviceToHost));
         HIP_TRY(hipMemcpy(out_errs, d_errs, nseries * sizeof(int32_t), hipMemcpyDeviceToHost));
     }
-    hipFree(d_blobs); hipFree(d_offsets); hipFree(d_lens);
-    hipFree(d_ts); hipFree(d_vals); hipFree(d_counts); hipFree(d_errs);
+    (void)hipFree(d_blobs); hipFree(d_offsets); hipFree(d_lens);
+    (void)hipFree(d_ts); hipFree(d_vals); hipFree(d_counts); hipFree(d_errs);
     return rc;
 }
 
@@ -1927,8 +1927,8 @@ int m3gpu_encode_batch(
         HIP_TRY(hipMemcpy(out_lens, d_lens, nseries * sizeof(uint32_t), hipMemcpyDeviceToHost));
         HIP_TRY(hipMemcpy(out_errs, d_errs, nseries * sizeof(int32_t), hipMemcpyDeviceToHost));
     }
-    hipFree(d_ts); hipFree(d_vals); hipFree(d_counts);
-    hipFree(d_out); hipFree(d_lens); hipFree(d_errs);
+    (void)hipFree(d_ts); hipFree(d_vals); hipFree(d_counts);
+    (void)hipFree(d_out); hipFree(d_lens); hipFree(d_errs);
     return rc;
 }
 
@@ -1966,8 +1966,8 @@ int m3gpu_rollup_batch(
             HIP_TRY(hipMemcpy(out_window_ts, d_wts, (uint64_t)nseries * nbuckets * sizeof(int64_t), hipMemcpyDeviceToHost));
         HIP_TRY(hipMemcpy(out_errs, d_errs, nseries * sizeof(int32_t), hipMemcpyDeviceToHost));
     }
-    hipFree(d_blobs); hipFree(d_offsets); hipFree(d_lens);
-    hipFree(d_out); hipFree(d_wts); hipFree(d_errs);
+    (void)hipFree(d_blobs); hipFree(d_offsets); hipFree(d_lens);
+    (void)hipFree(d_out); hipFree(d_wts); hipFree(d_errs);
     return rc;
 }
 
