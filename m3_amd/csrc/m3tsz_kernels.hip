@@ -705,33 +705,28 @@ k_decode_batch(const uint8_t* __restrict__ blobs,
     uint32_t k = 0;
 
     const bool discard = out_ts == nullptr; /* parse-only diagnostic */
-    /* flush geometry is loop-invariant: lane handles column p of rows
-     * r0 + j*8; precompute each row's output base offset once */
-    const uint32_t fl_p = lane & (DEC_TILE - 1);
-    const uint32_t fl_r0 = lane / DEC_TILE;
-    uint64_t fl_base[DEC_TILE];
-    for (uint32_t j = 0; j < DEC_TILE; j++) {
-        uint32_t r = fl_r0 + j * (WAVE / DEC_TILE);
-        fl_base[j] = (uint64_t)__shfl((int)series, (int)r) * stride + fl_p;
-    }
     auto flush = [&](uint32_t base_pt) {
         if (discard) return;
         __builtin_amdgcn_wave_barrier();
-        const uint32_t pt = base_pt + fl_p;
+        const uint32_t p = lane & (DEC_TILE - 1);
+        const uint32_t r0 = lane / DEC_TILE;
+        const uint32_t pt = base_pt + p;
         if (__all(cnt >= base_pt + DEC_TILE)) {
             /* all 64 rows full: unconditional stores, no per-row counts */
             for (uint32_t j = 0; j < DEC_TILE; j++) {
-                uint32_t r = fl_r0 + j * (WAVE / DEC_TILE);
-                out_ts[fl_base[j] + base_pt] = ts_tile[r][fl_p];
-                out_vals[fl_base[j] + base_pt] = val_tile[r][fl_p];
+                uint32_t r = r0 + j * (WAVE / DEC_TILE);
+                uint64_t row = (uint64_t)__shfl((int)series, (int)r);
+                out_ts[row * stride + pt] = ts_tile[r][p];
+                out_vals[row * stride + pt] = val_tile[r][p];
             }
         } else {
             for (uint32_t j = 0; j < DEC_TILE; j++) {
-                uint32_t r = fl_r0 + j * (WAVE / DEC_TILE);
+                uint32_t r = r0 + j * (WAVE / DEC_TILE);
                 uint32_t c = (uint32_t)__shfl((int)cnt, (int)r);
+                uint64_t row = (uint64_t)__shfl((int)series, (int)r);
                 if (pt < c) {
-                    out_ts[fl_base[j] + base_pt] = ts_tile[r][fl_p];
-                    out_vals[fl_base[j] + base_pt] = val_tile[r][fl_p];
+                    out_ts[row * stride + pt] = ts_tile[r][p];
+                    out_vals[row * stride + pt] = val_tile[r][p];
                 }
             }
         }
@@ -1890,8 +1885,8 @@ int m3gpu_decode_batch(
         HIP_TRY(hipMemcpy(out_counts, d_counts, nseries * sizeof(uint32_t), hipMemcpyDeviceToHost));
         HIP_TRY(hipMemcpy(out_errs, d_errs, nseries * sizeof(int32_t), hipMemcpyDeviceToHost));
     }
-    (void)hipFree(d_blobs); hipFree(d_offsets); hipFree(d_lens);
-    (void)hipFree(d_ts); hipFree(d_vals); hipFree(d_counts); hipFree(d_errs);
+    (void)hipFree(d_blobs); (void)hipFree(d_offsets); (void)hipFree(d_lens);
+    (void)hipFree(d_ts); (void)hipFree(d_vals); (void)hipFree(d_counts); (void)hipFree(d_errs);
     return rc;
 }
 
@@ -1927,8 +1922,8 @@ int m3gpu_encode_batch(
         HIP_TRY(hipMemcpy(out_lens, d_lens, nseries * sizeof(uint32_t), hipMemcpyDeviceToHost));
         HIP_TRY(hipMemcpy(out_errs, d_errs, nseries * sizeof(int32_t), hipMemcpyDeviceToHost));
     }
-    (void)hipFree(d_ts); hipFree(d_vals); hipFree(d_counts);
-    (void)hipFree(d_out); hipFree(d_lens); hipFree(d_errs);
+    (void)hipFree(d_ts); (void)hipFree(d_vals); (void)hipFree(d_counts);
+    (void)hipFree(d_out); (void)hipFree(d_lens); (void)hipFree(d_errs);
     return rc;
 }
 
@@ -1966,8 +1961,8 @@ int m3gpu_rollup_batch(
             HIP_TRY(hipMemcpy(out_window_ts, d_wts, (uint64_t)nseries * nbuckets * sizeof(int64_t), hipMemcpyDeviceToHost));
         HIP_TRY(hipMemcpy(out_errs, d_errs, nseries * sizeof(int32_t), hipMemcpyDeviceToHost));
     }
-    (void)hipFree(d_blobs); hipFree(d_offsets); hipFree(d_lens);
-    (void)hipFree(d_out); hipFree(d_wts); hipFree(d_errs);
+    (void)hipFree(d_blobs); (void)hipFree(d_offsets); (void)hipFree(d_lens);
+    (void)hipFree(d_out); (void)hipFree(d_wts); (void)hipFree(d_errs);
     return rc;
 }
 
