@@ -1476,6 +1476,11 @@ k_rollup_lane(const uint8_t* __restrict__ blobs,
     auto emit_bucket = [&](int64_t b) {
         if (b < 0 || b >= (int64_t)nbuckets) return;
         if (wts_row) wts_row[b] = base + (b + 1) * window_ns;
+        /* even naggs: emit value pairs as aligned 16B stores (the rows are
+         * naggs*8B apart, so paired stores halve the line-granular write
+         * amplification of the scattered per-lane emission) */
+        const bool paired = (plan.naggs & 1) == 0;
+        double pend = 0;
         for (int k = 0; k < plan.naggs; k++) {
             int32_t t = plan.agg_types[k];
             int8_t qi = plan.qidx[k];
@@ -1532,7 +1537,18 @@ k_rollup_lane(const uint8_t* __restrict__ blobs,
                     }
                 }
             }
-            out_row[(uint64_t)b * plan.naggs + k] = r;
+            if (paired) {
+                if (k & 1) {
+                    double2 pr;
+                    pr.x = pend;
+                    pr.y = r;
+                    *(double2*)(out_row + (uint64_t)b * plan.naggs + k - 1) = pr;
+                } else {
+                    pend = r;
+                }
+            } else {
+                out_row[(uint64_t)b * plan.naggs + k] = r;
+            }
         }
     };
 
