@@ -273,12 +273,17 @@ def aggregate_tiles_dev(torch, d_blob, d_offsets, d_lens, metric_type,
     rollup_batch_dev(d_blob, d_offsets, d_lens, metric_type, window_ns,
                      nbuckets, aggs, out, wts, errs,
                      int_optimized=int_optimized, default_unit=default_unit)
-    # compact out the empty tiles (stable: valid buckets keep time order)
+    # compact out the empty tiles (stable: valid buckets keep time order);
+    # when no series has gaps (the common dense case) skip the gather
     valid = out[:, :, 1] > 0
-    order = torch.argsort((~valid).to(torch.int8), dim=1, stable=True)
-    ts_c = torch.gather(wts, 1, order)
-    vals_c = torch.gather(out[:, :, 0].contiguous(), 1, order)
     counts = valid.sum(1).to(torch.int32)
+    if bool((counts == nbuckets).all().item()):
+        ts_c = wts
+        vals_c = out[:, :, 0]
+    else:
+        order = torch.argsort((~valid).to(torch.int8), dim=1, stable=True)
+        ts_c = torch.gather(wts, 1, order)
+        vals_c = torch.gather(out[:, :, 0].contiguous(), 1, order)
     out_stride = (24 * nbuckets + 32 + 15) & ~15
     d_tile_bytes = torch.zeros((nseries, out_stride), dtype=torch.uint8,
                                device=d_blob.device)
