@@ -774,19 +774,28 @@ struct BitWriter {
     uint64_t acc;       /* left-aligned bit accumulator */
     uint32_t used;      /* bits used in acc */
     uint32_t nwords;    /* full words emitted */
-    uint64_t* out;      /* aligned output row */
+    uint64_t pend;      /* buffered even word: pairs store as aligned 16B */
+    uint64_t* out;      /* 16B-aligned output row */
     uint32_t cap_words;
     int err;
 
     __device__ __forceinline__ void init(uint8_t* row, uint32_t cap_bytes) {
-        acc = 0; used = 0; nwords = 0;
+        acc = 0; used = 0; nwords = 0; pend = 0;
         out = (uint64_t*)row;
         cap_words = cap_bytes / 8;
         err = 0;
     }
     __device__ __forceinline__ void emit_word(uint64_t w) {
         if (nwords >= cap_words) { err = M3GPU_SERIES_CAPACITY; return; }
-        out[nwords++] = __builtin_bswap64(w);
+        if (nwords & 1) {
+            ulonglong2 pr;
+            pr.x = pend;
+            pr.y = __builtin_bswap64(w);
+            *(ulonglong2*)(out + nwords - 1) = pr;
+        } else {
+            pend = __builtin_bswap64(w);
+        }
+        nwords++;
     }
     __device__ __forceinline__ void write_bits(uint64_t v, uint32_t n) {
         if (n == 0 || err) return;
@@ -805,8 +814,9 @@ struct BitWriter {
         }
     }
     __device__ __forceinline__ void write_bit(uint32_t b) { write_bits(b, 1); }
-    /* Flush the partial tail word. Returns total byte length. */
+    /* Flush the buffered word + partial tail word. Returns byte length. */
     __device__ __forceinline__ uint32_t finish() {
+        if (nwords & 1) out[nwords - 1] = pend; /* odd word count: flush pend */
         uint32_t nbytes = nwords * 8;
         if (used > 0) {
             if (nwords >= cap_words) { err = M3GPU_SERIES_CAPACITY; return 0; }
