@@ -688,10 +688,13 @@ k_decode_batch(const uint8_t* __restrict__ blobs,
      * Purely a schedule: outputs still land in series order. */
     const uint32_t series = (perm && slot < nseries) ? (uint32_t)perm[slot] : slot;
 
-    __shared__ int64_t ts_tile_all[WAVES_PER_BLOCK][WAVE][DEC_TILE + 1];
-    __shared__ double val_tile_all[WAVES_PER_BLOCK][WAVE][DEC_TILE + 1];
-    int64_t (*ts_tile)[DEC_TILE + 1] = ts_tile_all[wave];
-    double (*val_tile)[DEC_TILE + 1] = val_tile_all[wave];
+    /* unpadded tiles with an XOR column swizzle (col ^ (row & 7)): bank
+     * conflict-free for both per-lane access and the coalesced tile pass,
+     * and the exact 32 KB/block admits 5 blocks/CU (5 waves/SIMD). */
+    __shared__ int64_t ts_tile_all[WAVES_PER_BLOCK][WAVE][DEC_TILE];
+    __shared__ double val_tile_all[WAVES_PER_BLOCK][WAVE][DEC_TILE];
+    int64_t (*ts_tile)[DEC_TILE] = ts_tile_all[wave];
+    double (*val_tile)[DEC_TILE] = val_tile_all[wave];
 
 
     const bool in_range = slot < nseries;
@@ -715,18 +718,20 @@ k_decode_batch(const uint8_t* __restrict__ blobs,
             /* all 64 rows full: unconditional stores, no per-row counts */
             for (uint32_t j = 0; j < DEC_TILE; j++) {
                 uint32_t r = r0 + j * (WAVE / DEC_TILE);
+                uint32_t pc = p ^ (r & 7);
                 uint64_t row = (uint64_t)__shfl((int)series, (int)r);
-                out_ts[row * stride + pt] = ts_tile[r][p];
-                out_vals[row * stride + pt] = val_tile[r][p];
+                out_ts[row * stride + pt] = ts_tile[r][pc];
+                out_vals[row * stride + pt] = val_tile[r][pc];
             }
         } else {
             for (uint32_t j = 0; j < DEC_TILE; j++) {
                 uint32_t r = r0 + j * (WAVE / DEC_TILE);
+                uint32_t pc = p ^ (r & 7);
                 uint32_t c = (uint32_t)__shfl((int)cnt, (int)r);
                 uint64_t row = (uint64_t)__shfl((int)series, (int)r);
                 if (pt < c) {
-                    out_ts[row * stride + pt] = ts_tile[r][p];
-                    out_vals[row * stride + pt] = val_tile[r][p];
+                    out_ts[row * stride + pt] = ts_tile[r][pc];
+                    out_vals[row * stride + pt] = val_tile[r][pc];
                 }
             }
         }
@@ -746,8 +751,9 @@ k_decode_batch(const uint8_t* __restrict__ blobs,
                 running = false;
             } else {
                 if (!discard) {
-                    ts_tile[lane][k & (DEC_TILE - 1)] = t;
-                    val_tile[lane][k & (DEC_TILE - 1)] = v;
+                    uint32_t col = (k & (DEC_TILE - 1)) ^ (lane & 7);
+                    ts_tile[lane][col] = t;
+                    val_tile[lane][col] = v;
                 }
                 cnt++;
             }
@@ -1130,10 +1136,13 @@ k_encode_batch(const int64_t* __restrict__ ts, const double* __restrict__ vals,
     const uint32_t s_base = blockIdx.x * BLOCK_THREADS + wave * WAVE;
     const uint32_t series = s_base + lane;
 
-    __shared__ int64_t ts_tile_all[WAVES_PER_BLOCK][WAVE][DEC_TILE + 1];
-    __shared__ double val_tile_all[WAVES_PER_BLOCK][WAVE][DEC_TILE + 1];
-    int64_t (*ts_tile)[DEC_TILE + 1] = ts_tile_all[wave];
-    double (*val_tile)[DEC_TILE + 1] = val_tile_all[wave];
+    /* unpadded tiles with an XOR column swizzle (col ^ (row & 7)): bank
+     * conflict-free for both per-lane access and the coalesced tile pass,
+     * and the exact 32 KB/block admits 5 blocks/CU (5 waves/SIMD). */
+    __shared__ int64_t ts_tile_all[WAVES_PER_BLOCK][WAVE][DEC_TILE];
+    __shared__ double val_tile_all[WAVES_PER_BLOCK][WAVE][DEC_TILE];
+    int64_t (*ts_tile)[DEC_TILE] = ts_tile_all[wave];
+    double (*val_tile)[DEC_TILE] = val_tile_all[wave];
 
     const bool in_range = series < nseries;
     uint32_t n = in_range ? counts[series] : 0;
@@ -1153,8 +1162,9 @@ k_encode_batch(const int64_t* __restrict__ ts, const double* __restrict__ vals,
             uint32_t pt = base_pt + p;
             if (pt < c) {
                 uint64_t row = (uint64_t)(s_base + r) * stride + pt;
-                ts_tile[r][p] = ts[row];
-                val_tile[r][p] = vals[row];
+                uint32_t pc = p ^ (r & 7);
+                ts_tile[r][pc] = ts[row];
+                val_tile[r][pc] = vals[row];
             }
         }
         __builtin_amdgcn_wave_barrier();
@@ -1167,8 +1177,9 @@ k_encode_batch(const int64_t* __restrict__ ts, const double* __restrict__ vals,
     while (__any(running)) {
         if ((j & (DEC_TILE - 1)) == 0) fill_tile(j);
         if (running) {
-            int64_t t = ts_tile[lane][j & (DEC_TILE - 1)];
-            double v = val_tile[lane][j & (DEC_TILE - 1)];
+            uint32_t col = (j & (DEC_TILE - 1)) ^ (lane & 7);
+            int64_t t = ts_tile[lane][col];
+            double v = val_tile[lane][col];
             err = e.encode(t, v, unit);
             if (err) {
                 running = false;
