@@ -269,3 +269,34 @@ def xxhash64(data):
         buf = np.zeros(1, dtype=np.uint8)
         return lib().oracle_xxhash64(_pu8(buf), 0)
     return lib().oracle_xxhash64(_pu8(buf), len(buf))
+
+
+def merge_batch(ts, vals, counts, out_stride=None, nthreads=0):
+    """Replica-deduplicating merge (MultiReaderIterator semantics over one
+    slice of R replica iterators). ts/vals: [R, nseries, stride]; counts:
+    [R, nseries]. Returns (out_ts, out_vals, out_counts, out_errs)."""
+    ts = np.ascontiguousarray(ts, dtype=np.int64)
+    vals = np.ascontiguousarray(vals, dtype=np.float64)
+    counts = np.ascontiguousarray(counts, dtype=np.uint32)
+    r, nseries, stride = ts.shape
+    if out_stride is None:
+        out_stride = stride * r
+    out_ts = np.zeros((nseries, out_stride), dtype=np.int64)
+    out_vals = np.zeros((nseries, out_stride), dtype=np.float64)
+    out_counts = np.zeros(nseries, dtype=np.uint32)
+    out_errs = np.zeros(nseries, dtype=np.int32)
+    L = lib()
+    L.oracle_merge_batch.restype = c_int
+    L.oracle_merge_batch.argtypes = [P(c_i64), P(c_f64), P(c_u32), c_int,
+                                     c_i64, c_i64, P(c_i64), P(c_f64),
+                                     P(c_u32), c_i64, P(c_i32), c_int]
+    if nthreads <= 0:
+        nthreads = os.cpu_count()
+    L.oracle_merge_batch(ts.ctypes.data_as(P(c_i64)),
+                         vals.ctypes.data_as(P(c_f64)),
+                         counts.ctypes.data_as(P(c_u32)), r, nseries, stride,
+                         out_ts.ctypes.data_as(P(c_i64)),
+                         out_vals.ctypes.data_as(P(c_f64)),
+                         out_counts.ctypes.data_as(P(c_u32)), out_stride,
+                         out_errs.ctypes.data_as(P(c_i32)), nthreads)
+    return out_ts, out_vals, out_counts, out_errs

@@ -627,3 +627,112 @@ int oracle_rollup_batch(
     }
     return err;
 }
+
+/* ==================== multi-replica deduplicating merge ====================
+ * Restates the reference's MultiReaderIterator over ONE slice of R replica
+ * iterators (dbnode/encoding/multi_reader_iterator.go:62-155 with the
+ * iterators collection, dbnode/encoding/iterators.go:56-237), operating on
+ * already-decoded (ts, val) arrays:
+ *  - `values` keeps push order; an exhausted iterator is removed by
+ *    swapping the TAIL into its slot (iterators.go:181-196);
+ *  - the `earliest` set is rebuilt by scanning `values` in order
+ *    (iterators.go:211-215 tryAddEarliest);
+ *  - equal timestamps resolve via IterateLastPushed (the default,
+ *    iterators_types.go:41-56): current() returns earliest[len-1]
+ *    (iterators.go:60-90);
+ *  - after advancing, an equal earliest time is deduped by advancing again
+ *    (multi_reader_iterator.go:139-158), a smaller one is
+ *    errOutOfOrderIterator (iterators.go:229-236).
+ */
+
+int oracle_merge_series(
+    const int64_t* const* ts_rows, const double* const* val_rows,
+    const uint32_t* counts, int nreplicas,
+    int64_t* out_ts, double* out_vals, int64_t cap, int64_t* out_n) {
+    int values[64]; /* replica ids in `values` order */
+    int64_t cursor[64];
+    int nvals = 0;
+    if (nreplicas > 64) return -2;
+    for (int r = 0; r < nreplicas; r++) {
+        if (counts[r] > 0) {
+            values[nvals] = r;
+            cursor[r] = 0;
+            nvals++;
+        }
+    }
+    int64_t n = 0;
+    int64_t prev_at = 0;
+    int first = 1;
+    while (nvals > 0) {
+        /* rebuild earliest: scan values in order */
+        int64_t earliest_at = INT64_MAX;
+        int winner = -1; /* last in scan order among minima */
+        for (int i = 0; i < nvals; i++) {
+            int r = values[i];
+            int64_t t = ts_rows[r][cursor[r]];
+            if (t < earliest_at) {
+                earliest_at = t;
+                winner = r;
+            } else if (t == earliest_at) {
+                winner = r; /* IterateLastPushed: later scan position wins */
+            }
+        }
+        if (!first) {
+            if (earliest_at < prev_at) { *out_n = n; return -1; } /* errOutOfOrderIterator */
+            if (earliest_at == prev_at) {
+                /* dedupe: advance the earliest set again without emitting */
+            }
+        }
+        if (first || earliest_at != prev_at) {
+            if (n >= cap) { *out_n = n; return -3; }
+            out_ts[n] = earliest_at;
+            out_vals[n] = val_rows[winner][cursor[winner]];
+            n++;
+            prev_at = earliest_at;
+            first = 0;
+        }
+        /* moveToValidNext: advance every iterator at earliest_at; remove
+         * exhausted ones by swap-with-tail */
+        for (int i = 0; i < nvals; i++) {
+            int r = values[i];
+            if (ts_rows[r][cursor[r]] == earliest_at) {
+                cursor[r]++;
+                if (cursor[r] >= (int64_t)counts[r]) {
+                    values[i] = values[nvals - 1];
+                    nvals--;
+                    i--; /* re-examine the swapped-in entry */
+                }
+            }
+        }
+    }
+    *out_n = n;
+    return 0;
+}
+
+/* Batch form over SoA rows: replica-major layout ts[(r*nseries + i)*stride]. */
+int oracle_merge_batch(
+    const int64_t* ts, const double* vals, const uint32_t* counts,
+    int nreplicas, int64_t nseries, int64_t stride,
+    int64_t* out_ts, double* out_vals, uint32_t* out_counts,
+    int64_t out_stride, int32_t* out_errs, int nthreads) {
+    (void)nthreads;
+#pragma omp parallel for schedule(dynamic, 64) num_threads(nthreads)
+    for (int64_t i = 0; i < nseries; i++) {
+        const int64_t* tr[64];
+        const double* vr[64];
+        uint32_t cr[64];
+        for (int r = 0; r < nreplicas && r < 64; r++) {
+            tr[r] = ts + ((int64_t)r * nseries + i) * stride;
+            vr[r] = vals + ((int64_t)r * nseries + i) * stride;
+            cr[r] = counts[(int64_t)r * nseries + i];
+        }
+        int64_t n = 0;
+        int rc = oracle_merge_series(tr, vr, cr, nreplicas,
+                                     out_ts + i * out_stride,
+                                     out_vals + i * out_stride,
+                                     out_stride, &n);
+        out_counts[i] = (uint32_t)n;
+        out_errs[i] = rc;
+    }
+    return 0;
+}
