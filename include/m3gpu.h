@@ -145,6 +145,18 @@ int m3gpu_rollup_batch_dev(
     double* d_out, int64_t* d_out_window_ts, int32_t* d_out_errs,
     void* hip_stream);
 
+/* -------- replica-deduplicating merge (replaces MultiReaderIterator over
+ * one slice of R replica iterators: dbnode/encoding/multi_reader_iterator.go
+ * :62-155 + iterators.go:56-237, IterateLastPushed default). Decoded replica
+ * rows (replica-major: row r*nseries+i) merge into one row per series with
+ * equal timestamps deduped (last in values-order wins) and decreasing
+ * timestamps flagged (err 100 = errOutOfOrderIterator). R <= 4. -------- */
+int m3gpu_merge_batch_dev(
+    const int64_t* d_ts, const double* d_vals, const uint32_t* d_counts,
+    uint32_t nreplicas, uint32_t nseries, uint32_t stride,
+    int64_t* d_out_ts, double* d_out_vals, uint32_t* d_out_counts,
+    int32_t* d_out_errs, uint32_t out_stride, void* hip_stream);
+
 int m3gpu_rollup_batch(
     const uint8_t* blobs, uint64_t blobs_len,
     const uint64_t* offsets, const uint32_t* lens,

@@ -1650,6 +1650,108 @@ k_rollup_lane(const uint8_t* __restrict__ blobs,
     if (in_range) out_errs[series] = err;
 }
 
+
+/* ===================== replica-deduplicating merge =====================
+ * MultiReaderIterator semantics over one slice of R (<=4) replica rows of
+ * already-decoded points (multi_reader_iterator.go:62-155 +
+ * iterators.go:56-237, IterateLastPushed default): the `values` order starts
+ * as replica order and mutates only by swap-with-tail on exhaustion; among
+ * equal earliest timestamps the LAST in `values` order wins; equal-to-prev
+ * timestamps are deduped; a decreasing one is errOutOfOrderIterator.
+ * One series per lane; order bookkeeping packed in a u32 (a byte per slot)
+ * so everything stays in registers. */
+#define MERGE_MAX_R 4
+#define MERGE_ERR_OUT_OF_ORDER 100
+
+__global__ void __launch_bounds__(BLOCK_THREADS)
+k_merge(const int64_t* __restrict__ ts, const double* __restrict__ vals,
+        const uint32_t* __restrict__ counts, uint32_t nreplicas,
+        uint32_t nseries, uint32_t stride,
+        int64_t* __restrict__ out_ts, double* __restrict__ out_vals,
+        uint32_t* __restrict__ out_counts, int32_t* __restrict__ out_errs,
+        uint32_t out_stride) {
+    const uint32_t series = blockIdx.x * BLOCK_THREADS + threadIdx.x;
+    if (series >= nseries) return;
+
+    const int64_t* tr[MERGE_MAX_R];
+    const double* vr[MERGE_MAX_R];
+    uint32_t cnt[MERGE_MAX_R];
+    uint32_t cur[MERGE_MAX_R];
+    int64_t cur_ts[MERGE_MAX_R];
+    double cur_val[MERGE_MAX_R];
+
+    uint32_t ord = 0; /* packed replica ids, byte per slot, slot 0 = lowest */
+    uint32_t nvals = 0;
+#pragma unroll
+    for (uint32_t r = 0; r < MERGE_MAX_R; r++) {
+        if (r < nreplicas) {
+            tr[r] = ts + ((uint64_t)r * nseries + series) * stride;
+            vr[r] = vals + ((uint64_t)r * nseries + series) * stride;
+            cnt[r] = counts[(uint64_t)r * nseries + series];
+            cur[r] = 0;
+            if (cnt[r] > 0) {
+                cur_ts[r] = tr[r][0];
+                cur_val[r] = vr[r][0];
+                ord |= r << (8 * nvals);
+                nvals++;
+            }
+        }
+    }
+
+    int64_t* orow_ts = out_ts + (uint64_t)series * out_stride;
+    double* orow_val = out_vals + (uint64_t)series * out_stride;
+    uint32_t n = 0;
+    int64_t prev_at = 0;
+    bool first = true;
+    int err = 0;
+
+    while (nvals > 0) {
+        /* earliest scan in `ord` order; last min wins (IterateLastPushed) */
+        int64_t earliest = INT64_MAX;
+        uint32_t winner = 0;
+#pragma unroll
+        for (uint32_t i = 0; i < MERGE_MAX_R; i++) {
+            if (i < nvals) {
+                uint32_t r = (ord >> (8 * i)) & 0xff;
+                int64_t t = cur_ts[r];
+                if (t <= earliest) {
+                    earliest = t;
+                    winner = r;
+                }
+            }
+        }
+        if (!first && earliest < prev_at) { err = MERGE_ERR_OUT_OF_ORDER; break; }
+        if (first || earliest != prev_at) {
+            if (n >= out_stride) { err = M3GPU_SERIES_CAPACITY; break; }
+            orow_ts[n] = earliest;
+            orow_val[n] = cur_val[winner];
+            n++;
+            prev_at = earliest;
+            first = false;
+        }
+        /* advance every replica at `earliest`; swap-with-tail on exhaustion
+         * (ascending slot order with re-examination == the reference's
+         * removal order, see DESIGN.md merge note) */
+        for (uint32_t i = 0; i < nvals;) {
+            uint32_t r = (ord >> (8 * i)) & 0xff;
+            if (cur_ts[r] == earliest) {
+                cur[r]++;
+                if (cur[r] >= cnt[r]) {
+                    uint32_t tail = (ord >> (8 * (nvals - 1))) & 0xff;
+                    ord = (ord & ~(0xffu << (8 * i))) | (tail << (8 * i));
+                    nvals--;
+                    continue; /* re-examine the swapped-in slot */
+                }
+                cur_ts[r] = tr[r][cur[r]];
+                cur_val[r] = vr[r][cur[r]];
+            }
+            i++;
+        }
+    }
+    out_counts[series] = n;
+    out_errs[series] = err;
+}
+
 } // namespace m3
 
 /* ============================ C-ABI host layer ============================ */
@@ -1883,6 +1985,26 @@ int m3gpu_rollup_batch_dev(
         }
         free(h_errs);
     }
+    return M3GPU_OK;
+}
+
+int m3gpu_merge_batch_dev(
+    const int64_t* d_ts, const double* d_vals, const uint32_t* d_counts,
+    uint32_t nreplicas, uint32_t nseries, uint32_t stride,
+    int64_t* d_out_ts, double* d_out_vals, uint32_t* d_out_counts,
+    int32_t* d_out_errs, uint32_t out_stride, void* hip_stream) {
+    if (!nseries) return M3GPU_OK;
+    if (nreplicas < 1 || nreplicas > MERGE_MAX_R) {
+        snprintf(g_err, sizeof(g_err), "nreplicas must be 1..%d", MERGE_MAX_R);
+        return M3GPU_ERR_BADARG;
+    }
+    hipStream_t s = (hipStream_t)hip_stream;
+    hipLaunchKernelGGL(m3::k_merge, dim3(grid_lane(nseries)),
+                       dim3(BLOCK_THREADS), 0, s,
+                       d_ts, d_vals, d_counts, nreplicas, nseries, stride,
+                       d_out_ts, d_out_vals, d_out_counts, d_out_errs,
+                       out_stride);
+    HIP_TRY(hipGetLastError());
     return M3GPU_OK;
 }
 

@@ -296,3 +296,25 @@ def aggregate_tiles_dev(torch, d_blob, d_offsets, d_lens, metric_type,
     _raise_series_errors(errs.cpu().numpy(), "tiles rollup")
     _raise_series_errors(d_tile_errs.cpu().numpy(), "tiles encode")
     return d_tile_bytes, d_tile_lens, counts
+
+
+def merge_batch_dev(d_ts, d_vals, d_counts, nreplicas, out_ts, out_vals,
+                    out_counts, out_errs):
+    """Replica-deduplicating merge (MultiReaderIterator semantics,
+    IterateLastPushed). d_ts/d_vals: [nreplicas*nseries, stride] rows
+    (replica-major); outputs [nseries, out_stride]."""
+    nseries = d_counts.numel() // nreplicas
+    stride = d_ts.shape[-1]
+    out_stride = out_ts.shape[-1]
+    L = lib()
+    if not hasattr(L.m3gpu_merge_batch_dev, "_configured"):
+        L.m3gpu_merge_batch_dev.restype = c_int
+        L.m3gpu_merge_batch_dev.argtypes = [c_vp, c_vp, c_vp, c_u32, c_u32,
+                                            c_u32, c_vp, c_vp, c_vp, c_vp,
+                                            c_u32, c_vp]
+        L.m3gpu_merge_batch_dev._configured = True
+    rc = L.m3gpu_merge_batch_dev(
+        _dev_ptr(d_ts), _dev_ptr(d_vals), _dev_ptr(d_counts), nreplicas,
+        nseries, stride, _dev_ptr(out_ts), _dev_ptr(out_vals),
+        _dev_ptr(out_counts), _dev_ptr(out_errs), out_stride, _torch_stream())
+    _check(rc, "m3gpu_merge_batch_dev")

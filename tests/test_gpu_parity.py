@@ -429,3 +429,66 @@ def test_aggregate_tiles_vs_oracle(torch, engine, metric, agg):
     tb = tile_bytes.cpu().numpy()
     for i in range(nseries):
         assert bytes(tb[i, :tl[i]]) == exp_streams[i], i
+
+
+def test_merge_replicas_vs_oracle(torch, engine):
+    """GPU replica merge (k_merge) vs the oracle MultiReaderIterator
+    restatement: random ragged overlapping replicas, bit-exact (incl. the
+    IterateLastPushed tie winner and the out-of-order error point)."""
+    rng = np.random.default_rng(47)
+    nseries, stride, R = 512, 64, 3
+    START = 1700000000 * 10**9
+    ts = np.zeros((R, nseries, stride), np.int64)
+    vals = np.zeros((R, nseries, stride), np.float64)
+    counts = np.zeros((R, nseries), np.uint32)
+    for r in range(R):
+        for i in range(nseries):
+            n = int(rng.integers(0, stride))
+            t = np.sort(rng.choice(np.arange(200), size=n, replace=False))
+            ts[r, i, :n] = START + t * 10**9
+            vals[r, i, :n] = np.round(rng.random(n) * 100, 2)
+            counts[r, i] = n
+    o_ts, o_vals, o_counts, o_errs = oracle.merge_batch(ts, vals, counts,
+                                                        out_stride=stride * R)
+    d_ts = torch.from_numpy(ts.reshape(R * nseries, stride)).to("cuda:0")
+    d_vals = torch.from_numpy(vals.reshape(R * nseries, stride)).to("cuda:0")
+    d_counts = torch.from_numpy(counts.reshape(-1).astype(np.int32)).to("cuda:0")
+    out_ts = torch.zeros((nseries, stride * R), dtype=torch.int64, device="cuda:0")
+    out_vals = torch.zeros((nseries, stride * R), dtype=torch.float64, device="cuda:0")
+    out_counts = torch.empty(nseries, dtype=torch.int32, device="cuda:0")
+    out_errs = torch.empty(nseries, dtype=torch.int32, device="cuda:0")
+    engine.merge_batch_dev(d_ts, d_vals, d_counts, R, out_ts, out_vals,
+                           out_counts, out_errs)
+    torch.cuda.synchronize()
+    assert np.all(out_errs.cpu().numpy() == 0)
+    assert np.array_equal(out_counts.cpu().numpy().astype(np.uint32), o_counts)
+    g_ts = out_ts.cpu().numpy()
+    g_vals = out_vals.cpu().numpy()
+    for i in range(nseries):
+        n = int(o_counts[i])
+        assert np.array_equal(g_ts[i, :n], o_ts[i, :n]), i
+        assert np.array_equal(g_vals[i, :n], o_vals[i, :n]), i
+
+
+def test_merge_out_of_order_flag(torch, engine):
+    """A replica with a decreasing timestamp flags err 100 after emitting
+    the prefix — exactly like errOutOfOrderIterator."""
+    START = 1700000000 * 10**9
+    ts = np.zeros((1, 1, 8), np.int64)
+    vals = np.zeros((1, 1, 8), np.float64)
+    ts[0, 0, :3] = [START + 1, START + 3, START + 2]
+    vals[0, 0, :3] = [1.0, 3.0, 2.0]
+    counts = np.array([[3]], np.uint32)
+    d_ts = torch.from_numpy(ts.reshape(1, 8)).to("cuda:0")
+    d_vals = torch.from_numpy(vals.reshape(1, 8)).to("cuda:0")
+    d_counts = torch.from_numpy(counts.reshape(-1).astype(np.int32)).to("cuda:0")
+    out_ts = torch.zeros((1, 8), dtype=torch.int64, device="cuda:0")
+    out_vals = torch.zeros((1, 8), dtype=torch.float64, device="cuda:0")
+    out_counts = torch.empty(1, dtype=torch.int32, device="cuda:0")
+    out_errs = torch.empty(1, dtype=torch.int32, device="cuda:0")
+    engine.merge_batch_dev(d_ts, d_vals, d_counts, 1, out_ts, out_vals,
+                           out_counts, out_errs)
+    torch.cuda.synchronize()
+    assert int(out_errs[0].item()) == 100
+    assert int(out_counts[0].item()) == 2
+    assert out_ts[0, :2].cpu().numpy().tolist() == [START + 1, START + 3]
