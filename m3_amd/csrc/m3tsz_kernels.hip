@@ -1212,7 +1212,7 @@ k_encode_batch(const int64_t* __restrict__ ts, const double* __restrict__ vals,
  *                     M3GPU_SERIES_BUCKET_OVERFLOW.
  */
 
-#define QCAP 64
+#define QCAP 512 /* wave-kernel LDS staging capacity per series */
 #define MAX_AGGS 16
 
 struct RollupPlan {
@@ -1221,6 +1221,12 @@ struct RollupPlan {
     double qs[MAX_AGGS];     /* sorted unique quantiles */
     int32_t nq;
     int32_t naggs;
+    /* Largest bucket size for which the production-default CKMS
+     * (eps=1e-3, insertAndCompressEvery=1024) provably never compresses
+     * for THIS quantile set, so its quantiles are exact order statistics
+     * (the calcQuantiles walk). Computed on the host; buckets beyond it
+     * flag M3GPU_SERIES_BUCKET_OVERFLOW rather than approximate. */
+    int32_t exact_cap;
 };
 
 struct BucketState {
@@ -1292,17 +1298,17 @@ k_rollup_batch(const uint8_t* __restrict__ blobs,
          * form: k_i = max(rank_i, k_{i-1}+1), value = sorted[min(k_i,n)-1];
          * n<=3: sorted[min(int(q*n), n-1)] (quantilesFromBuf) */
         if (metric_type == M3GPU_METRIC_TIMER && plan.nq > 0 && n > 0) {
-            /* rank-select: lane l < n computes the rank of qvals[l].
+            /* rank-select: each lane ranks elements lane, lane+64, ...
              * LDS ops from one wave retire in order; the fences only stop
              * compiler reordering around the cross-lane LDS use. */
             __builtin_amdgcn_wave_barrier();
             __threadfence_block();
-            if (lane < n) {
-                double v = qvals[lane];
+            for (uint32_t e = lane; e < n; e += WAVE) {
+                double v = qvals[e];
                 uint32_t rank = 0;
                 for (uint32_t j = 0; j < n; j++) {
                     double o = qvals[j];
-                    rank += (o < v) || (o == v && j < lane);
+                    rank += (o < v) || (o == v && j < e);
                 }
                 sorted[rank] = v;
             }
@@ -1421,7 +1427,11 @@ k_rollup_batch(const uint8_t* __restrict__ blobs,
             bs.fsum += v;
             bs.fsumsq += v * v;
             if (plan.nq > 0) {
-                if (nq_in_bucket >= QCAP) { err = M3GPU_SERIES_BUCKET_OVERFLOW; break; }
+                if (nq_in_bucket >= (uint32_t)plan.exact_cap ||
+                    nq_in_bucket >= QCAP) {
+                    err = M3GPU_SERIES_BUCKET_OVERFLOW;
+                    break;
+                }
                 if (lane == 0) qvals[nq_in_bucket] = v;
                 nq_in_bucket++;
             }
@@ -1631,7 +1641,8 @@ k_rollup_lane(const uint8_t* __restrict__ blobs,
                         bs.fsum += v;
                         bs.fsumsq += v * v;
                         if (WITH_Q && plan.nq > 0) {
-                            if (nq >= RQCAP_LANE) {
+                            if (nq >= RQCAP_LANE ||
+                                nq >= (uint32_t)plan.exact_cap) {
                                 err = M3GPU_SERIES_BUCKET_OVERFLOW;
                                 running = false;
                             } else {
@@ -1757,6 +1768,7 @@ k_merge(const int64_t* __restrict__ ts, const double* __restrict__ vals,
 /* ============================ C-ABI host layer ============================ */
 
 #include <mutex>
+#include <climits>
 
 static __thread char g_err[512];
 static int g_device = 0;
@@ -1907,6 +1919,31 @@ int m3gpu_rollup_batch_dev(
     nq = m;
     plan.nq = nq;
     for (int i = 0; i < nq; i++) plan.qs[i] = qs[i];
+    /* exact_cap: largest n such that for every numVals v <= n and every
+     * maxRank mr <= v, the compress merge threshold (stream.go:362-385,
+     * int64 truncation exactly as the reference computes it) stays < 2 =
+     * the minimum testVal — i.e. compression NEVER merges and quantiles
+     * are exact order statistics. */
+    plan.exact_cap = QCAP;
+    {
+        double eps2 = 2.0 * 1e-3; /* cm defaults, options.go:30-32 */
+        for (int v = 4; v <= QCAP + 1; v++) /* minSamplesToCompress=3 */ {
+            bool merges = false;
+            for (int mr = 0; mr <= v && !merges; mr++) {
+                long long thr = LLONG_MAX;
+                for (int i = 0; i < nq; i++) {
+                    long long qmin;
+                    if (mr >= (long long)(qs[i] * (double)v))
+                        qmin = (long long)(eps2 * (double)mr / qs[i]);
+                    else
+                        qmin = (long long)(eps2 * (double)(v - mr) / (1.0 - qs[i]));
+                    if (qmin < thr) thr = qmin;
+                }
+                if (nq > 0 && thr >= 2) merges = true;
+            }
+            if (merges) { plan.exact_cap = v - 1; break; }
+        }
+    }
     for (int i = 0; i < naggs; i++) {
         double q = -1;
         switch (agg_types[i]) {

@@ -492,3 +492,68 @@ def test_merge_out_of_order_flag(torch, engine):
     assert int(out_errs[0].item()) == 100
     assert int(out_counts[0].item()) == 2
     assert out_ts[0, :2].cpu().numpy().tolist() == [START + 1, START + 3]
+
+
+def test_rollup_large_exact_windows(torch, engine):
+    """Buckets up to ~300 values stay EXACT (the host-computed no-compression
+    cap for the default timer quantile set is ~498): bit-equal to the oracle
+    full-CKMS rollup via the wave-kernel retry path."""
+    from m3_amd.engine import pack_streams
+    rng = np.random.default_rng(53)
+    nseries, npts = 64, 900
+    window = 600 * 10**9  # 10m windows, 2s cadence -> 300 values/bucket
+    start = (1427162462 * 10**9 // window) * window
+    ts = start + np.arange(npts, dtype=np.int64) * 2 * 10**9
+    ts = np.broadcast_to(ts, (nseries, npts)).copy()
+    vals = np.round(rng.random((nseries, npts)) * 1e4, 4)
+    counts = np.full(nseries, npts, np.uint32)
+    aggs = ["median", "p95", "p99", "count"]
+    base = (ts[:, 0] // window) * window
+    nbuckets = int(((ts[:, -1] - base) // window).max()) + 1
+    o_out, o_wts = oracle.rollup_batch(ts, vals, counts, oracle.METRIC_TIMER,
+                                       window, nbuckets, aggs)
+    streams = [oracle.encode_series(ts[i], vals[i], start_ns=int(ts[i, 0]))
+               for i in range(nseries)]
+    blob, offsets, lens = pack_streams(streams)
+    d_blob = torch.from_numpy(blob).to("cuda:0")
+    d_off = torch.from_numpy(offsets.astype(np.int64)).to("cuda:0")
+    d_lens = torch.from_numpy(lens.astype(np.int32)).to("cuda:0")
+    out = torch.empty((nseries, nbuckets, len(aggs)), dtype=torch.float64,
+                      device="cuda:0")
+    wts = torch.empty((nseries, nbuckets), dtype=torch.int64, device="cuda:0")
+    errs = torch.empty(nseries, dtype=torch.int32, device="cuda:0")
+    engine.rollup_batch_dev(d_blob, d_off, d_lens, engine.METRIC_TIMER,
+                            window, nbuckets, aggs, out, wts, errs)
+    torch.cuda.synchronize()
+    assert np.all(errs.cpu().numpy() == 0)
+    g = out.cpu().numpy()
+    assert np.array_equal(g.view(np.uint64), o_out.view(np.uint64))
+
+
+def test_rollup_beyond_exact_cap_errors(torch, engine):
+    """Buckets beyond the no-compression cap flag BUCKET_OVERFLOW (the
+    engine never approximates where the reference CKMS would compress)."""
+    from m3_amd.engine import pack_streams
+    npts = 600
+    window = 600 * 10**9  # 1s cadence -> 600 values/bucket > cap (~498)
+    start = (1427162462 * 10**9 // window) * window
+    ts = start + np.arange(npts, dtype=np.int64) * 10**9
+    vals = np.arange(npts, dtype=np.float64)
+    streams = [oracle.encode_series(ts, vals, start_ns=int(ts[0]))]
+    blob, offsets, lens = pack_streams(streams)
+    d_blob = torch.from_numpy(blob).to("cuda:0")
+    d_off = torch.from_numpy(offsets.astype(np.int64)).to("cuda:0")
+    d_lens = torch.from_numpy(lens.astype(np.int32)).to("cuda:0")
+    out = torch.empty((1, 1, 2), dtype=torch.float64, device="cuda:0")
+    wts = torch.empty((1, 1), dtype=torch.int64, device="cuda:0")
+    errs = torch.empty(1, dtype=torch.int32, device="cuda:0")
+    import m3_amd.engine as e
+    # bypass check_errors to inspect the flag
+    aggs = np.asarray([e.M3GPU_AGG["p99"], e.M3GPU_AGG["count"]], dtype=np.int32)
+    rc = e.lib().m3gpu_rollup_batch_dev(
+        e._dev_ptr(d_blob), e._dev_ptr(d_off), e._dev_ptr(d_lens), 1, 1, 1,
+        e.METRIC_TIMER, window, 1, aggs.ctypes.data_as(e.P(e.c_i32)), 2,
+        e._dev_ptr(out), e._dev_ptr(wts), e._dev_ptr(errs), e._torch_stream())
+    assert rc == 0
+    torch.cuda.synchronize()
+    assert int(errs[0].item()) == 8  # M3GPU_SERIES_BUCKET_OVERFLOW
