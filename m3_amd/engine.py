@@ -106,7 +106,7 @@ def _raise_series_errors(errs, what):
 
 def pack_streams(streams):
     """Pack a list of encoded streams into (blob, offsets, lens) with the
-    8-byte-aligned zero-padded layout the C ABI requires."""
+    16-byte-aligned zero-padded layout the C ABI requires."""
     n = len(streams)
     lens = np.fromiter((len(s) for s in streams), dtype=np.uint32, count=n)
     padded = (lens.astype(np.uint64) + 15) & ~np.uint64(15)

@@ -3,7 +3,8 @@
 Pure numpy + m3_amd product APIs — no oracle imports. Timestamps follow the
 reference benchmark shape: start 1427162462e9 ns + i*10s (exact 10 s cadence
 => DoD == 0 => 1 timestamp bit/pt after the 2nd point). Values cycle through
-four seeded distributions per series index i (i % 4):
+four seeded distributions, grouped in 64-series runs
+((i >> 6) % 4) so each wavefront's 64 lane parsers stay branch-coherent:
   0: counter-like random-walk 12-digit ints
   1: timer-like 7-digit.6-decimal floats
   2: gauge-like small one-decimal floats
