@@ -545,134 +545,91 @@ struct Decoder {
      * path (bit-identical) for markers, default buckets, float mode,
      * wide sig, unit changes and near-EOS. Returns 1/0/-err like next().
      * Returns -1000 to mean "take the general path" (nothing consumed). */
-    /* XOR field classification from the window w2 (value bits
-     * left-aligned, `avail` valid bits). On success updates float state and
-     * returns bits consumed (INCLUDING the `pre` bits before the XOR
-     * control); 0 = does not fit / take stepwise path. Never touches the
-     * reader. */
-    __device__ __forceinline__ uint32_t classify_xor(uint64_t w2, uint32_t pre,
-                                                     uint32_t avail) {
-        if (avail < pre + 2) return 0;
+    /* XOR field fused from w2 (value bits left-aligned), pre = bits already
+     * classified before the XOR control (ts field + optional mode prefix).
+     * Returns 0 (consumed + state updated) or -1000 (take stepwise path). */
+    __device__ __forceinline__ int fused_xor(uint64_t w2, uint32_t pre) {
         if (!(w2 >> 63)) { /* '0': same value */
+            r.consume(pre + 1);
             prev_xor = 0;
-            return pre + 1;
+            return 0;
         }
         if ((w2 >> 62) == 0x2) { /* '10' contained */
             uint32_t lead = prev_xor ? __builtin_clzll(prev_xor) : 64;
             uint32_t trail = prev_xor ? __builtin_ctzll(prev_xor) : 0;
             uint32_t nmean = 64 - lead - trail;
-            if (pre + 2 + nmean > avail || nmean > 62) return 0;
+            if (pre + 2 + nmean > 64) return -1000;
             uint64_t mb = nmean ? ((w2 << 2) >> (64 - nmean)) : 0;
+            r.consume(pre + 2 + nmean);
             prev_xor = mb << trail;
             prev_float_bits ^= prev_xor;
-            return pre + 2 + nmean;
+            return 0;
         }
-        /* '11' + 6b lead + 6b (nmean-1) + payload */
-        if (avail < pre + 14) return 0;
+        /* '11' + 6b lead + 6b (nmean-1), then payload (may exceed the peek) */
         uint64_t lead = (w2 >> 56) & 0x3f;
         uint64_t nmean = ((w2 >> 50) & 0x3f) + 1;
-        if (pre + 14 + nmean > avail) return 0;
-        uint64_t mb = (w2 << 14) >> (64 - nmean);
+        r.consume(pre + 14);
+        uint64_t mb;
+        int err = r.read_bits((uint32_t)nmean, &mb);
+        if (err) return -err; /* same EOF point as the stepwise reads */
         uint64_t trail = 64 - lead - nmean;
         prev_xor = mb << trail;
         prev_float_bits ^= prev_xor;
-        return pre + 14 + (uint32_t)nmean;
+        return 0;
     }
 
-    /* Classify one complete point (ts field + value field) from the window
-     * w (`avail` valid bits). On success commits the state updates and
-     * returns the consumed bit count; 0 = fall back to the stepwise path
-     * (state untouched). Gating (markers, unit changes, wide fields,
-     * first point) is the caller's job for the FIRST point; for follow-up
-     * points in the same window the same checks re-run here. */
-    __device__ __forceinline__ uint32_t classify_point(uint64_t w, uint32_t avail,
-                                                       int64_t* t, double* v) {
-        if (avail == 0) return 0;
+    __device__ __forceinline__ int next_fused(int64_t* t, double* v) {
+        uint64_t w;
+        if (tu_changed || !have_scheme || prev_time == 0 ||
+            r.peek_bits(64, &w) != 0)
+            return -1000;
         uint32_t c1;
         int64_t dod;
         if (!(w >> 63)) {
             c1 = 1;
             dod = 0;
         } else {
-            if (avail < 11) return 0;          /* cannot rule out a marker */
-            if ((w >> 55) == MARKER_OPCODE) return 0; /* marker: stepwise */
+            if ((w >> 55) == MARKER_OPCODE) return -1000; /* marker */
             uint32_t top4 = (uint32_t)(w >> 60);
             uint32_t L = __builtin_clz(~(top4 << 28)); /* leading ones, 1..4 */
-            if (L >= 4) return 0; /* default bucket: stepwise */
+            if (L >= 4) return -1000; /* default bucket: stepwise */
             uint32_t vb = (L == 1) ? 7 : (L == 2) ? 9 : 12;
             uint32_t ob = L + 1;
-            if (ob + vb > avail) return 0;
             dod = sign_extend((w << ob) >> (64 - vb), vb) * UNIT_NS_D[time_unit];
             c1 = ob + vb;
         }
         uint64_t w2 = w << c1;
-        uint32_t rest = avail - c1;
-        uint32_t c2 = 0;
-        double out_val;
         if (int_optimized && !is_float) { /* int mode: the dominant path */
-            if (rest < 2) return 0;
             if (w2 >> 63) { /* opcodeNoUpdate: sign + sig diff */
-                if (sig > 62 || 2 + sig > rest) return 0;
+                if (sig > 45 || c1 + 2 + sig > 64) return -1000;
                 uint64_t bits = (w2 << 1) >> (63 - sig);
+                r.consume(c1 + 2 + sig);
                 double sgn = -1.0;
                 if ((bits >> sig) == 1) { sgn = 1.0; bits ^= (1ULL << sig); }
                 int_val += sgn * (double)bits;
-                c2 = 2 + sig;
             } else if ((w2 >> 62) == 0x1) { /* repeat */
-                c2 = 2;
+                r.consume(c1 + 2);
             } else {
-                return 0; /* '00': mode/sig/mult update -> stepwise */
+                return -1000; /* '00': mode/sig/mult update -> stepwise */
             }
-            out_val = (mult == 0) ? int_val : int_val / exp10_table(mult);
         } else if (!int_optimized) { /* pure float stream: XOR directly */
-            c2 = classify_xor(w2, 0, rest);
-            if (!c2) return 0;
-            out_val = bits2f(prev_float_bits);
+            int rx = fused_xor(w2, c1);
+            if (rx == -1000) return -1000;
+            if (rx) return rx; /* negative error, same point as stepwise */
         } else {
             /* int-opt float mode: '1' + XOR | '01' repeat | '00..' stepwise */
-            if (rest < 2) return 0;
             if (w2 >> 63) {
-                c2 = classify_xor(w2 << 1, 1, rest);
-                if (!c2) return 0;
+                int rx = fused_xor(w2 << 1, c1 + 1);
+                if (rx == -1000) return -1000;
+                if (rx) return rx;
             } else if ((w2 >> 62) == 0x1) {
-                c2 = 2;
+                r.consume(c1 + 2);
             } else {
-                return 0;
+                return -1000;
             }
-            out_val = bits2f(prev_float_bits);
         }
         prev_time_delta += dod;
         prev_time += prev_time_delta;
-        *t = prev_time;
-        *v = out_val;
-        return c1 + c2;
-    }
-
-    /* Up to TWO points from one 64-bit peek and one consume. Returns like
-     * next(); *got2 set when a second point was decoded into (*t2,*v2). */
-    __device__ __forceinline__ int next2(int64_t* t, double* v,
-                                         int64_t* t2, double* v2, bool* got2) {
-        *got2 = false;
-        if (done) return 0;
-        uint64_t w;
-        if (!tu_changed && have_scheme && prev_time != 0 &&
-            r.peek_bits(64, &w) == 0) {
-            uint32_t c1 = classify_point(w, 64, t, v);
-            if (c1 > 0) {
-                uint32_t c2 = classify_point(w << c1, 64 - c1, t2, v2);
-                r.consume(c1 + c2);
-                *got2 = c2 > 0;
-                return 1;
-            }
-        }
-        /* stepwise path (markers, unit changes, mode/sig updates, default
-         * buckets, first point, near-EOS) */
-        bool first;
-        int err = read_timestamp(&first);
-        if (err) return -err;
-        if (done) return 0;
-        err = first ? read_first_value() : read_next_value();
-        if (err) return -err;
         *t = prev_time;
         if (!int_optimized || is_float) *v = bits2f(prev_float_bits);
         else *v = (mult == 0) ? int_val : int_val / exp10_table(mult);
@@ -682,15 +639,8 @@ struct Decoder {
     /* One point. Returns 1 = value in (*t,*v), 0 = done, -err on error. */
     __device__ __forceinline__ int next(int64_t* t, double* v) {
         if (done) return 0;
-        uint64_t w;
-        if (!tu_changed && have_scheme && prev_time != 0 &&
-            r.peek_bits(64, &w) == 0) {
-            uint32_t c1 = classify_point(w, 64, t, v);
-            if (c1 > 0) {
-                r.consume(c1);
-                return 1;
-            }
-        }
+        int f = next_fused(t, v);
+        if (f != -1000) return f;
         bool first;
         int err = read_timestamp(&first);
         if (err) return -err;
@@ -788,44 +738,24 @@ k_decode_batch(const uint8_t* __restrict__ blobs,
         __builtin_amdgcn_wave_barrier();
     };
 
-    /* two-point decode: next2 classifies up to two points from one peek +
-     * one consume; the second waits in registers and makes the following
-     * iteration parse-free for this lane (the staging protocol still moves
-     * exactly one point per lane per iteration). */
-    bool pend = false;
-    int64_t pend_t = 0;
-    double pend_v = 0;
     while (__any(running)) {
         if (running) {
             int64_t t;
             double v;
-            bool have = false;
-            if (pend) {
-                t = pend_t;
-                v = pend_v;
-                pend = false;
-                have = true;
+            int rstat = d.next(&t, &v);
+            if (rstat <= 0) {
+                err = -rstat;
+                running = false;
+            } else if (cnt >= stride) {
+                err = M3GPU_SERIES_CAPACITY;
+                running = false;
             } else {
-                int rstat = d.next2(&t, &v, &pend_t, &pend_v, &pend);
-                if (rstat <= 0) {
-                    err = -rstat;
-                    running = false;
-                } else {
-                    have = true;
+                if (!discard) {
+                    uint32_t col = (k & (DEC_TILE - 1)) ^ (lane & 7);
+                    ts_tile[lane][col] = t;
+                    val_tile[lane][col] = v;
                 }
-            }
-            if (have) {
-                if (cnt >= stride) {
-                    err = M3GPU_SERIES_CAPACITY;
-                    running = false;
-                } else {
-                    if (!discard) {
-                        uint32_t col = (k & (DEC_TILE - 1)) ^ (lane & 7);
-                        ts_tile[lane][col] = t;
-                        val_tile[lane][col] = v;
-                    }
-                    cnt++;
-                }
+                cnt++;
             }
         }
         k++;
