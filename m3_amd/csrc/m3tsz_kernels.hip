@@ -599,25 +599,12 @@ struct Decoder {
             c1 = ob + vb;
         }
         uint64_t w2 = w << c1;
-        if (int_optimized && !is_float) { /* int mode: the dominant path */
-            if (w2 >> 63) { /* opcodeNoUpdate: sign + sig diff */
-                if (sig > 45 || c1 + 2 + sig > 64) return -1000;
-                uint64_t bits = (w2 << 1) >> (63 - sig);
-                r.consume(c1 + 2 + sig);
-                double sgn = -1.0;
-                if ((bits >> sig) == 1) { sgn = 1.0; bits ^= (1ULL << sig); }
-                int_val += sgn * (double)bits;
-            } else if ((w2 >> 62) == 0x1) { /* repeat */
-                r.consume(c1 + 2);
-            } else {
-                return -1000; /* '00': mode/sig/mult update -> stepwise */
-            }
-        } else if (!int_optimized) { /* pure float stream: XOR directly */
+        if (!int_optimized) { /* pure float stream: XOR field directly */
             int rx = fused_xor(w2, c1);
             if (rx == -1000) return -1000;
             if (rx) return rx; /* negative error, same point as stepwise */
-        } else {
-            /* int-opt float mode: '1' + XOR | '01' repeat | '00..' stepwise */
+        } else if (is_float) {
+            /* float mode: '1' + XOR | '01' repeat | '00...' stepwise */
             if (w2 >> 63) {
                 int rx = fused_xor(w2 << 1, c1 + 1);
                 if (rx == -1000) return -1000;
@@ -627,6 +614,17 @@ struct Decoder {
             } else {
                 return -1000;
             }
+        } else if (w2 >> 63) { /* int mode, opcodeNoUpdate: sign + sig diff */
+            if (sig > 45 || c1 + 2 + sig > 64) return -1000;
+            uint64_t bits = (w2 << 1) >> (63 - sig);
+            r.consume(c1 + 2 + sig);
+            double sgn = -1.0;
+            if ((bits >> sig) == 1) { sgn = 1.0; bits ^= (1ULL << sig); }
+            int_val += sgn * (double)bits;
+        } else if ((w2 >> 62) == 0x1) { /* repeat */
+            r.consume(c1 + 2);
+        } else {
+            return -1000; /* '00': mode/sig/mult update -> stepwise */
         }
         prev_time_delta += dod;
         prev_time += prev_time_delta;
