@@ -712,18 +712,14 @@ k_decode_batch(const uint8_t* __restrict__ blobs,
         const uint32_t p = lane & (DEC_TILE - 1);
         const uint32_t r0 = lane / DEC_TILE;
         const uint32_t pt = base_pt + p;
-        /* non-temporal output stores: the decoded rows are pure streaming
-         * output never re-read here — keeping them out of L2 leaves the
-         * cache to the compressed stream lines the 1280 resident parsers
-         * gather from */
         if (__all(cnt >= base_pt + DEC_TILE)) {
             /* all 64 rows full: unconditional stores, no per-row counts */
             for (uint32_t j = 0; j < DEC_TILE; j++) {
                 uint32_t r = r0 + j * (WAVE / DEC_TILE);
                 uint32_t pc = p ^ (r & 7);
                 uint64_t row = (uint64_t)__shfl((int)series, (int)r);
-                __builtin_nontemporal_store(ts_tile[r][pc], &out_ts[row * stride + pt]);
-                __builtin_nontemporal_store(val_tile[r][pc], &out_vals[row * stride + pt]);
+                out_ts[row * stride + pt] = ts_tile[r][pc];
+                out_vals[row * stride + pt] = val_tile[r][pc];
             }
         } else {
             for (uint32_t j = 0; j < DEC_TILE; j++) {
@@ -732,8 +728,8 @@ k_decode_batch(const uint8_t* __restrict__ blobs,
                 uint32_t c = (uint32_t)__shfl((int)cnt, (int)r);
                 uint64_t row = (uint64_t)__shfl((int)series, (int)r);
                 if (pt < c) {
-                    __builtin_nontemporal_store(ts_tile[r][pc], &out_ts[row * stride + pt]);
-                    __builtin_nontemporal_store(val_tile[r][pc], &out_vals[row * stride + pt]);
+                    out_ts[row * stride + pt] = ts_tile[r][pc];
+                    out_vals[row * stride + pt] = val_tile[r][pc];
                 }
             }
         }
