@@ -426,6 +426,7 @@ static int enc_maybe_write_time_unit_change(m3tsz_enc* e, m3_ostream* os, uint8_
 /* timestamp_encoder.go:205-246 */
 static int enc_write_dod_unchanged(m3tsz_enc* e, m3_ostream* os,
                                    int64_t prev_delta, int64_t cur_delta, uint8_t time_unit) {
+    (void)e;
     if (!unit_is_valid(time_unit)) return -M3_ERR_NO_SCHEME; /* timeUnit.Value() error */
     int64_t u = UNIT_NS[time_unit];
     int64_t dod = (cur_delta - prev_delta) / u; /* ToNormalizedDuration, time.go:55-57 */
@@ -1131,7 +1132,7 @@ int64_t oracle_decode_series(
     int64_t n = 0;
     int64_t ann_used = 0;
     for (;;) {
-        int64_t t; double v; uint8_t u;
+        int64_t t = 0; double v = 0; uint8_t u = 0;
         int r = dec_next(&d, &t, &v, &u);
         if (r == 0) break;
         if (r < 0) return r;
