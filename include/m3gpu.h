@@ -49,7 +49,9 @@ enum {
     M3GPU_SERIES_ANNOTATION = 5,   /* bad annotation */
     M3GPU_SERIES_CAPACITY = 6,     /* stride/out buffer too small */
     M3GPU_SERIES_UNSORTED = 7,     /* rollup input crosses buckets backwards */
-    M3GPU_SERIES_BUCKET_OVERFLOW = 8, /* >64 values in one rollup bucket */
+    M3GPU_SERIES_BUCKET_OVERFLOW = 8, /* rollup bucket beyond CKMS capacity
+                                         (sample list > 3072; unreachable for
+                                         the reference-default eps/cadence) */
 };
 
 /* metric types for the rollup entry (aggregator/aggregation) */

@@ -1761,6 +1761,402 @@ k_merge(const int64_t* __restrict__ ts, const double* __restrict__ vals,
     out_errs[series] = err;
 }
 
+
+/* ===================== full-CKMS deep-bucket rollup =====================
+ * Third tier of the timer-quantile chain (after the per-lane <=16 staging
+ * and the wave-kernel exact-order-statistics cap): the reference CKMS
+ * stream WITH compression (quantile/cm/stream.go:77-429), ported from the
+ * oracle's restatement onto LDS arrays. One WAVE per workgroup, one series
+ * per workgroup (rare retry path; correctness, not throughput):
+ *  - samples double-buffer (value,numRanks,delta) replaces the linked list;
+ *    the insert cursor walk becomes a two-pointer merge (equivalent:
+ *    incoming v inserted before the first list element with value >= v,
+ *    delta = that element's numRanks+delta-1; the tail appends delta 0);
+ *  - compress walks backward tracking the surviving next element; removed
+ *    samples are compacted after the walk (same removal order);
+ *  - in the AddBatch(1)/Flush usage the compress cursor is always nil on
+ *    insert entry, so the compValue/compressMinRank-in-insert branch of the
+ *    reference is dead (stream.go:284-287,303) and omitted;
+ *  - buffers bufLess/bufMore are value MULTISETS here (the reference's heap
+ *    order only permutes ties, which insert() cannot distinguish);
+ *  - capacity overflow (sample list or buffer > CKMS_CAP) flags
+ *    M3GPU_SERIES_BUCKET_OVERFLOW — never an approximation. */
+#define CKMS_CAP 3072
+#define CKMS_BUF 2080 /* bufMore peak = carried bufLess (<=1024) + 1024 new */
+
+struct CkmsLds {
+    double* val[2];
+    int32_t* nr[2];
+    int32_t* dl[2];
+    double* less;
+    double* more;
+};
+
+struct CkmsState {
+    int cur;           /* active buffer index */
+    int32_t len;       /* samples */
+    int32_t nless, nmore;
+    int64_t num_values;
+    int32_t counter;   /* insertAndCompressCounter */
+    int err;
+};
+
+__device__ __forceinline__ void ckms_reset(CkmsState& st) {
+    st.cur = 0;
+    st.len = 0;
+    st.nless = 0;
+    st.nmore = 0;
+    st.num_values = 0;
+    st.counter = 0;
+}
+
+/* wave-parallel ascending sort of buf[0..n) into tmp[0..n) (rank-select) */
+__device__ __forceinline__ void ckms_sort(const double* buf, double* tmp,
+                                          int32_t n, uint32_t lane) {
+    __builtin_amdgcn_wave_barrier();
+    __threadfence_block();
+    for (int32_t e = lane; e < n; e += WAVE) {
+        double v = buf[e];
+        int32_t rank = 0;
+        for (int32_t j = 0; j < n; j++) {
+            double o = buf[j];
+            rank += (o < v) || (o == v && j < e);
+        }
+        tmp[rank] = v;
+    }
+    __builtin_amdgcn_wave_barrier();
+    __threadfence_block();
+}
+
+/* stream.go:362-377 threshold(rank), exact int64 truncation */
+__device__ __forceinline__ int64_t ckms_threshold(const RollupPlan& plan,
+                                                  int64_t num_vals, int64_t rank) {
+    int64_t min_val = INT64_MAX;
+    double eps = 2.0 * 1e-3;
+    for (int i = 0; i < plan.nq; i++) {
+        int64_t qmin;
+        if (rank >= (int64_t)(plan.qs[i] * (double)num_vals))
+            qmin = (int64_t)(eps * (double)rank / plan.qs[i]);
+        else
+            qmin = (int64_t)(eps * (double)(num_vals - rank) / (1.0 - plan.qs[i]));
+        if (qmin < min_val) min_val = qmin;
+    }
+    return min_val;
+}
+
+/* insert sorted batch (stream.go:280-331) + compress (:333-401); lane 0
+ * does the sequential merge/walk, the wave sorts. `sorted` is dedicated LDS scratch. */
+__device__ void ckms_insert_compress(CkmsLds& L, CkmsState& st,
+                                     const RollupPlan& plan, double* sorted,
+                                     uint32_t lane) {
+    int32_t m = st.nmore;
+    ckms_sort(L.more, sorted, m, lane);
+    if (st.len + m > CKMS_CAP) { /* uniform: all lanes see the same state */
+        st.err = M3GPU_SERIES_BUCKET_OVERFLOW;
+        return;
+    }
+    if (lane == 0) {
+        int src = st.cur, dst = st.cur ^ 1;
+        int32_t n = st.len;
+        {
+            /* two-pointer merge == the reference's cursor walk */
+            int32_t i = 0, j = 0, o = 0;
+            while (i < n) {
+                if (j < m && sorted[j] <= L.val[src][i]) {
+                    L.val[dst][o] = sorted[j];
+                    L.nr[dst][o] = 1;
+                    L.dl[dst][o] = L.nr[src][i] + L.dl[src][i] - 1;
+                    o++; j++;
+                } else {
+                    L.val[dst][o] = L.val[src][i];
+                    L.nr[dst][o] = L.nr[src][i];
+                    L.dl[dst][o] = L.dl[src][i];
+                    o++; i++;
+                }
+            }
+            while (j < m) { /* PushBack tail, delta 0 (:316-328) */
+                L.val[dst][o] = sorted[j];
+                L.nr[dst][o] = 1;
+                L.dl[dst][o] = 0;
+                o++; j++;
+            }
+            st.cur = dst;
+            st.len = o;
+            st.num_values += m;
+            /* compress (:333-401); entry cursor nil in this usage */
+            int32_t len = st.len;
+            if (len >= 3) {
+                int b = st.cur;
+                int64_t cmr = st.num_values - 1 - L.nr[b][len - 2];
+                int32_t nxt = len - 2; /* surviving element after curr */
+                for (int32_t c = len - 3; c >= 1; c--) {
+                    int64_t max_rank = cmr + L.nr[b][c] + L.dl[b][c];
+                    int64_t thr = ckms_threshold(plan, st.num_values, max_rank);
+                    cmr -= L.nr[b][c];
+                    int64_t test_val = L.nr[b][c] + L.nr[b][nxt] + L.dl[b][nxt];
+                    if (test_val <= thr) {
+                        L.nr[b][nxt] += L.nr[b][c];
+                        L.nr[b][c] = -1; /* removed mark */
+                    } else {
+                        nxt = c;
+                    }
+                }
+                /* compact, preserving order */
+                int32_t o2 = 0;
+                for (int32_t c = 0; c < len; c++) {
+                    if (L.nr[b][c] >= 0) {
+                        if (o2 != c) {
+                            L.val[b][o2] = L.val[b][c];
+                            L.nr[b][o2] = L.nr[b][c];
+                            L.dl[b][o2] = L.dl[b][c];
+                        }
+                        o2++;
+                    }
+                }
+                st.len = o2;
+            }
+        }
+    }
+    __builtin_amdgcn_wave_barrier();
+    __threadfence_block();
+    /* uniform parts every lane must perform itself: the buffer swap
+     * (resetInsertCursor, stream.go:425-429) swaps POINTERS held per lane */
+    double* t = L.less; L.less = L.more; L.more = t;
+    st.nmore = st.nless;
+    st.nless = 0;
+    /* lane-0-computed scalars: broadcast via readfirstlane */
+    st.cur = __builtin_amdgcn_readfirstlane(st.cur);
+    st.len = __builtin_amdgcn_readfirstlane(st.len);
+    st.num_values = ((int64_t)(uint32_t)__builtin_amdgcn_readfirstlane(
+                         (uint32_t)(st.num_values >> 32)) << 32) |
+                    (uint32_t)__builtin_amdgcn_readfirstlane((uint32_t)st.num_values);
+}
+
+/* stream.go:77-116 AddBatch(values[0..1)) — one value at a time, exactly
+ * like Timer.AddBatch -> stream per-value adds in the rollup loop. */
+__device__ __forceinline__ void ckms_add(CkmsLds& L, CkmsState& st,
+                                         const RollupPlan& plan, double v,
+                                         double* sorted, uint32_t lane) {
+    if (st.err) return;
+    if (st.len == 0 && st.nmore == 0 && st.nless == 0 && st.num_values == 0) {
+        /* first value becomes the first sample (:88-97) */
+        if (lane == 0) {
+            L.val[st.cur][0] = v;
+            L.nr[st.cur][0] = 1;
+            L.dl[st.cur][0] = 0;
+        }
+        st.len = 1;
+        st.num_values = 1;
+        __builtin_amdgcn_wave_barrier();
+        __threadfence_block();
+        return;
+    }
+    double insert_point = L.val[st.cur][0]; /* insertCursor == head here */
+    if (st.nless >= CKMS_BUF || st.nmore >= CKMS_BUF) {
+        st.err = M3GPU_SERIES_BUCKET_OVERFLOW;
+        return;
+    }
+    if (v < insert_point) {
+        if (lane == 0) L.less[st.nless] = v;
+        st.nless++;
+    } else {
+        if (lane == 0) L.more[st.nmore] = v;
+        st.nmore++;
+    }
+    __builtin_amdgcn_wave_barrier();
+    __threadfence_block();
+    if (st.counter == 1024) { /* insertAndCompressEvery, checked BEFORE ++ */
+        ckms_insert_compress(L, st, plan, sorted, lane);
+        st.counter = 0;
+    }
+    st.counter++;
+}
+
+/* stream.go:123-137 Flush */
+__device__ __forceinline__ void ckms_flush(CkmsLds& L, CkmsState& st,
+                                           const RollupPlan& plan,
+                                           double* sorted, uint32_t lane) {
+    while (!st.err && (st.nless > 0 || st.nmore > 0)) {
+        if (st.nmore == 0) { /* resetInsertCursor: swap */
+            double* t = L.less; L.less = L.more; L.more = t;
+            st.nmore = st.nless;
+            st.nless = 0;
+        }
+        ckms_insert_compress(L, st, plan, sorted, lane);
+    }
+}
+
+/* stream.go:231-277 calcQuantiles (+ :210-229 quantilesFromBuf), filling
+ * computed[0..nq) for the plan's sorted unique quantile list. computed[]
+ * must be zeroed per bucket first: the reference leaves un-emitted
+ * quantiles at the zero value (the post-loop condition can fail for the
+ * top quantiles at certain numValues — real behavior, preserved). Lane 0
+ * only; thr_rank/thr_thresh are LDS scratch (>= plan.nq entries). */
+__device__ void ckms_calc_quantiles(const CkmsLds& L, const CkmsState& st,
+                                    const RollupPlan& plan, double* computed,
+                                    int64_t* thr_rank, int64_t* thr_thresh) {
+    if (plan.nq == 0 || st.num_values == 0) return;
+    const int b = st.cur;
+    if (st.num_values <= 3) { /* quantilesFromBuf over the sample list */
+        for (int i = 0; i < plan.nq; i++) {
+            int32_t idx = (int32_t)(plan.qs[i] * (double)st.len);
+            if (idx >= st.len) idx = st.len - 1;
+            computed[i] = L.val[b][idx];
+        }
+        return;
+    }
+    for (int i = 0; i < plan.nq; i++) {
+        int64_t rank = (int64_t)ceil(plan.qs[i] * (double)st.num_values);
+        thr_rank[i] = rank;
+        thr_thresh[i] =
+            (int64_t)ceil((double)ckms_threshold(plan, st.num_values, rank) / 2.0);
+    }
+    int64_t min_rank = 0, max_rank = 0;
+    int idx = 0;
+    int32_t prev = 0;
+    for (int32_t c = 0; c < st.len && idx < plan.nq; c++) {
+        max_rank = min_rank + L.nr[b][c] + L.dl[b][c];
+        if (max_rank > thr_rank[idx] + thr_thresh[idx] || min_rank > thr_rank[idx]) {
+            computed[idx] = L.val[b][prev];
+            idx++;
+        }
+        min_rank += L.nr[b][c];
+        prev = c;
+    }
+    for (int i = idx; i < plan.nq; i++) { /* post-loop (:268-276), note >= */
+        if (max_rank >= thr_rank[i] + thr_thresh[i] || min_rank > thr_rank[i])
+            computed[i] = L.val[b][prev];
+    }
+}
+
+/* ------------- the deep-bucket rollup kernel (third tier) -------------
+ * One series per 64-thread workgroup (one wave), full 150 KB dynamic LDS
+ * for the CKMS state. Only TIMER series with quantile aggs can reach this
+ * tier (nothing else sets BUCKET_OVERFLOW). Decode/bucket walk identical
+ * to k_rollup_batch. */
+#define CKMS_BLOCK 64
+
+__global__ void __launch_bounds__(CKMS_BLOCK)
+k_rollup_ckms(const uint8_t* __restrict__ blobs,
+              const uint64_t* __restrict__ offsets,
+              const uint32_t* __restrict__ lens,
+              const int32_t* __restrict__ select, uint32_t nseries,
+              int int_optimized, uint8_t default_unit,
+              int64_t window_ns, uint32_t nbuckets, RollupPlan plan,
+              double* __restrict__ out, int64_t* __restrict__ out_window_ts,
+              int32_t* __restrict__ out_errs) {
+    extern __shared__ uint8_t lds_raw[];
+    const uint32_t lane = threadIdx.x % WAVE;
+    const uint32_t slot = blockIdx.x;
+    if (slot >= nseries) return;
+    const uint32_t series = __builtin_amdgcn_readfirstlane(
+        select ? (uint32_t)select[slot] : slot);
+
+    /* carve the dynamic LDS (layout mirrored by rollup_ckms_lds_bytes) */
+    uint8_t* p = lds_raw;
+    CkmsLds L;
+    L.val[0] = (double*)p; p += CKMS_CAP * 8;
+    L.val[1] = (double*)p; p += CKMS_CAP * 8;
+    L.nr[0] = (int32_t*)p; p += CKMS_CAP * 4;
+    L.nr[1] = (int32_t*)p; p += CKMS_CAP * 4;
+    L.dl[0] = (int32_t*)p; p += CKMS_CAP * 4;
+    L.dl[1] = (int32_t*)p; p += CKMS_CAP * 4;
+    L.less = (double*)p; p += CKMS_BUF * 8;
+    L.more = (double*)p; p += CKMS_BUF * 8;
+    double* sorted = (double*)p; p += CKMS_BUF * 8;
+    double* computed = (double*)p; p += MAX_AGGS * 8;
+    int64_t* thr_rank = (int64_t*)p; p += MAX_AGGS * 8;
+    int64_t* thr_thresh = (int64_t*)p; p += MAX_AGGS * 8;
+
+    Decoder d;
+    d.init(blobs, offsets[series], lens[series], int_optimized != 0, default_unit);
+
+    double* out_row = out + (uint64_t)series * nbuckets * plan.naggs;
+    int64_t* wts_row = out_window_ts ? out_window_ts + (uint64_t)series * nbuckets : nullptr;
+
+    BucketState bs;
+    bs.reset();
+    CkmsState st;
+    ckms_reset(st);
+    st.err = 0;
+    int64_t base = 0;
+    int64_t cur_bucket = -1;
+    int err = 0;
+
+    auto emit_bucket = [&](int64_t bk) {
+        if (bk < 0 || bk >= (int64_t)nbuckets) return;
+        if (wts_row && lane == 0) wts_row[bk] = base + (bk + 1) * window_ns;
+        ckms_flush(L, st, plan, sorted, lane);
+        if (st.err) return;
+        if (lane == 0) {
+            for (int i = 0; i < plan.nq; i++) computed[i] = 0.0;
+            ckms_calc_quantiles(L, st, plan, computed, thr_rank, thr_thresh);
+        }
+        __builtin_amdgcn_wave_barrier();
+        __threadfence_block();
+        if (lane < (uint32_t)plan.naggs) { /* timer.go:131-153 ValueOf */
+            int32_t t = plan.agg_types[lane];
+            int8_t qi = plan.qidx[lane];
+            double r = 0;
+            int cb = st.cur;
+            if (qi >= 0) {
+                r = st.len ? computed[qi] : 0.0; /* Quantile(q): computed */
+            } else {
+                switch (t) {
+                case M3GPU_AGG_MIN: r = st.len ? L.val[cb][0] : 0.0; break;
+                case M3GPU_AGG_MAX: r = st.len ? L.val[cb][st.len - 1] : 0.0; break;
+                case M3GPU_AGG_MEAN: r = bs.count ? bs.fsum / (double)bs.count : 0.0; break;
+                case M3GPU_AGG_COUNT: r = (double)bs.count; break;
+                case M3GPU_AGG_SUM: r = bs.fsum; break;
+                case M3GPU_AGG_SUMSQ: r = bs.fsumsq; break;
+                case M3GPU_AGG_STDEV: r = m3_stdev(bs.count, bs.fsumsq, bs.fsum); break;
+                default: r = 0; break;
+                }
+            }
+            out_row[(uint64_t)bk * plan.naggs + lane] = r;
+        }
+    };
+
+    for (;;) {
+        int64_t t;
+        double v;
+        int rstat = d.next(&t, &v);
+        if (rstat < 0) { err = -rstat; break; }
+        bool have = rstat == 1;
+        int64_t b = -1;
+        if (have) {
+            if (cur_bucket < 0) base = (t / window_ns) * window_ns;
+            if (t < base) { err = M3GPU_SERIES_UNSORTED; break; }
+            b = (t - base) / window_ns;
+            if (b >= (int64_t)nbuckets) { err = M3GPU_SERIES_CAPACITY; break; }
+            if (b < cur_bucket) { err = M3GPU_SERIES_UNSORTED; break; }
+        }
+        if (!have || b != cur_bucket) {
+            if (cur_bucket >= 0) {
+                emit_bucket(cur_bucket);
+                if (st.err) { err = st.err; break; }
+            }
+            int64_t stop = have ? b : (int64_t)nbuckets;
+            for (int64_t eb = (cur_bucket < 0 ? 0 : cur_bucket + 1); eb < stop; eb++) {
+                bs.reset();
+                ckms_reset(st);
+                emit_bucket(eb);
+            }
+            if (!have) break;
+            bs.reset();
+            ckms_reset(st);
+            cur_bucket = b;
+        }
+        /* timer.go:56-75 AddBatch, one value at a time */
+        bs.count++;
+        bs.fsum += v;
+        bs.fsumsq += v * v;
+        ckms_add(L, st, plan, v, sorted, lane);
+        if (st.err) { err = st.err; break; }
+    }
+    if (lane == 0) out_errs[series] = err;
+}
+
 } // namespace m3
 
 /* ============================ C-ABI host layer ============================ */
@@ -2017,6 +2413,48 @@ int m3gpu_rollup_batch_dev(
             (void)hipFree(d_sel);
             free(h_sel);
             if (e2 != hipSuccess) { free(h_errs); return set_hip_err("rollup retry", e2); }
+
+            /* Third tier: buckets deeper than the exact-order-statistics cap
+             * run the real compressed CKMS (k_rollup_ckms, one series per
+             * workgroup, ~122 KB dynamic LDS). Rescan the error flags the
+             * wave kernel just rewrote. */
+            ce = hipMemcpy(h_errs, d_out_errs, nseries * sizeof(int32_t),
+                           hipMemcpyDeviceToHost);
+            if (ce != hipSuccess) { free(h_errs); return set_hip_err("errs copy2", ce); }
+            uint32_t ndeep = 0;
+            for (uint32_t i = 0; i < nseries; i++)
+                if (h_errs[i] == M3GPU_SERIES_BUCKET_OVERFLOW) ndeep++;
+            if (ndeep) {
+                size_t lds_bytes = (size_t)CKMS_CAP * 32 /* val+nr+dl x2 */
+                                 + (size_t)CKMS_BUF * 24 /* less+more+sorted */
+                                 + (size_t)MAX_AGGS * 24; /* computed+thr */
+                hipError_t e3 = hipFuncSetAttribute(
+                    reinterpret_cast<const void*>(m3::k_rollup_ckms),
+                    hipFuncAttributeMaxDynamicSharedMemorySize, (int)lds_bytes);
+                int32_t* h_sel2 = (int32_t*)malloc(ndeep * sizeof(int32_t));
+                uint32_t k2 = 0;
+                for (uint32_t i = 0; i < nseries; i++)
+                    if (h_errs[i] == M3GPU_SERIES_BUCKET_OVERFLOW)
+                        h_sel2[k2++] = (int32_t)i;
+                int32_t* d_sel2 = nullptr;
+                if (e3 == hipSuccess) e3 = hipMalloc(&d_sel2, ndeep * sizeof(int32_t));
+                if (e3 == hipSuccess)
+                    e3 = hipMemcpy(d_sel2, h_sel2, ndeep * sizeof(int32_t),
+                                   hipMemcpyHostToDevice);
+                if (e3 == hipSuccess) {
+                    hipLaunchKernelGGL(m3::k_rollup_ckms, dim3(ndeep),
+                                       dim3(CKMS_BLOCK), lds_bytes, s,
+                                       d_blobs, d_offsets, d_lens, d_sel2, ndeep,
+                                       int_optimized, default_unit,
+                                       window_ns, nbuckets, plan,
+                                       d_out, d_out_window_ts, d_out_errs);
+                    e3 = hipGetLastError();
+                    if (e3 == hipSuccess) e3 = hipStreamSynchronize(s);
+                }
+                (void)hipFree(d_sel2);
+                free(h_sel2);
+                if (e3 != hipSuccess) { free(h_errs); return set_hip_err("rollup ckms", e3); }
+            }
         }
         free(h_errs);
     }

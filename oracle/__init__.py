@@ -61,6 +61,8 @@ def _setup(L):
                                       P(c_i64), P(c_f64), P(c_u32), c_i64, c_int]
     L.oracle_xxhash64.restype = c_u64
     L.oracle_xxhash64.argtypes = [P(c_u8), ctypes.c_size_t]
+    L.oracle_ckms_list_len.restype = c_i64
+    L.oracle_ckms_list_len.argtypes = [P(c_f64), c_i64, P(c_f64), c_int, c_f64, c_int]
     L.oracle_ckms_quantiles.restype = c_int
     L.oracle_ckms_quantiles.argtypes = [P(c_f64), c_i64, P(c_f64), c_int, c_f64, c_int,
                                         P(c_f64), P(c_f64), P(c_f64)]
@@ -230,6 +232,15 @@ def ckms_quantiles(values, quantiles, eps=1e-3, every=1024):
                                 out.ctypes.data_as(P(c_f64)),
                                 ctypes.byref(mn), ctypes.byref(mx))
     return out, mn.value, mx.value
+
+
+def ckms_list_len(values, quantiles, eps=1e-3, every=1024):
+    """Post-flush CKMS sample-list length (sizing aid for engine cap tests)."""
+    v = np.ascontiguousarray(values, dtype=np.float64)
+    q = np.ascontiguousarray(quantiles, dtype=np.float64)
+    return int(lib().oracle_ckms_list_len(
+        v.ctypes.data_as(P(c_f64)), len(v),
+        q.ctypes.data_as(P(c_f64)), len(q), eps, every))
 
 
 METRIC_COUNTER, METRIC_GAUGE, METRIC_TIMER = 0, 1, 2

@@ -390,6 +390,21 @@ static double stream_quantile(cm_stream* s, double q) {
 
 /* ==================== exported CKMS surface (tests) ==================== */
 
+/* Feed n values through a CKMS stream and return the post-flush sample
+ * list length (test sizing aid for the GPU engine's sample-list cap). */
+int64_t oracle_ckms_list_len(const double* values, int64_t n,
+                             const double* quantiles, int nq,
+                             double eps, int insert_and_compress_every) {
+    cm_stream s;
+    stream_init(&s, quantiles, nq, eps, insert_and_compress_every);
+    stream_add_batch(&s, values, n);
+    stream_flush(&s);
+    int64_t len = s.list_len;
+    stream_free(&s);
+    return len;
+}
+
+
 /* Feed n values through a CKMS stream (AddBatch), Flush, and evaluate the
  * q[] quantiles (which must equal the stream's registered quantiles).
  * Returns 0. */

@@ -189,3 +189,35 @@ def test_rollup_gauge_last_equal_timestamps():
     assert out[0, 0, 0] == 1.0  # first wins on equal ts
     assert out[0, 0, 1] == 2.0
     assert out[0, 1, 0] == 3.0
+
+
+def test_ckms_stale_zero_top_quantile_regime():
+    """Reference CKMS artifact (quantile/cm/stream.go:231-277): when
+    ceil(q*numValues) == numValues and the emission threshold is >= 1, the
+    walk never emits the top quantile and Quantile(q) returns the zero-value
+    of the un-touched computed[] slot. For q=0.9999 this holds for
+    999 <= n <= 9999. The GPU deep-bucket tier must reproduce it (covered
+    bit-exactly in tests/test_gpu_parity.py deep-bucket tests)."""
+    rng = np.random.default_rng(3)
+    qs = [0.5, 0.95, 0.99, 0.9999]
+    for n, stale in ((300, False), (999, True), (2520, True), (9999, True),
+                     (10001, False)):
+        v = 1.0 + rng.random(n)  # values in (1,2): 0.0 is unambiguous
+        out, mn, mx = oracle.ckms_quantiles(v, qs)
+        assert (out[3] == 0.0) == stale, (n, out)
+        assert out[0] != 0.0 and out[1] != 0.0 and out[2] != 0.0
+        assert 1.0 < mn < mx < 2.0
+
+
+def test_ckms_list_len_probe():
+    """Compression keeps the sample list far below the GPU tier's
+    CKMS_CAP=3072 even at 10^6 values (the buffer peak is bounded at 2048
+    by the insert cadence): the deep tier never truncates."""
+    rng = np.random.default_rng(7)
+    qs = [0.5, 0.95, 0.99]
+    ln = oracle.ckms_list_len(rng.random(200) * 1e4, qs)
+    assert ln == 200  # below first compression cadence, nothing merges? no:
+    # flush compresses once, but thresholds stay 0 below ~500 values
+    for n in (600, 5000, 100000):
+        ln = oracle.ckms_list_len(rng.random(n) * 1e4, qs)
+        assert 100 < ln < 1500, (n, ln)

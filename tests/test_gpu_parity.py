@@ -530,30 +530,118 @@ def test_rollup_large_exact_windows(torch, engine):
     assert np.array_equal(g.view(np.uint64), o_out.view(np.uint64))
 
 
-def test_rollup_beyond_exact_cap_errors(torch, engine):
-    """Buckets beyond the no-compression cap flag BUCKET_OVERFLOW (the
-    engine never approximates where the reference CKMS would compress)."""
+def test_rollup_deep_buckets_full_ckms(torch, engine):
+    """Buckets beyond the exact-order-statistics cap run the real compressed
+    CKMS on the GPU (k_rollup_ckms, third retry tier): bit-equal to the
+    oracle's quantile/cm/stream.go:77-429 restatement, including compression
+    merges, the 1024-value insert cadence, and the reference's stale-zero
+    top-quantile regime (un-emitted computed[] stays 0.0 for some
+    numValues). Depths straddle the insertAndCompressEvery boundary."""
     from m3_amd.engine import pack_streams
-    npts = 600
-    window = 600 * 10**9  # 1s cadence -> 600 values/bucket > cap (~498)
+    rng = np.random.default_rng(61)
+    window_s = 5040
+    window = window_s * 10**9
+    cadences = [8, 5, 4, 2, 1]     # depths 630, 1008, 1260, 2520, 5040
+    aggs = ["median", "p95", "p99", "p9999", "min", "max", "mean", "count",
+            "sum", "stdev"]
     start = (1427162462 * 10**9 // window) * window
-    ts = start + np.arange(npts, dtype=np.int64) * 10**9
-    vals = np.arange(npts, dtype=np.float64)
-    streams = [oracle.encode_series(ts, vals, start_ns=int(ts[0]))]
+    series = []
+    for cad in cadences:
+        depth = window_s // cad
+        npts = 2 * depth           # two full buckets
+        t = start + np.arange(npts, dtype=np.int64) * cad * 10**9
+        for dist in range(5):
+            if dist == 0:
+                v = np.round(rng.random(npts) * 1e4, 4)
+            elif dist == 1:
+                v = np.round(np.sort(rng.random(npts)) * 1e3, 4)
+            elif dist == 2:
+                v = np.round(np.sort(rng.random(npts))[::-1] * 1e3, 4).copy()
+            elif dist == 3:        # heavy ties
+                v = np.round(rng.random(npts) * 10, 1)
+            else:                  # constant
+                v = np.full(npts, 42.5)
+            series.append((t, v))
+    nseries = len(series)
+    width = max(len(t) for t, _ in series)
+    ts = np.zeros((nseries, width), np.int64)
+    vals = np.zeros((nseries, width), np.float64)
+    counts = np.zeros(nseries, np.uint32)
+    for i, (t, v) in enumerate(series):
+        ts[i, :len(t)] = t
+        vals[i, :len(t)] = v
+        counts[i] = len(t)
+    nbuckets = 2
+    o_out, o_wts = oracle.rollup_batch(ts, vals, counts, oracle.METRIC_TIMER,
+                                       window, nbuckets, aggs)
+    streams = [oracle.encode_series(ts[i, :counts[i]], vals[i, :counts[i]],
+                                    start_ns=int(ts[i, 0]))
+               for i in range(nseries)]
     blob, offsets, lens = pack_streams(streams)
     d_blob = torch.from_numpy(blob).to("cuda:0")
     d_off = torch.from_numpy(offsets.astype(np.int64)).to("cuda:0")
     d_lens = torch.from_numpy(lens.astype(np.int32)).to("cuda:0")
-    out = torch.empty((1, 1, 2), dtype=torch.float64, device="cuda:0")
-    wts = torch.empty((1, 1), dtype=torch.int64, device="cuda:0")
-    errs = torch.empty(1, dtype=torch.int32, device="cuda:0")
-    import m3_amd.engine as e
-    # bypass check_errors to inspect the flag
-    aggs = np.asarray([e.M3GPU_AGG["p99"], e.M3GPU_AGG["count"]], dtype=np.int32)
-    rc = e.lib().m3gpu_rollup_batch_dev(
-        e._dev_ptr(d_blob), e._dev_ptr(d_off), e._dev_ptr(d_lens), 1, 1, 1,
-        e.METRIC_TIMER, window, 1, aggs.ctypes.data_as(e.P(e.c_i32)), 2,
-        e._dev_ptr(out), e._dev_ptr(wts), e._dev_ptr(errs), e._torch_stream())
-    assert rc == 0
+    out = torch.empty((nseries, nbuckets, len(aggs)), dtype=torch.float64,
+                      device="cuda:0")
+    wts = torch.empty((nseries, nbuckets), dtype=torch.int64, device="cuda:0")
+    errs = torch.empty(nseries, dtype=torch.int32, device="cuda:0")
+    engine.rollup_batch_dev(d_blob, d_off, d_lens, engine.METRIC_TIMER,
+                            window, nbuckets, aggs, out, wts, errs)
     torch.cuda.synchronize()
-    assert int(errs[0].item()) == 8  # M3GPU_SERIES_BUCKET_OVERFLOW
+    assert np.all(errs.cpu().numpy() == 0)
+    g = out.cpu().numpy()
+    gw = wts.cpu().numpy()
+    assert np.array_equal(gw, o_wts)
+    for i in range(nseries):
+        assert np.array_equal(g[i].view(np.uint64), o_out[i].view(np.uint64)), \
+            (i, g[i], o_out[i])
+
+
+def test_rollup_deep_buckets_mixed_with_shallow(torch, engine):
+    """A batch mixing shallow (per-lane tier), medium (wave tier) and deep
+    (CKMS tier) series resolves each series on the right tier with every
+    result bit-equal to the oracle."""
+    from m3_amd.engine import pack_streams
+    rng = np.random.default_rng(67)
+    window_s = 1200
+    window = window_s * 10**9
+    start = (1427162462 * 10**9 // window) * window
+    aggs = ["p95", "p99", "max", "count", "sum"]
+    cadences = [120, 10, 1]        # depths 10 (lane), 120 (wave), 1200 (ckms)
+    series = []
+    for cad in cadences:
+        for _ in range(6):
+            npts = 3 * (window_s // cad)
+            t = start + np.arange(npts, dtype=np.int64) * cad * 10**9
+            v = np.round(rng.random(npts) * 1e3, 3)
+            series.append((t, v))
+    nseries = len(series)
+    width = max(len(t) for t, _ in series)
+    ts = np.zeros((nseries, width), np.int64)
+    vals = np.zeros((nseries, width), np.float64)
+    counts = np.zeros(nseries, np.uint32)
+    for i, (t, v) in enumerate(series):
+        ts[i, :len(t)] = t
+        vals[i, :len(t)] = v
+        counts[i] = len(t)
+    nbuckets = 3
+    o_out, o_wts = oracle.rollup_batch(ts, vals, counts, oracle.METRIC_TIMER,
+                                       window, nbuckets, aggs)
+    streams = [oracle.encode_series(ts[i, :counts[i]], vals[i, :counts[i]],
+                                    start_ns=int(ts[i, 0]))
+               for i in range(nseries)]
+    blob, offsets, lens = pack_streams(streams)
+    d_blob = torch.from_numpy(blob).to("cuda:0")
+    d_off = torch.from_numpy(offsets.astype(np.int64)).to("cuda:0")
+    d_lens = torch.from_numpy(lens.astype(np.int32)).to("cuda:0")
+    out = torch.empty((nseries, nbuckets, len(aggs)), dtype=torch.float64,
+                      device="cuda:0")
+    wts = torch.empty((nseries, nbuckets), dtype=torch.int64, device="cuda:0")
+    errs = torch.empty(nseries, dtype=torch.int32, device="cuda:0")
+    engine.rollup_batch_dev(d_blob, d_off, d_lens, engine.METRIC_TIMER,
+                            window, nbuckets, aggs, out, wts, errs)
+    torch.cuda.synchronize()
+    assert np.all(errs.cpu().numpy() == 0)
+    g = out.cpu().numpy()
+    assert np.array_equal(g.view(np.uint64), o_out.view(np.uint64))
+    assert np.array_equal(wts.cpu().numpy(), o_wts)
