@@ -167,6 +167,47 @@ int m3gpu_rollup_batch(
     const int32_t* agg_types, int naggs,
     double* out, int64_t* out_window_ts, int32_t* out_errs);
 
+/* ======================= fileset volume reader =======================
+ * Native reader for the reference's dbnode fileset volumes (persist/fs
+ * read.go:145-457 + msgpack/decoder.go + digest): open a volume from a
+ * shard directory (reference naming fileset-<blockStartNs>-<volume>-
+ * <suffix>.db, legacy no-volume names for volume 0), validate the
+ * checkpoint, all five file digests, every index-entry checksum (V3) and
+ * every data-block checksum, then expose the entries sorted by data
+ * offset ascending and repack the blocks into the decode-batch blob
+ * layout above. Pure host code (usable without a GPU). */
+
+/* Returns a handle >= 0, or a negative M3GPU_FS_ERR_* code. */
+int m3gpu_fileset_open(const char* shard_dir, int64_t block_start_ns,
+                       int volume_index);
+int m3gpu_fileset_close(int handle);
+const char* m3gpu_fileset_last_error(void);
+int m3gpu_fileset_info(int handle, int64_t* block_start, int64_t* block_size,
+                       int64_t* entries, int64_t* major_version,
+                       int64_t* minor_version, int* volume_index,
+                       int64_t* bloom_m, int64_t* bloom_k,
+                       int64_t* summaries);
+int m3gpu_fileset_entry(int handle, int64_t i, int64_t* size, int64_t* offset,
+                        int64_t* data_checksum, const uint8_t** id,
+                        int64_t* id_len, const uint8_t** tags,
+                        int64_t* tags_len);
+int64_t m3gpu_fileset_packed_size(int handle);
+int m3gpu_fileset_pack(int handle, uint8_t* blob, uint64_t blob_cap,
+                       uint64_t* offsets, uint32_t* lens);
+
+enum {
+    M3GPU_FS_ERR_IO = -101,
+    M3GPU_FS_ERR_CHECKPOINT = -102,
+    M3GPU_FS_ERR_DIGEST = -103,
+    M3GPU_FS_ERR_MSGPACK = -104,
+    M3GPU_FS_ERR_SCHEMA = -105,
+    M3GPU_FS_ERR_ENTRY_CHECKSUM = -106,
+    M3GPU_FS_ERR_DATA_CHECKSUM = -107,
+    M3GPU_FS_ERR_BOUNDS = -108,
+    M3GPU_FS_ERR_BADHANDLE = -109,
+    M3GPU_FS_ERR_CAPACITY = -110,
+};
+
 #ifdef __cplusplus
 }
 #endif

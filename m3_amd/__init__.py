@@ -14,6 +14,9 @@ the oracle/ package is test infrastructure only and is never imported by
 this product path.
 """
 from .engine import (  # noqa: F401
+    FS_ERRORS,
+    FilesetVolume,
+    fileset_ingest_dev,
     M3GPU_AGG,
     METRIC_COUNTER,
     METRIC_GAUGE,

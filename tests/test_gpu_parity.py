@@ -645,3 +645,43 @@ def test_rollup_deep_buckets_mixed_with_shallow(torch, engine):
     g = out.cpu().numpy()
     assert np.array_equal(g.view(np.uint64), o_out.view(np.uint64))
     assert np.array_equal(wts.cpu().numpy(), o_wts)
+
+
+def test_fileset_ingest_end_to_end(torch, engine, tmp_path):
+    """§8f row 1 end-to-end: oracle-written fileset volume -> native reader
+    (validate digests/checksums, repack) -> HIP batch decode on the GPU ->
+    bit-exact vs the original points."""
+    from oracle import fileset_writer as fsw
+    rng = np.random.default_rng(41)
+    BLOCK_START = 1427162400 * 10**9
+    series = []
+    raw = {}
+    for i in range(200):
+        npts = int(rng.integers(2, 700))
+        ts = BLOCK_START + np.cumsum(rng.integers(1, 30, npts)) * 10**9
+        kind = i % 3
+        if kind == 0:
+            vals = np.round(rng.random(npts) * 100, 2)
+        elif kind == 1:
+            vals = np.cumsum(rng.integers(-1000, 1000, npts)).astype(float)
+        else:
+            vals = rng.random(npts) * 1e6
+        blob = oracle.encode_series(ts, vals, start_ns=int(ts[0]))
+        sid = f"ns.metric.{i:05d}".encode()
+        series.append((sid, blob, None))
+        raw[sid] = (ts, vals)
+    fsw.write_volume(str(tmp_path), BLOCK_START, series)
+    ids, d_ts, d_vals, d_counts, d_errs = engine.fileset_ingest_dev(
+        torch, str(tmp_path), BLOCK_START)
+    torch.cuda.synchronize()
+    assert np.all(d_errs.cpu().numpy() == 0)
+    g_ts = d_ts.cpu().numpy()
+    g_vals = d_vals.cpu().numpy()
+    g_counts = d_counts.cpu().numpy()
+    assert len(ids) == len(series)
+    for i, sid in enumerate(ids):
+        ts, vals = raw[sid]
+        assert g_counts[i] == len(ts), sid
+        assert np.array_equal(g_ts[i, :len(ts)], ts), sid
+        assert np.array_equal(g_vals[i, :len(ts)].view(np.uint64),
+                              vals.view(np.uint64)), sid
