@@ -208,6 +208,36 @@ enum {
     M3GPU_FS_ERR_CAPACITY = -110,
 };
 
+/* ======================== commit log reader =========================
+ * Native reader for the reference's commit log files (persist/fs/
+ * commitlog: 12-byte chunk headers with adler32 size/data checksums,
+ * uvarint-framed msgpack LogInfo/LogEntry records, nested LogMetadata on
+ * each series' first entry). Series come back in first-seen order with
+ * datapoints in log order — the bootstrap-from-commitlog input to the GPU
+ * batch encoder. Pure host code. */
+int m3gpu_commitlog_open(const char* path); /* handle >= 0 or error */
+int m3gpu_commitlog_close(int handle);
+const char* m3gpu_commitlog_last_error(void);
+int m3gpu_commitlog_info(int handle, int64_t* index, int64_t* num_entries,
+                         int64_t* num_series);
+int m3gpu_commitlog_series(int handle, int64_t i, uint64_t* unique_index,
+                           const uint8_t** id, int64_t* id_len,
+                           const uint8_t** ns, int64_t* ns_len,
+                           uint32_t* shard, const uint8_t** tags,
+                           int64_t* tags_len, int64_t* num_points,
+                           int64_t* num_annotations);
+int m3gpu_commitlog_series_points(int handle, int64_t i, int64_t* ts,
+                                  double* vals, uint8_t* units);
+int m3gpu_commitlog_series_annotation(int handle, int64_t i, int64_t j,
+                                      int64_t* point_index,
+                                      const uint8_t** bytes, int64_t* len);
+
+enum {
+    M3GPU_CL_ERR_CHUNK_CHECKSUM = -111,
+    M3GPU_CL_ERR_MISSING_METADATA = -112,
+    M3GPU_CL_ERR_TRUNCATED = -113,
+};
+
 #ifdef __cplusplus
 }
 #endif

@@ -14,6 +14,9 @@ the oracle/ package is test infrastructure only and is never imported by
 this product path.
 """
 from .engine import (  # noqa: F401
+    CL_ERRORS,
+    CommitLog,
+    commitlog_bootstrap_dev,
     FS_ERRORS,
     FilesetVolume,
     fileset_ingest_dev,

@@ -454,3 +454,146 @@ def fileset_ingest_dev(torch, shard_dir, block_start_ns, volume_index=0,
     decode_batch_dev(d_blob, d_off, d_lens, d_ts, d_vals, d_counts, d_errs,
                      int_optimized=int_optimized, default_unit=default_unit)
     return ids, d_ts, d_vals, d_counts, d_errs
+
+
+# =========================== commit log reader ===========================
+
+CL_ERRORS = dict(FS_ERRORS)
+CL_ERRORS.update({-111: "chunk_checksum", -112: "missing_metadata",
+                  -113: "truncated"})
+
+
+def _cl_configure(L):
+    if getattr(L, "_cl_configured", False):
+        return
+    L.m3gpu_commitlog_open.restype = c_int
+    L.m3gpu_commitlog_open.argtypes = [ctypes.c_char_p]
+    L.m3gpu_commitlog_close.restype = c_int
+    L.m3gpu_commitlog_close.argtypes = [c_int]
+    L.m3gpu_commitlog_last_error.restype = ctypes.c_char_p
+    L.m3gpu_commitlog_info.restype = c_int
+    L.m3gpu_commitlog_info.argtypes = [c_int, P(c_i64), P(c_i64), P(c_i64)]
+    L.m3gpu_commitlog_series.restype = c_int
+    L.m3gpu_commitlog_series.argtypes = [c_int, c_i64, P(c_u64), P(c_vp),
+                                         P(c_i64), P(c_vp), P(c_i64),
+                                         P(c_u32), P(c_vp), P(c_i64),
+                                         P(c_i64), P(c_i64)]
+    L.m3gpu_commitlog_series_points.restype = c_int
+    L.m3gpu_commitlog_series_points.argtypes = [c_int, c_i64, P(c_i64),
+                                                P(c_f64), P(c_u8)]
+    L.m3gpu_commitlog_series_annotation.restype = c_int
+    L.m3gpu_commitlog_series_annotation.argtypes = [c_int, c_i64, c_i64,
+                                                    P(c_i64), P(c_vp),
+                                                    P(c_i64)]
+    L._cl_configured = True
+
+
+class CommitLog:
+    """A parsed, checksum-validated commit log file (commitlog.cpp).
+
+    Replaces the reference's commitlog Reader loop
+    (persist/fs/commitlog/reader.go:161-209) for bootstrap: series in
+    first-seen order, each with its datapoints in log order."""
+
+    def __init__(self, path):
+        L = lib()
+        _cl_configure(L)
+        h = L.m3gpu_commitlog_open(str(path).encode())
+        if h < 0:
+            detail = L.m3gpu_commitlog_last_error().decode()
+            raise M3GpuError(
+                f"commitlog open failed: {CL_ERRORS.get(h, h)} ({detail})")
+        self._h = h
+        self._lib = L
+        idx, ne, ns = c_i64(), c_i64(), c_i64()
+        L.m3gpu_commitlog_info(h, ctypes.byref(idx), ctypes.byref(ne),
+                               ctypes.byref(ns))
+        self.index = idx.value
+        self.num_entries = ne.value
+        self.num_series = ns.value
+
+    def close(self):
+        if self._h >= 0:
+            self._lib.m3gpu_commitlog_close(self._h)
+            self._h = -1
+
+    def __enter__(self):
+        return self
+
+    def __exit__(self, *a):
+        self.close()
+
+    def series(self):
+        """[{id, namespace, shard, tags, unique_index, ts, vals, units,
+        annotations: [(point_idx, bytes)]}] in first-seen order."""
+        out = []
+        for i in range(self.num_series):
+            ui = c_u64()
+            idp, nsp, tgp = c_vp(), c_vp(), c_vp()
+            idl, nsl, tgl = c_i64(), c_i64(), c_i64()
+            shard = c_u32()
+            npts, nant = c_i64(), c_i64()
+            rc = self._lib.m3gpu_commitlog_series(
+                self._h, i, ctypes.byref(ui), ctypes.byref(idp),
+                ctypes.byref(idl), ctypes.byref(nsp), ctypes.byref(nsl),
+                ctypes.byref(shard), ctypes.byref(tgp), ctypes.byref(tgl),
+                ctypes.byref(npts), ctypes.byref(nant))
+            _check(rc, "m3gpu_commitlog_series")
+            n = npts.value
+            ts = np.empty(n, np.int64)
+            vals = np.empty(n, np.float64)
+            units = np.empty(n, np.uint8)
+            rc = self._lib.m3gpu_commitlog_series_points(
+                self._h, i, ts.ctypes.data_as(P(c_i64)),
+                vals.ctypes.data_as(P(c_f64)), units.ctypes.data_as(P(c_u8)))
+            _check(rc, "m3gpu_commitlog_series_points")
+            annotations = []
+            for j in range(nant.value):
+                pi, ap, al = c_i64(), c_vp(), c_i64()
+                self._lib.m3gpu_commitlog_series_annotation(
+                    self._h, i, j, ctypes.byref(pi), ctypes.byref(ap),
+                    ctypes.byref(al))
+                annotations.append((pi.value, ctypes.string_at(ap, al.value)))
+            out.append(dict(
+                unique_index=ui.value,
+                id=ctypes.string_at(idp, idl.value) if idl.value else b"",
+                namespace=ctypes.string_at(nsp, nsl.value) if nsl.value else b"",
+                shard=shard.value,
+                tags=ctypes.string_at(tgp, tgl.value) if tgl.value else b"",
+                ts=ts, vals=vals, units=units, annotations=annotations))
+        return out
+
+
+def commitlog_bootstrap_dev(torch, path, int_optimized=True, device="cuda:0"):
+    """Bootstrap-from-commitlog on the GPU: parse + validate the log
+    (native reader), group per series, batch-encode every series into
+    M3TSZ blocks with the HIP encoder. Returns (series_meta, d_bytes,
+    d_lens, d_errs) where series_meta is the CommitLog.series() list.
+
+    Replaces the reference's commitlog bootstrapper read-and-re-encode
+    (bootstrap/bootstrapper/commitlog + series buffer encoders) with one
+    batched GPU encode."""
+    with CommitLog(path) as cl:
+        meta = cl.series()
+    if not meta:
+        return [], None, None, None
+    n = len(meta)
+    width = max(len(m["ts"]) for m in meta)
+    ts = np.zeros((n, width), np.int64)
+    vals = np.zeros((n, width), np.float64)
+    counts = np.zeros(n, np.int32)
+    for i, m in enumerate(meta):
+        k = len(m["ts"])
+        ts[i, :k] = m["ts"]
+        vals[i, :k] = m["vals"]
+        counts[i] = k
+    d_ts = torch.from_numpy(ts).to(device)
+    d_vals = torch.from_numpy(vals).to(device)
+    d_counts = torch.from_numpy(counts).to(device)
+    out_stride = (24 * width + 32 + 15) & ~15
+    d_bytes = torch.zeros((n, out_stride), dtype=torch.uint8, device=device)
+    d_lens = torch.empty(n, dtype=torch.int32, device=device)
+    d_errs = torch.empty(n, dtype=torch.int32, device=device)
+    encode_batch_dev(d_ts, d_vals, d_counts, d_bytes, d_lens, d_errs,
+                     int_optimized=int_optimized)
+    return meta, d_bytes, d_lens, d_errs

@@ -700,3 +700,41 @@ def test_fileset_ingest_end_to_end(torch, engine, tmp_path):
         # timestamps always survive exactly; values match input except
         # where the codec canonicalized (still equal as oracle output)
         assert np.array_equal(o_ts[i, :n], ts), sid
+
+
+def test_commitlog_bootstrap_end_to_end(torch, engine, tmp_path):
+    """Commitlog half of §8f row 1: oracle-written commit log -> native
+    reader (chunk validation, metadata registration) -> grouped series ->
+    HIP batch encode -> byte-exact vs oracle encode of the same points."""
+    from oracle import commitlog_writer as clw
+    rng = np.random.default_rng(71)
+    START = 1427162462 * 10**9
+    nseries, npts = 150, 200
+    per = {}
+    entries = []
+    for i in range(nseries):
+        ts = START + np.cumsum(rng.integers(1, 30, npts)) * 10**9
+        vals = np.round(rng.random(npts) * 1e4, 3)
+        per[i] = (ts, vals)
+    for t in range(npts):          # interleaved, like live ingest
+        for i in range(nseries):
+            entries.append((i, f"boot.{i:04d}".encode(), i % 8,
+                            int(per[i][0][t]), float(per[i][1][t]), 4,
+                            None, None))
+    path = tmp_path / "commitlog-0-0.db"
+    clw.write_commitlog(str(path), entries)
+    meta, d_bytes, d_lens, d_errs = engine.commitlog_bootstrap_dev(
+        torch, str(path))
+    torch.cuda.synchronize()
+    assert len(meta) == nseries
+    assert np.all(d_errs.cpu().numpy() == 0)
+    g_lens = d_lens.cpu().numpy()
+    g_rows = d_bytes.cpu().numpy()
+    ts = np.stack([per[i][0] for i in range(nseries)])
+    vals = np.stack([per[i][1] for i in range(nseries)])
+    counts = np.full(nseries, npts, np.uint32)
+    o_rows, o_lens = oracle.encode_batch(ts, vals, counts, int_optimized=True)
+    assert np.array_equal(g_lens.astype(np.uint32), o_lens)
+    for i in range(nseries):
+        assert meta[i]["id"] == f"boot.{i:04d}".encode()
+        assert bytes(g_rows[i, :g_lens[i]]) == bytes(o_rows[i, :o_lens[i]]), i
