@@ -55,6 +55,10 @@ def parse_args():
     p.add_argument("--skip-cpu-baseline", action="store_true")
     p.add_argument("--no-sort", action="store_true",
                    help="disable length-sorted wave scheduling")
+    p.add_argument("--layout", choices=["index", "sorted"], default="sorted",
+                   help="physical blob layout: 'sorted' repacks streams in "
+                        "schedule order before the timed region (free at "
+                        "pack time in production; k_regather here)")
     p.add_argument("--data", choices=["synthetic", "production"],
                    default="synthetic",
                    help="production = the 10 real M3TSZ streams embedded in "
@@ -124,6 +128,20 @@ def main():
     d_perm = None
     if not args.no_sort:
         d_perm = torch.argsort(d_lens).to(torch.int32)
+    if args.layout == "sorted" and d_perm is not None:
+        # physical layout pass: repack the blob in schedule order so each
+        # wave's 64 streams are HBM neighbors; scheduling perm becomes
+        # identity and is dropped
+        new_lens = d_lens[d_perm.long()]
+        aligned = ((new_lens.to(torch.int64) + 15) // 16) * 16
+        new_off = torch.zeros_like(aligned)
+        torch.cumsum(aligned[:-1], 0, out=new_off[1:])
+        d_blob2 = torch.empty_like(d_blob)
+        engine.regather_dev(d_blob, d_offsets, d_lens, d_perm,
+                            new_off.to(torch.int64), d_blob2)
+        torch.cuda.synchronize()
+        d_blob, d_offsets, d_lens = d_blob2, new_off, new_lens.contiguous()
+        d_perm = None
 
     parse_only = args.mode == "parse-only"
 

@@ -167,6 +167,15 @@ int m3gpu_rollup_batch(
     const int32_t* agg_types, int naggs,
     double* out, int64_t* out_window_ts, int32_t* out_errs);
 
+/* Layout pass: physically reorder packed streams so dst series i holds
+ * src series perm[i] (dst_offsets precomputed from lens[perm], 16B
+ * aligned). Lays a batch out in wave-schedule order for L2 locality. */
+int m3gpu_regather_dev(
+    const uint8_t* d_src, const uint64_t* d_src_offsets,
+    const uint32_t* d_lens, const int32_t* d_perm,
+    const uint64_t* d_dst_offsets, uint32_t nseries, uint8_t* d_dst,
+    void* hip_stream);
+
 /* ======================= fileset volume reader =======================
  * Native reader for the reference's dbnode fileset volumes (persist/fs
  * read.go:145-457 + msgpack/decoder.go + digest): open a volume from a

@@ -1438,6 +1438,28 @@ k_rollup_batch(const uint8_t* __restrict__ blobs,
     if (lane == 0) out_errs[series] = err;
 }
 
+/* ===================== blob regather (layout pass) ===================== */
+/* Physically reorders packed streams: dst series i <- src series perm[i].
+ * Used to lay the blob out in wave-schedule order (length-sorted), so the
+ * 64 lanes of each decoding wavefront read NEIGHBORING memory instead of
+ * scattered streams — the L2-locality companion to the perm scheduling.
+ * One series per wavefront, u64 copies (offsets are 16B aligned). */
+__global__ void __launch_bounds__(BLOCK_THREADS)
+k_regather(const uint8_t* __restrict__ src, const uint64_t* __restrict__ src_offsets,
+           const uint32_t* __restrict__ lens, const int32_t* __restrict__ perm,
+           const uint64_t* __restrict__ dst_offsets, uint32_t nseries,
+           uint8_t* __restrict__ dst) {
+    const uint32_t wave = __builtin_amdgcn_readfirstlane(threadIdx.x / WAVE);
+    const uint32_t lane = threadIdx.x % WAVE;
+    const uint32_t i = blockIdx.x * WAVES_PER_BLOCK + wave;
+    if (i >= nseries) return;
+    const uint32_t s = (uint32_t)perm[i];
+    const uint64_t* sp = (const uint64_t*)(src + src_offsets[s]);
+    uint64_t* dp = (uint64_t*)(dst + dst_offsets[i]);
+    uint32_t nwords = (lens[s] + 7) / 8;
+    for (uint32_t w = lane; w < nwords; w += WAVE) dp[w] = sp[w];
+}
+
 /* ===================== stream compaction kernel ===================== */
 /* Packs strided encoder output rows into the tight 8B-aligned blob layout
  * (m3gpu.h): one series per wavefront, u64 copies, coalesced within a row. */
@@ -2252,6 +2274,21 @@ int m3gpu_compact_dev(
     hipLaunchKernelGGL(m3::k_compact, dim3(grid_for(nseries)),
                        dim3(BLOCK_THREADS), 0, s,
                        d_src, src_stride, d_lens, d_dst_offsets, nseries, d_dst);
+    HIP_TRY(hipGetLastError());
+    return M3GPU_OK;
+}
+
+int m3gpu_regather_dev(
+    const uint8_t* d_src, const uint64_t* d_src_offsets,
+    const uint32_t* d_lens, const int32_t* d_perm,
+    const uint64_t* d_dst_offsets, uint32_t nseries, uint8_t* d_dst,
+    void* hip_stream) {
+    if (!nseries) return M3GPU_OK;
+    hipStream_t s = (hipStream_t)hip_stream;
+    hipLaunchKernelGGL(m3::k_regather, dim3(grid_for(nseries)),
+                       dim3(BLOCK_THREADS), 0, s,
+                       d_src, d_src_offsets, d_lens, d_perm, d_dst_offsets,
+                       nseries, d_dst);
     HIP_TRY(hipGetLastError());
     return M3GPU_OK;
 }

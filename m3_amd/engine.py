@@ -241,6 +241,23 @@ def compact_dev(d_src, src_stride, d_lens, d_dst_offsets, d_dst):
     _check(rc, "m3gpu_compact_dev")
 
 
+def regather_dev(d_blob, d_offsets, d_lens, d_perm, d_dst_offsets, d_dst):
+    """Layout pass: physically reorder packed streams into perm order
+    (k_regather) so each decoding wavefront's 64 streams are neighbors in
+    HBM. Pair with decode_batch_dev(d_perm=None) on the result."""
+    L = lib()
+    if not hasattr(L.m3gpu_regather_dev, "_configured"):
+        L.m3gpu_regather_dev.restype = c_int
+        L.m3gpu_regather_dev.argtypes = [c_vp, c_vp, c_vp, c_vp, c_vp, c_u32,
+                                         c_vp, c_vp]
+        L.m3gpu_regather_dev._configured = True
+    rc = L.m3gpu_regather_dev(
+        _dev_ptr(d_blob), _dev_ptr(d_offsets), _dev_ptr(d_lens),
+        _dev_ptr(d_perm), _dev_ptr(d_dst_offsets), d_lens.numel(),
+        _dev_ptr(d_dst), _torch_stream())
+    _check(rc, "m3gpu_regather_dev")
+
+
 def rollup_batch_dev(d_blob, d_offsets, d_lens, metric_type, window_ns,
                      nbuckets, agg_types, out, out_window_ts, out_errs,
                      int_optimized=True, default_unit=1):
