@@ -141,6 +141,10 @@ def main():
                             new_off.to(torch.int64), d_blob2)
         torch.cuda.synchronize()
         d_blob, d_offsets, d_lens = d_blob2, new_off, new_lens.contiguous()
+        # series identity = pack order: output row i now holds the stream
+        # that was at perm[i] before the repack
+        if expected_counts is not None:
+            expected_counts = expected_counts[d_perm.long()]
         d_perm = None
 
     parse_only = args.mode == "parse-only"
