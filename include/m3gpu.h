@@ -247,6 +247,32 @@ enum {
     M3GPU_CL_ERR_TRUNCATED = -113,
 };
 
+/* ================= unaggregated metric wire parser =================
+ * Batch parser for the m3aggregator ingest wire format
+ * (metrics/encoding/protobuf): zigzag-varint size-prefixed metricpb
+ * MetricWithMetadatas messages. Untimed counter/batch-timer/gauge and
+ * timed unions are parsed; metadatas/policies kept as opaque bytes.
+ * Pure host code; grouped values feed the GPU encode/rollup path. */
+int m3gpu_unagg_parse(const uint8_t* buf, uint64_t len); /* handle or <0 */
+int m3gpu_unagg_close(int handle);
+const char* m3gpu_unagg_last_error(void);
+int64_t m3gpu_unagg_count(int handle);
+int m3gpu_unagg_metric(int handle, int64_t i, int32_t* union_type,
+                       int32_t* metric_type, const uint8_t** id,
+                       int64_t* id_len, int64_t* num_values,
+                       int64_t* counter_value, int64_t* time_nanos,
+                       const uint8_t** annotation, int64_t* annotation_len,
+                       const uint8_t** metadatas, int64_t* metadatas_len);
+int m3gpu_unagg_values(int handle, int64_t i, double* out);
+
+enum {
+    M3GPU_UA_ERR_TRUNCATED = -121,
+    M3GPU_UA_ERR_PROTO = -122,
+    M3GPU_UA_ERR_TYPE = -123,
+    M3GPU_UA_ERR_BADHANDLE = -124,
+    M3GPU_UA_ERR_SIZE = -125,
+};
+
 #ifdef __cplusplus
 }
 #endif

@@ -614,3 +614,70 @@ def commitlog_bootstrap_dev(torch, path, int_optimized=True, device="cuda:0"):
     encode_batch_dev(d_ts, d_vals, d_counts, d_bytes, d_lens, d_errs,
                      int_optimized=int_optimized)
     return meta, d_bytes, d_lens, d_errs
+
+
+# ==================== unaggregated metric wire parser ====================
+
+UA_ERRORS = {-121: "truncated", -122: "proto", -123: "type",
+             -124: "badhandle", -125: "size"}
+UA_TYPES = {1: "counter", 2: "batch_timer", 3: "gauge",
+            5: "timed_with_metadata", 6: "timed_with_metadatas",
+            7: "timed_with_storage_policy"}
+
+
+def parse_unaggregated(buf):
+    """Parse an unaggregated-metric wire buffer (m3aggregator ingest,
+    metrics/encoding/protobuf framing) with the native parser (unagg.cpp).
+    Returns a list of dicts: {type, metric_type, id, values, counter_value,
+    time_nanos, annotation, metadatas}."""
+    L = lib()
+    if not getattr(L, "_ua_configured", False):
+        L.m3gpu_unagg_parse.restype = c_int
+        L.m3gpu_unagg_parse.argtypes = [P(c_u8), c_u64]
+        L.m3gpu_unagg_close.restype = c_int
+        L.m3gpu_unagg_close.argtypes = [c_int]
+        L.m3gpu_unagg_last_error.restype = ctypes.c_char_p
+        L.m3gpu_unagg_count.restype = c_i64
+        L.m3gpu_unagg_count.argtypes = [c_int]
+        L.m3gpu_unagg_metric.restype = c_int
+        L.m3gpu_unagg_metric.argtypes = [c_int, c_i64, P(c_i32), P(c_i32),
+                                         P(c_vp), P(c_i64), P(c_i64),
+                                         P(c_i64), P(c_i64), P(c_vp),
+                                         P(c_i64), P(c_vp), P(c_i64)]
+        L.m3gpu_unagg_values.restype = c_int
+        L.m3gpu_unagg_values.argtypes = [c_int, c_i64, P(c_f64)]
+        L._ua_configured = True
+    b = np.frombuffer(bytes(buf), dtype=np.uint8)
+    h = L.m3gpu_unagg_parse(b.ctypes.data_as(P(c_u8)), b.nbytes)
+    if h < 0:
+        detail = L.m3gpu_unagg_last_error().decode()
+        raise M3GpuError(
+            f"unagg parse failed: {UA_ERRORS.get(h, h)} ({detail})")
+    try:
+        out = []
+        for i in range(L.m3gpu_unagg_count(h)):
+            ut, mt = c_i32(), c_i32()
+            idp, anp, mdp = c_vp(), c_vp(), c_vp()
+            idl, nv, cv, tn, anl, mdl = (c_i64() for _ in range(6))
+            rc = L.m3gpu_unagg_metric(
+                h, i, ctypes.byref(ut), ctypes.byref(mt), ctypes.byref(idp),
+                ctypes.byref(idl), ctypes.byref(nv), ctypes.byref(cv),
+                ctypes.byref(tn), ctypes.byref(anp), ctypes.byref(anl),
+                ctypes.byref(mdp), ctypes.byref(mdl))
+            _check(rc, "m3gpu_unagg_metric")
+            vals = np.empty(nv.value, np.float64)
+            if nv.value:
+                L.m3gpu_unagg_values(h, i, vals.ctypes.data_as(P(c_f64)))
+            out.append(dict(
+                type=UA_TYPES.get(ut.value, ut.value),
+                metric_type=mt.value,
+                id=ctypes.string_at(idp, idl.value) if idl.value else b"",
+                values=vals,
+                counter_value=cv.value,
+                time_nanos=tn.value,
+                annotation=ctypes.string_at(anp, anl.value) if anl.value else b"",
+                metadatas=ctypes.string_at(mdp, mdl.value) if mdl.value else b"",
+            ))
+        return out
+    finally:
+        L.m3gpu_unagg_close(h)
