@@ -101,6 +101,48 @@ def soak(budget_s=120):
         for i in range(nseries):
             assert bytes(g_rows[i, :g_lens[i]]) == bytes(o2_rows[i, :o2_lens[i]]), \
                 (trial, i)
+        # randomized rollup parity (all three dispatch tiers incl. the
+        # compressed-CKMS deep-bucket kernel) every few trials
+        if trial % 3 == 0:
+            metric = int(rng.integers(0, 3))
+            window_s = int(rng.choice([60, 600, 1200, 3600]))
+            window = window_s * 10**9
+            start = (START // window) * window
+            rn = int(rng.integers(8, 80))
+            cad = int(rng.choice([1, 2, 5, 10, 60]))
+            depth = window_s // cad
+            rnp = min(int(rng.integers(1, 4)) * depth, 6000)
+            rts = start + np.arange(rnp, dtype=np.int64) * cad * 10**9
+            rts = np.broadcast_to(rts, (rn, rnp)).copy()
+            rvals = np.round(rng.random((rn, rnp)) * 1e4, 3)
+            if metric == 0:
+                rvals = np.trunc(rvals)
+            rcounts = np.full(rn, rnp, np.uint32)
+            aggs = (["median", "p95", "p99", "p9999", "min", "max", "count",
+                     "sum", "stdev"] if metric == 2 else
+                    ["min", "max", "mean", "count", "sum", "sumsq", "stdev"])
+            nbuckets = (rnp * cad + window_s - 1) // window_s
+            o_out, o_wts = oracle.rollup_batch(rts, rvals, rcounts, metric,
+                                               window, nbuckets, aggs)
+            streams_r = [oracle.encode_series(rts[i], rvals[i],
+                                              start_ns=int(rts[i, 0]))
+                         for i in range(rn)]
+            rblob, roff, rlens = pack_streams(streams_r)
+            d_rblob = torch.from_numpy(rblob).to("cuda:0")
+            d_roff = torch.from_numpy(roff.astype(np.int64)).to("cuda:0")
+            d_rlens = torch.from_numpy(rlens.astype(np.int32)).to("cuda:0")
+            r_out = torch.empty((rn, nbuckets, len(aggs)),
+                                dtype=torch.float64, device="cuda:0")
+            r_wts = torch.empty((rn, nbuckets), dtype=torch.int64,
+                                device="cuda:0")
+            r_errs = torch.empty(rn, dtype=torch.int32, device="cuda:0")
+            engine.rollup_batch_dev(d_rblob, d_roff, d_rlens, metric, window,
+                                    nbuckets, aggs, r_out, r_wts, r_errs)
+            torch.cuda.synchronize()
+            assert np.all(r_errs.cpu().numpy() == 0), (trial, "rollup errs")
+            assert np.array_equal(r_out.cpu().numpy().view(np.uint64),
+                                  o_out.view(np.uint64)), (trial, "rollup",
+                                                           metric, depth)
         if trial % 10 == 0:
             print(f"  soak trial {trial} ok ({time.time()-t0:.0f}s)", flush=True)
     print(f"SOAK PASSED: {trial} random batches bit-exact", flush=True)
