@@ -15,6 +15,11 @@ from hypothesis import given, settings, strategies as st
 
 import oracle
 
+
+def _engine_available():
+    from m3_amd import engine as _e
+    return _e.engine_available()
+
 START = 1427162462 * 10**9
 
 finite_floats = st.floats(allow_nan=False, allow_infinity=False, width=64)
@@ -156,3 +161,43 @@ def test_extreme_magnitude_drift_is_reference_faithful():
     enc3 = oracle.encode_series(dec2["ts"], dec2["vals"], start_ns=START - 1)
     dec3 = oracle.decode_series(enc3)
     assert list(dec3["vals"]) == list(dec2["vals"])  # now a fixpoint
+
+
+@pytest.mark.skipif(not _engine_available(), reason="libm3gpu.so not built")
+@settings(max_examples=120, deadline=None)
+@given(st.binary(min_size=0, max_size=400), st.integers(0, 2))
+def test_ingest_parsers_never_crash(data, which):
+    """The native ingestion parsers (fileset index/commitlog/unagg) must
+    error cleanly (never crash or hang) on arbitrary bytes."""
+    from m3_amd import engine as _e
+    if which == 0:
+        try:
+            _e.parse_unaggregated(data)
+        except _e.M3GpuError:
+            pass
+    elif which == 1:
+        import tempfile, os
+        with tempfile.TemporaryDirectory() as td:
+            p = os.path.join(td, "commitlog-0-0.db")
+            open(p, "wb").write(data)
+            try:
+                _e.CommitLog(p).close()
+            except _e.M3GpuError:
+                pass
+    else:
+        # fileset: drop random bytes into every file of a valid volume
+        import tempfile, os
+        import oracle as _o
+        from oracle import fileset_writer as _fw
+        with tempfile.TemporaryDirectory() as td:
+            ts = 1427162400 * 10**9 + np.arange(5) * 10**9
+            blob = _o.encode_series(ts, np.ones(5), start_ns=int(ts[0]))
+            paths = _fw.write_volume(td, 1427162400 * 10**9,
+                                     [(b"s", blob, None)])
+            if data:
+                victim = paths[len(data) % len(paths)]
+                open(victim, "wb").write(data)
+            try:
+                _e.FilesetVolume(td, 1427162400 * 10**9).close()
+            except _e.M3GpuError:
+                pass
