@@ -676,16 +676,10 @@ k_decode_batch(const uint8_t* __restrict__ blobs,
                uint32_t nseries, int int_optimized, uint8_t default_unit,
                int64_t* __restrict__ out_ts, double* __restrict__ out_vals,
                uint32_t* __restrict__ out_counts, int32_t* __restrict__ out_errs,
-               uint32_t stride, int xcd_swz) {
+               uint32_t stride) {
     const uint32_t lane = threadIdx.x & (WAVE - 1);
     const uint32_t wave = __builtin_amdgcn_readfirstlane(threadIdx.x >> 6);
-    /* optional XCD-contiguous block swizzle: block b runs on XCD b%8
-     * (MI355X_MICROARCH "Workgroup dispatch"); remapping gives each XCD's
-     * L2 a CONTIGUOUS slice of the (schedule-sorted) blob instead of an
-     * 8-way stride. Exact bijection: dispatch pads the grid to 8k blocks. */
-    uint32_t blk = blockIdx.x;
-    if (xcd_swz) blk = (blockIdx.x & 7) * (gridDim.x >> 3) + (blockIdx.x >> 3);
-    const uint32_t s_base = blk * BLOCK_THREADS + wave * WAVE;
+    const uint32_t s_base = blockIdx.x * BLOCK_THREADS + wave * WAVE;
     const uint32_t slot = s_base + lane;
     /* optional scheduling permutation (e.g. length-sorted): waves then get
      * 64 similar-cost streams, removing intra-wave and intra-CU skew.
@@ -2243,18 +2237,11 @@ int m3gpu_decode_batch_dev_perm(
     int32_t* d_out_errs, uint32_t stride, void* hip_stream) {
     if (!nseries) return M3GPU_OK;
     hipStream_t s = (hipStream_t)hip_stream;
-    static int xcd_swz = -1;
-    if (xcd_swz < 0) {
-        const char* e = getenv("M3GPU_DECODE_XCD");
-        xcd_swz = (e && e[0] == '1') ? 1 : 0;
-    }
-    uint32_t nblk = grid_lane(nseries);
-    if (xcd_swz) nblk = (nblk + 7) & ~7u; /* exact 8-way swizzle bijection */
-    hipLaunchKernelGGL(m3::k_decode_batch, dim3(nblk),
+    hipLaunchKernelGGL(m3::k_decode_batch, dim3(grid_lane(nseries)),
                        dim3(BLOCK_THREADS), 0, s,
                        d_blobs, d_offsets, d_lens, d_perm, nseries,
                        int_optimized, default_unit, d_out_ts, d_out_vals,
-                       d_out_counts, d_out_errs, stride, xcd_swz);
+                       d_out_counts, d_out_errs, stride);
     HIP_TRY(hipGetLastError());
     return M3GPU_OK;
 }
