@@ -47,7 +47,10 @@ def parse_args():
     p.add_argument("--gpus", type=int, default=1)
     p.add_argument("--steps", type=int, default=10)
     p.add_argument("--warmup", type=int, default=3)
-    p.add_argument("--nseries", type=int, default=1_000_000)
+    p.add_argument("--nseries", type=int, default=None,
+                   help="TOTAL series. Default: 1M PER GPU (weak scaling, "
+                        "BASELINE config 4: 8M across 8 GPUs); an explicit "
+                        "value is a fixed total (strong scaling)")
     p.add_argument("--npts", type=int, default=1440)
     p.add_argument("--chunk", type=int, default=65536)
     p.add_argument("--cpu-sample-series", type=int, default=0,
@@ -91,6 +94,11 @@ def main():
     device = f"cuda:{local_rank}"
     torch.cuda.set_device(device)
 
+    if args.nseries is None:
+        args.nseries = 1_000_000 * world  # weak scaling: 1M series per GPU
+        args.scaling = "weak"
+    else:
+        args.scaling = "strong" if world > 1 else "weak"
     lo, hi = shard_range(args.nseries, world, rank)
     n_local = hi - lo
     t0 = time.time()
@@ -241,7 +249,7 @@ def main():
             "warmup": args.warmup,
             "ms_per_step": elapsed / args.steps * 1e3,
             "higher_is_better": True,
-            "scaling": "weak",
+            "scaling": args.scaling,
             "vs_baseline": None,
             "dtype": "f64",
             "data": "synthetic" if args.data == "synthetic" else "replicated production streams",
@@ -353,7 +361,7 @@ def bench_encode(args, torch, engine, workload, device, n_local, lo):
         "value": dps, "unit": "datapoints/sec", "n_gpus": 1,
         "steps": args.steps, "warmup": args.warmup,
         "ms_per_step": dt / args.steps * 1e3, "higher_is_better": True,
-        "scaling": "weak", "vs_baseline": None, "dtype": "f64",
+        "scaling": args.scaling, "vs_baseline": None, "dtype": "f64",
         "data": "synthetic",
         "config": {"workload": "1M series batched M3TSZ encode, 1x MI355X",
                    "nseries": n_local, "npts": npts,
@@ -400,7 +408,7 @@ def bench_rollup(args, torch, engine, d_blob, d_offsets, d_lens, enc_bytes,
         "value": dps, "unit": "datapoints/sec", "n_gpus": 1,
         "steps": args.steps, "warmup": args.warmup,
         "ms_per_step": dt / args.steps * 1e3, "higher_is_better": True,
-        "scaling": "weak", "vs_baseline": None, "dtype": "f64",
+        "scaling": args.scaling, "vs_baseline": None, "dtype": "f64",
         "data": "synthetic",
         "config": {"workload": "Fused decode->downsample 10s->1m rollup, 1x MI355X",
                    "nseries": n_local, "npts": args.npts, "nbuckets": nbuckets,
@@ -468,7 +476,7 @@ def bench_tiles(args, torch, engine, d_blob, d_offsets, d_lens, enc_bytes,
         "value": dps, "unit": "datapoints/sec", "n_gpus": 1,
         "steps": args.steps, "warmup": args.warmup,
         "ms_per_step": dt / args.steps * 1e3, "higher_is_better": True,
-        "scaling": "weak", "vs_baseline": None, "dtype": "f64",
+        "scaling": args.scaling, "vs_baseline": None, "dtype": "f64",
         "data": "synthetic",
         "config": {"workload": "AggregateTiles 10s->1m, 1x MI355X",
                    "nseries": n_local, "npts": args.npts,
