@@ -1490,13 +1490,14 @@ k_compact(const uint8_t* __restrict__ src, uint32_t src_stride,
  * wave-per-series kernel (64-deep staging). */
 #define RQCAP_LANE 16
 
-template <bool WITH_Q>
-__global__ void __launch_bounds__(BLOCK_THREADS)
+template <bool WITH_Q, int METRIC>
+__global__ void __launch_bounds__(BLOCK_THREADS, 5) /* target 5 waves/SIMD:
+    the kernel sits a few VGPRs over the 96 granule; the hint trades them */
 k_rollup_lane(const uint8_t* __restrict__ blobs,
               const uint64_t* __restrict__ offsets,
               const uint32_t* __restrict__ lens,
               uint32_t nseries, int int_optimized, uint8_t default_unit,
-              int metric_type, int64_t window_ns, uint32_t nbuckets,
+              int64_t window_ns, uint32_t nbuckets,
               RollupPlan plan,
               double* __restrict__ out, int64_t* __restrict__ out_window_ts,
               int32_t* __restrict__ out_errs) {
@@ -1536,7 +1537,7 @@ k_rollup_lane(const uint8_t* __restrict__ blobs,
             int32_t t = plan.agg_types[k];
             int8_t qi = plan.qidx[k];
             double r = 0;
-            if (metric_type == M3GPU_METRIC_COUNTER) { /* counter.go:112-131 */
+            if (METRIC == M3GPU_METRIC_COUNTER) { /* counter.go:112-131 */
                 switch (t) {
                 case M3GPU_AGG_MIN: r = (double)bs.imin; break;
                 case M3GPU_AGG_MAX: r = (double)bs.imax; break;
@@ -1547,7 +1548,7 @@ k_rollup_lane(const uint8_t* __restrict__ blobs,
                 case M3GPU_AGG_STDEV: r = m3_stdev(bs.count, (double)bs.isumsq, (double)bs.isum); break;
                 default: break;
                 }
-            } else if (metric_type == M3GPU_METRIC_GAUGE) { /* gauge.go:144-165 */
+            } else if (METRIC == M3GPU_METRIC_GAUGE) { /* gauge.go:144-165 */
                 switch (t) {
                 case M3GPU_AGG_LAST: r = bs.last; break;
                 case M3GPU_AGG_MIN: r = bs.fmin; break;
@@ -1640,14 +1641,14 @@ k_rollup_lane(const uint8_t* __restrict__ blobs,
                     }
                 }
                 if (ok) {
-                    if (metric_type == M3GPU_METRIC_COUNTER) { /* counter.go:52-78 */
+                    if (METRIC == M3GPU_METRIC_COUNTER) { /* counter.go:52-78 */
                         int64_t iv = go_f2i(v);
                         bs.isum += iv;
                         bs.count++;
                         if (bs.imax < iv) bs.imax = iv;
                         if (bs.imin > iv) bs.imin = iv;
                         bs.isumsq += iv * iv;
-                    } else if (metric_type == M3GPU_METRIC_GAUGE) { /* gauge.go:73-103 */
+                    } else if (METRIC == M3GPU_METRIC_GAUGE) { /* gauge.go:73-103 */
                         if (bs.last_at == 0 || t > bs.last_at) { bs.last_at = t; bs.last = v; }
                         bs.count++;
                         if (!isnan(v)) {
@@ -2402,17 +2403,35 @@ int m3gpu_rollup_batch_dev(
 
     hipStream_t s = (hipStream_t)hip_stream;
     bool with_q = plan.nq > 0 && metric_type == M3GPU_METRIC_TIMER;
+    /* metric type is a template parameter: dead per-metric BucketState
+     * fields drop out of the register file */
     if (with_q)
-        hipLaunchKernelGGL(m3::k_rollup_lane<true>, dim3(grid_lane(nseries)),
+        hipLaunchKernelGGL((m3::k_rollup_lane<true, M3GPU_METRIC_TIMER>),
+                           dim3(grid_lane(nseries)),
                            dim3(BLOCK_THREADS), 0, s,
                            d_blobs, d_offsets, d_lens, nseries, int_optimized,
-                           default_unit, metric_type, window_ns, nbuckets, plan,
+                           default_unit, window_ns, nbuckets, plan,
+                           d_out, d_out_window_ts, d_out_errs);
+    else if (metric_type == M3GPU_METRIC_COUNTER)
+        hipLaunchKernelGGL((m3::k_rollup_lane<false, M3GPU_METRIC_COUNTER>),
+                           dim3(grid_lane(nseries)),
+                           dim3(BLOCK_THREADS), 0, s,
+                           d_blobs, d_offsets, d_lens, nseries, int_optimized,
+                           default_unit, window_ns, nbuckets, plan,
+                           d_out, d_out_window_ts, d_out_errs);
+    else if (metric_type == M3GPU_METRIC_GAUGE)
+        hipLaunchKernelGGL((m3::k_rollup_lane<false, M3GPU_METRIC_GAUGE>),
+                           dim3(grid_lane(nseries)),
+                           dim3(BLOCK_THREADS), 0, s,
+                           d_blobs, d_offsets, d_lens, nseries, int_optimized,
+                           default_unit, window_ns, nbuckets, plan,
                            d_out, d_out_window_ts, d_out_errs);
     else
-        hipLaunchKernelGGL(m3::k_rollup_lane<false>, dim3(grid_lane(nseries)),
+        hipLaunchKernelGGL((m3::k_rollup_lane<false, M3GPU_METRIC_TIMER>),
+                           dim3(grid_lane(nseries)),
                            dim3(BLOCK_THREADS), 0, s,
                            d_blobs, d_offsets, d_lens, nseries, int_optimized,
-                           default_unit, metric_type, window_ns, nbuckets, plan,
+                           default_unit, window_ns, nbuckets, plan,
                            d_out, d_out_window_ts, d_out_errs);
     HIP_TRY(hipGetLastError());
     if (with_q) {
