@@ -236,7 +236,7 @@ def main():
             pass
 
     cpu_baseline = None
-    if rank == 0 and not args.skip_cpu_baseline:
+    if rank == 0 and world == 1 and not args.skip_cpu_baseline:
         cpu_baseline = run_cpu_baseline(args, d_blob, d_offsets, d_lens, n_local)
 
     if rank == 0:
