@@ -287,19 +287,24 @@ def run_cpu_baseline(args, d_blob, d_offsets, d_lens, n_local):
     import oracle  # permitted: cpu_baseline leg only
 
     cores = os.cpu_count()
-    # probe with 2k series to size the sample
-    probe_n = min(2000, n_local)
+    # probe to size the sample (large enough that thread spin-up noise
+    # does not dominate the per-series rate)
+    probe_n = min(50000, n_local)
     off = d_offsets[:probe_n + 1].cpu().numpy().astype(np.uint64)
     lens = d_lens[:probe_n].cpu().numpy().astype(np.uint32)
     blob = d_blob[: int(off[-1])].cpu().numpy()
     t0 = time.time()
     oracle.decode_batch(blob, off, stride=args.npts, nthreads=cores)
     probe_t = time.time() - t0
-    target_series = probe_n
+    per_series = probe_t / probe_n
     if args.cpu_sample_series:
         target_series = min(args.cpu_sample_series, n_local)
+    elif per_series * n_local <= 30.0:
+        # the whole batch fits the 10-30s budget: unbiased full sample
+        # (a PREFIX of the schedule-sorted blob would over-sample the
+        # shortest streams)
+        target_series = n_local
     else:
-        per_series = probe_t / probe_n
         target_series = int(min(n_local, max(probe_n, 15.0 / per_series)))
     off = d_offsets[:target_series + 1].cpu().numpy().astype(np.uint64)
     lens = d_lens[:target_series].cpu().numpy().astype(np.uint32)
