@@ -147,13 +147,17 @@ def main():
         # identity and is dropped
         new_lens = d_lens[d_perm.long()]
         aligned = ((new_lens.to(torch.int64) + 15) // 16) * 16
-        new_off = torch.zeros_like(aligned)
-        torch.cumsum(aligned[:-1], 0, out=new_off[1:])
+        # n+1 offsets: the end sentinel is needed by the oracle-side
+        # cpu_baseline slice (the kernels only use offsets[series])
+        new_off = torch.zeros(aligned.numel() + 1, dtype=torch.int64,
+                              device=aligned.device)
+        torch.cumsum(aligned, 0, out=new_off[1:])
         d_blob2 = torch.empty_like(d_blob)
         engine.regather_dev(d_blob, d_offsets, d_lens, d_perm,
-                            new_off.to(torch.int64), d_blob2)
+                            new_off[:-1].contiguous(), d_blob2)
         torch.cuda.synchronize()
         d_blob, d_offsets, d_lens = d_blob2, new_off, new_lens.contiguous()
+        # (d_offsets now has n+1 entries; kernels index only [0, n))
         # series identity = pack order: output row i now holds the stream
         # that was at perm[i] before the repack
         if expected_counts is not None:
