@@ -33,7 +33,10 @@ from .engine import (  # noqa: F401
     encode_batch_dev,
     engine_available,
     lib,
+    merge_batch_dev,
     pack_streams,
+    parse_unaggregated,
+    regather_dev,
     aggregate_tiles_dev,
     rollup_batch,
     rollup_batch_dev,
