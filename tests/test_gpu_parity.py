@@ -738,3 +738,49 @@ def test_commitlog_bootstrap_end_to_end(torch, engine, tmp_path):
     for i in range(nseries):
         assert meta[i]["id"] == f"boot.{i:04d}".encode()
         assert bytes(g_rows[i, :g_lens[i]]) == bytes(o_rows[i, :o_lens[i]]), i
+
+
+def test_regather_layout_pass(torch, engine):
+    """k_regather physical reorder: dst row i holds src stream perm[i];
+    decoding the repacked blob without a scheduling perm must equal the
+    oracle decode of the original streams in perm order."""
+    from m3_amd.engine import pack_streams
+    rng = np.random.default_rng(79)
+    nseries, npts = 300, 120
+    START = 1427162462 * 10**9
+    ts = START + np.cumsum(rng.integers(1, 60, (nseries, npts)), axis=1) * 10**9
+    vals = np.round(rng.random((nseries, npts)) * 1e3, 3)
+    streams = [oracle.encode_series(ts[i], vals[i], start_ns=int(ts[i, 0]))
+               for i in range(nseries)]
+    blob, offsets, lens = pack_streams(streams)
+    d_blob = torch.from_numpy(blob).to("cuda:0")
+    d_off = torch.from_numpy(offsets.astype(np.int64)).to("cuda:0")
+    d_lens = torch.from_numpy(lens.astype(np.int32)).to("cuda:0")
+    perm = torch.randperm(nseries, device="cuda:0").to(torch.int32)
+    new_lens = d_lens[perm.long()]
+    aligned = ((new_lens.to(torch.int64) + 15) // 16) * 16
+    new_off = torch.zeros(nseries + 1, dtype=torch.int64, device="cuda:0")
+    torch.cumsum(aligned, 0, out=new_off[1:])
+    d_blob2 = torch.zeros(int(new_off[-1].item()), dtype=torch.uint8,
+                          device="cuda:0")
+    engine.regather_dev(d_blob, d_off, d_lens, perm, new_off[:-1].contiguous(),
+                        d_blob2)
+    out_ts = torch.zeros((nseries, npts), dtype=torch.int64, device="cuda:0")
+    out_vals = torch.zeros((nseries, npts), dtype=torch.float64, device="cuda:0")
+    out_counts = torch.empty(nseries, dtype=torch.int32, device="cuda:0")
+    out_errs = torch.empty(nseries, dtype=torch.int32, device="cuda:0")
+    engine.decode_batch_dev(d_blob2, new_off[:-1].contiguous(),
+                            new_lens.contiguous().to(torch.int32), out_ts,
+                            out_vals, out_counts, out_errs)
+    torch.cuda.synchronize()
+    assert np.all(out_errs.cpu().numpy() == 0)
+    p = perm.cpu().numpy()
+    o_ts, o_vals, _ = oracle.decode_batch(
+        blob, np.concatenate([offsets, [np.uint64(len(blob))]]), stride=npts)
+    g_ts = out_ts.cpu().numpy()
+    g_vals = out_vals.cpu().numpy()
+    for i in range(nseries):
+        s = p[i]
+        assert np.array_equal(g_ts[i], o_ts[s, :npts]), i
+        assert np.array_equal(g_vals[i].view(np.uint64),
+                              o_vals[s, :npts].view(np.uint64)), i
