@@ -775,8 +775,8 @@ def test_regather_layout_pass(torch, engine):
     torch.cuda.synchronize()
     assert np.all(out_errs.cpu().numpy() == 0)
     p = perm.cpu().numpy()
-    o_ts, o_vals, _ = oracle.decode_batch(
-        blob, np.concatenate([offsets, [np.uint64(len(blob))]]), stride=npts)
+    # pack_streams offsets already carry the n+1 end sentinel
+    o_ts, o_vals, _ = oracle.decode_batch(blob, offsets, stride=npts)
     g_ts = out_ts.cpu().numpy()
     g_vals = out_vals.cpu().numpy()
     for i in range(nseries):
