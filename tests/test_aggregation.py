@@ -221,3 +221,20 @@ def test_ckms_list_len_probe():
     for n in (600, 5000, 100000):
         ln = oracle.ckms_list_len(rng.random(n) * 1e4, qs)
         assert 100 < ln < 1500, (n, ln)
+
+
+def test_rollup_duplicate_quantile_aggs():
+    """median and p50 are the same quantile: the registered list is sorted
+    unique (m3 registers deduped quantiles), both output slots read the
+    same computed value."""
+    rng = np.random.default_rng(83)
+    n = 40
+    ts = 1427162462 * 10**9 + np.arange(n, dtype=np.int64) * 10**9
+    vals = np.round(rng.random(n) * 100, 2)
+    window = n * 10**9
+    out, wts = oracle.rollup_batch(ts[None, :], vals[None, :],
+                                   np.array([n], np.uint32),
+                                   oracle.METRIC_TIMER, window, 1,
+                                   ["median", "p50", "p95", "count"])
+    assert out[0, 0, 0] == out[0, 0, 1]  # median == p50
+    assert out[0, 0, 3] == n

@@ -784,3 +784,37 @@ def test_regather_layout_pass(torch, engine):
         assert np.array_equal(g_ts[i], o_ts[s, :npts]), i
         assert np.array_equal(g_vals[i].view(np.uint64),
                               o_vals[s, :npts].view(np.uint64)), i
+
+
+def test_rollup_duplicate_quantile_aggs_gpu(torch, engine):
+    """median + p50 share one quantile slot in the plan (sorted unique
+    list); GPU bit-equal to oracle for the duplicated agg outputs."""
+    from m3_amd.engine import pack_streams
+    rng = np.random.default_rng(89)
+    nseries, npts = 64, 90
+    START = 1427162462 * 10**9
+    window = npts * 10**9
+    start = (START // window) * window
+    ts = start + np.arange(npts, dtype=np.int64) * 10**9
+    ts = np.broadcast_to(ts, (nseries, npts)).copy()
+    vals = np.round(rng.random((nseries, npts)) * 100, 2)
+    counts = np.full(nseries, npts, np.uint32)
+    aggs = ["median", "p50", "p95", "count"]
+    o_out, _ = oracle.rollup_batch(ts, vals, counts, oracle.METRIC_TIMER,
+                                   window, 1, aggs)
+    streams = [oracle.encode_series(ts[i], vals[i], start_ns=int(ts[i, 0]))
+               for i in range(nseries)]
+    blob, offsets, lens = pack_streams(streams)
+    d_blob = torch.from_numpy(blob).to("cuda:0")
+    d_off = torch.from_numpy(offsets.astype(np.int64)).to("cuda:0")
+    d_lens = torch.from_numpy(lens.astype(np.int32)).to("cuda:0")
+    out = torch.empty((nseries, 1, 4), dtype=torch.float64, device="cuda:0")
+    wts = torch.empty((nseries, 1), dtype=torch.int64, device="cuda:0")
+    errs = torch.empty(nseries, dtype=torch.int32, device="cuda:0")
+    engine.rollup_batch_dev(d_blob, d_off, d_lens, engine.METRIC_TIMER,
+                            window, 1, aggs, out, wts, errs)
+    torch.cuda.synchronize()
+    assert np.all(errs.cpu().numpy() == 0)
+    g = out.cpu().numpy()
+    assert np.array_equal(g.view(np.uint64), o_out.view(np.uint64))
+    assert np.array_equal(g[:, 0, 0], g[:, 0, 1])
