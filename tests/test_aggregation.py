@@ -229,7 +229,9 @@ def test_rollup_duplicate_quantile_aggs():
     same computed value."""
     rng = np.random.default_rng(83)
     n = 40
-    ts = 1427162462 * 10**9 + np.arange(n, dtype=np.int64) * 10**9
+    window_align = n * 10**9
+    ts = ((1427162462 * 10**9) // window_align) * window_align + \
+        np.arange(n, dtype=np.int64) * 10**9
     vals = np.round(rng.random(n) * 100, 2)
     window = n * 10**9
     out, wts = oracle.rollup_batch(ts[None, :], vals[None, :],
