@@ -147,6 +147,17 @@ int m3gpu_rollup_batch_dev(
     double* d_out, int64_t* d_out_window_ts, int32_t* d_out_errs,
     void* hip_stream);
 
+/* As above with explicit CKMS stream options (quantile/cm/options.go:30-32;
+ * defaults eps=1e-3, insert_and_compress_every=1024). eps in (0, 0.5),
+ * every in [1, 1024] (deep-tier buffers are sized for a 2*1024 peak). */
+int m3gpu_rollup_batch_dev_opts(
+    const uint8_t* d_blobs, const uint64_t* d_offsets, const uint32_t* d_lens,
+    uint32_t nseries, int int_optimized, uint8_t default_unit,
+    int metric_type, int64_t window_ns, uint32_t nbuckets,
+    const int32_t* agg_types, int naggs,
+    double* d_out, int64_t* d_out_window_ts, int32_t* d_out_errs,
+    void* hip_stream, double eps, int insert_and_compress_every);
+
 /* -------- replica-deduplicating merge (replaces MultiReaderIterator over
  * one slice of R replica iterators: dbnode/encoding/multi_reader_iterator.go
  * :62-155 + iterators.go:56-237, IterateLastPushed default). Decoded replica

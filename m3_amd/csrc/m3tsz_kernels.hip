@@ -1225,6 +1225,12 @@ struct RollupPlan {
      * (the calcQuantiles walk). Computed on the host; buckets beyond it
      * flag M3GPU_SERIES_BUCKET_OVERFLOW rather than approximate. */
     int32_t exact_cap;
+    /* CKMS stream options (quantile/cm/options.go:30-32 defaults:
+     * eps=1e-3, insertAndCompressEvery=1024); configurable via the
+     * *_opts ABI entries. every is capped at 1024 by validation (the
+     * deep-tier LDS buffers are sized for a 2*1024 peak). */
+    double eps;
+    int32_t every;
 };
 
 struct BucketState {
@@ -1855,7 +1861,7 @@ __device__ __forceinline__ void ckms_sort(const double* buf, double* tmp,
 __device__ __forceinline__ int64_t ckms_threshold(const RollupPlan& plan,
                                                   int64_t num_vals, int64_t rank) {
     int64_t min_val = INT64_MAX;
-    double eps = 2.0 * 1e-3;
+    double eps = 2.0 * plan.eps;
     for (int i = 0; i < plan.nq; i++) {
         int64_t qmin;
         if (rank >= (int64_t)(plan.qs[i] * (double)num_vals))
@@ -1988,7 +1994,7 @@ __device__ __forceinline__ void ckms_add(CkmsLds& L, CkmsState& st,
     }
     __builtin_amdgcn_wave_barrier();
     __threadfence_block();
-    if (st.counter == 1024) { /* insertAndCompressEvery, checked BEFORE ++ */
+    if (st.counter == plan.every) { /* insertAndCompressEvery, checked BEFORE ++ */
         ckms_insert_compress(L, st, plan, sorted, lane);
         st.counter = 0;
     }
@@ -2294,21 +2300,28 @@ int m3gpu_regather_dev(
     return M3GPU_OK;
 }
 
-int m3gpu_rollup_batch_dev(
+int m3gpu_rollup_batch_dev_opts(
     const uint8_t* d_blobs, const uint64_t* d_offsets, const uint32_t* d_lens,
     uint32_t nseries, int int_optimized, uint8_t default_unit,
     int metric_type, int64_t window_ns, uint32_t nbuckets,
     const int32_t* agg_types, int naggs,
     double* d_out, int64_t* d_out_window_ts, int32_t* d_out_errs,
-    void* hip_stream) {
+    void* hip_stream, double eps, int every) {
     if (!nseries) return M3GPU_OK;
     if (naggs < 1 || naggs > MAX_AGGS || window_ns <= 0) {
         snprintf(g_err, sizeof(g_err), "bad naggs/window");
         return M3GPU_ERR_BADARG;
     }
+    if (!(eps > 0.0 && eps < 0.5) || every < 1 || every > 1024) {
+        snprintf(g_err, sizeof(g_err),
+                 "eps must be in (0,0.5), every in [1,1024]");
+        return M3GPU_ERR_BADARG;
+    }
     m3::RollupPlan plan;
     memset(&plan, 0, sizeof(plan));
     plan.naggs = naggs;
+    plan.eps = eps;
+    plan.every = every;
     /* sorted unique quantile list + per-agg mapping (timer stream is
      * registered with the sorted unique quantile set) */
     double qs[MAX_AGGS];
@@ -2358,7 +2371,7 @@ int m3gpu_rollup_batch_dev(
      * are exact order statistics. */
     plan.exact_cap = QCAP;
     {
-        double eps2 = 2.0 * 1e-3; /* cm defaults, options.go:30-32 */
+        double eps2 = 2.0 * plan.eps;
         for (int v = 4; v <= QCAP + 1; v++) /* minSamplesToCompress=3 */ {
             bool merges = false;
             for (int mr = 0; mr <= v && !merges; mr++) {
@@ -2515,6 +2528,19 @@ int m3gpu_rollup_batch_dev(
         free(h_errs);
     }
     return M3GPU_OK;
+}
+
+int m3gpu_rollup_batch_dev(
+    const uint8_t* d_blobs, const uint64_t* d_offsets, const uint32_t* d_lens,
+    uint32_t nseries, int int_optimized, uint8_t default_unit,
+    int metric_type, int64_t window_ns, uint32_t nbuckets,
+    const int32_t* agg_types, int naggs,
+    double* d_out, int64_t* d_out_window_ts, int32_t* d_out_errs,
+    void* hip_stream) {
+    return m3gpu_rollup_batch_dev_opts(
+        d_blobs, d_offsets, d_lens, nseries, int_optimized, default_unit,
+        metric_type, window_ns, nbuckets, agg_types, naggs, d_out,
+        d_out_window_ts, d_out_errs, hip_stream, 1e-3, 1024);
 }
 
 int m3gpu_merge_batch_dev(

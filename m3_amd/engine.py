@@ -260,15 +260,24 @@ def regather_dev(d_blob, d_offsets, d_lens, d_perm, d_dst_offsets, d_dst):
 
 def rollup_batch_dev(d_blob, d_offsets, d_lens, metric_type, window_ns,
                      nbuckets, agg_types, out, out_window_ts, out_errs,
-                     int_optimized=True, default_unit=1):
+                     int_optimized=True, default_unit=1,
+                     eps=1e-3, every=1024):
+    """eps/every: CKMS stream options (reference defaults; cm
+    options.go:30-32). Non-default values route via the _opts ABI."""
     nseries = d_lens.numel()
     aggs = np.asarray([M3GPU_AGG[a] if isinstance(a, str) else a
                        for a in agg_types], dtype=np.int32)
-    rc = lib().m3gpu_rollup_batch_dev(
+    L = lib()
+    if not hasattr(L.m3gpu_rollup_batch_dev_opts, "_configured"):
+        L.m3gpu_rollup_batch_dev_opts.restype = c_int
+        L.m3gpu_rollup_batch_dev_opts.argtypes =             list(L.m3gpu_rollup_batch_dev.argtypes) + [c_f64, c_int]
+        L.m3gpu_rollup_batch_dev_opts._configured = True
+    rc = L.m3gpu_rollup_batch_dev_opts(
         _dev_ptr(d_blob), _dev_ptr(d_offsets), _dev_ptr(d_lens), nseries,
         1 if int_optimized else 0, default_unit, metric_type, window_ns,
         nbuckets, _pp(aggs, c_i32), len(aggs), _dev_ptr(out),
-        _dev_ptr(out_window_ts), _dev_ptr(out_errs), _torch_stream())
+        _dev_ptr(out_window_ts), _dev_ptr(out_errs), _torch_stream(),
+        eps, every)
     _check(rc, "m3gpu_rollup_batch_dev")
 
 

@@ -73,6 +73,8 @@ def _setup(L):
     L.oracle_rollup_batch.argtypes = [P(c_i64), P(c_f64), P(c_u32), c_i64, c_i64,
                                       c_int, c_i64, c_i64, P(c_i32), c_int,
                                       P(c_f64), P(c_i64), c_int]
+    L.oracle_rollup_batch_opts.restype = c_int
+    L.oracle_rollup_batch_opts.argtypes = L.oracle_rollup_batch.argtypes + [c_f64, c_int]
     # test wrappers
     for name in ("oracle_test_write_dod_unchanged", "oracle_test_write_dod_changed",
                  "oracle_test_write_xor", "oracle_test_write_annotation",
@@ -251,7 +253,8 @@ AGG = dict(last=1, min=2, max=3, mean=4, median=5, count=6, sum=7, sumsq=8,
 
 
 def rollup_batch(ts_ns, vals, counts, metric_type, window_ns, nbuckets,
-                 agg_types, nthreads=0, want_window_ts=True):
+                 agg_types, nthreads=0, want_window_ts=True,
+                 eps=1e-3, every=1024):
     ts = np.ascontiguousarray(ts_ns, dtype=np.int64)
     v = np.ascontiguousarray(vals, dtype=np.float64)
     c = np.ascontiguousarray(counts, dtype=np.uint32)
@@ -262,13 +265,14 @@ def rollup_batch(ts_ns, vals, counts, metric_type, window_ns, nbuckets,
     wts = np.zeros((nseries, nbuckets), dtype=np.int64) if want_window_ts else None
     if nthreads <= 0:
         nthreads = os.cpu_count()
-    r = lib().oracle_rollup_batch(
+    r = lib().oracle_rollup_batch_opts(
         ts.ctypes.data_as(P(c_i64)), v.ctypes.data_as(P(c_f64)),
         c.ctypes.data_as(P(c_u32)), nseries, stride,
         metric_type, window_ns, nbuckets,
         aggs.ctypes.data_as(P(c_i32)), len(aggs),
         out.ctypes.data_as(P(c_f64)),
-        wts.ctypes.data_as(P(c_i64)) if wts is not None else None, nthreads)
+        wts.ctypes.data_as(P(c_i64)) if wts is not None else None, nthreads,
+        eps, every)
     if r != 0:
         raise RuntimeError(f"oracle rollup error {r}")
     return (out, wts) if want_window_ts else out

@@ -447,11 +447,12 @@ enum { METRIC_COUNTER = 0, METRIC_GAUGE = 1, METRIC_TIMER = 2 };
  */
 static double agg_quantile_of(int t);
 
-int oracle_rollup_series(
+int oracle_rollup_series_opts(
     const int64_t* ts_ns, const double* vals, int64_t n,
     int metric_type, int64_t base_ns, int64_t window_ns, int64_t nbuckets,
     const int32_t* agg_types, int naggs,
-    double* out /* nbuckets x naggs */, int64_t* out_window_ts /* nbuckets, or NULL */) {
+    double* out /* nbuckets x naggs */, int64_t* out_window_ts /* nbuckets, or NULL */,
+    double eps, int every) {
     /* collect per-bucket state */
     typedef struct {
         int64_t count;
@@ -490,7 +491,7 @@ int oracle_rollup_series(
     if (need_stream && metric_type == METRIC_TIMER) {
         streams = (cm_stream*)malloc((size_t)nbuckets * sizeof(cm_stream));
         for (int64_t b = 0; b < nbuckets; b++)
-            stream_init(&streams[b], qlist, nql, 1e-3, 1024);
+            stream_init(&streams[b], qlist, nql, eps, every);
     }
 
     int err = 0;
@@ -616,31 +617,54 @@ static double agg_quantile_of(int t) {
 }
 
 /* Batch rollup over SoA decoded series (one thread per series chunk). */
-int oracle_rollup_batch(
+int oracle_rollup_series(
+    const int64_t* ts_ns, const double* vals, int64_t n,
+    int metric_type, int64_t base_ns, int64_t window_ns, int64_t nbuckets,
+    const int32_t* agg_types, int naggs,
+    double* out, int64_t* out_window_ts) {
+    /* m3 CKMS production defaults (cm options.go:30-32) */
+    return oracle_rollup_series_opts(ts_ns, vals, n, metric_type, base_ns,
+                                     window_ns, nbuckets, agg_types, naggs,
+                                     out, out_window_ts, 1e-3, 1024);
+}
+
+int oracle_rollup_batch_opts(
     const int64_t* ts_ns, const double* vals, const uint32_t* counts,
     int64_t nseries, int64_t stride,
     int metric_type, int64_t window_ns, int64_t nbuckets,
     const int32_t* agg_types, int naggs,
     double* out /* nseries x nbuckets x naggs */,
     int64_t* out_window_ts /* nseries x nbuckets, or NULL */,
-    int nthreads) {
+    int nthreads, double eps, int every) {
     int err = 0;
     (void)nthreads;
 #pragma omp parallel for schedule(dynamic, 16) num_threads(nthreads)
     for (int64_t i = 0; i < nseries; i++) {
         if (err) continue;
         int64_t base = (ts_ns[i * stride] / window_ns) * window_ns;
-        int r = oracle_rollup_series(
+        int r = oracle_rollup_series_opts(
             ts_ns + i * stride, vals + i * stride, (int64_t)counts[i],
             metric_type, base, window_ns, nbuckets, agg_types, naggs,
             out + i * nbuckets * naggs,
-            out_window_ts ? out_window_ts + i * nbuckets : NULL);
+            out_window_ts ? out_window_ts + i * nbuckets : NULL, eps, every);
         if (r) {
 #pragma omp critical
             err = r;
         }
     }
     return err;
+}
+
+int oracle_rollup_batch(
+    const int64_t* ts_ns, const double* vals, const uint32_t* counts,
+    int64_t nseries, int64_t stride,
+    int metric_type, int64_t window_ns, int64_t nbuckets,
+    const int32_t* agg_types, int naggs,
+    double* out, int64_t* out_window_ts, int nthreads) {
+    return oracle_rollup_batch_opts(ts_ns, vals, counts, nseries, stride,
+                                    metric_type, window_ns, nbuckets,
+                                    agg_types, naggs, out, out_window_ts,
+                                    nthreads, 1e-3, 1024);
 }
 
 /* ==================== multi-replica deduplicating merge ====================

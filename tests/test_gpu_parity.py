@@ -818,3 +818,67 @@ def test_rollup_duplicate_quantile_aggs_gpu(torch, engine):
     g = out.cpu().numpy()
     assert np.array_equal(g.view(np.uint64), o_out.view(np.uint64))
     assert np.array_equal(g[:, 0, 0], g[:, 0, 1])
+
+
+def test_rollup_nondefault_ckms_options(torch, engine):
+    """Configurable CKMS stream options (eps/insertAndCompressEvery, cm
+    options.go:30-32) through the _opts ABI: non-default (eps=0.01,
+    every=64) compresses far earlier — exact_cap shrinks and the deep
+    CKMS tier engages at shallow depths; bit-equal to the oracle with the
+    same options at depths straddling the insert cadence."""
+    from m3_amd.engine import pack_streams
+    rng = np.random.default_rng(97)
+    eps, every = 0.01, 64
+    window_s = 1040
+    window = window_s * 10**9
+    start = (1427162462 * 10**9 // window) * window
+    aggs = ["median", "p95", "p99", "min", "max", "count", "sum"]
+    series = []
+    for cad in (26, 16, 13, 8, 2):   # depths 40, 65, 80, 130, 520
+        depth = window_s // cad
+        npts = 2 * depth
+        t = start + np.arange(npts, dtype=np.int64) * cad * 10**9
+        for dist in range(3):
+            if dist == 0:
+                v = np.round(rng.random(npts) * 1e3, 3)
+            elif dist == 1:
+                v = np.round(np.sort(rng.random(npts)) * 100, 4)
+            else:
+                v = np.round(rng.random(npts) * 5, 1)  # ties
+            series.append((t, v))
+    nseries = len(series)
+    width = max(len(t) for t, _ in series)
+    ts = np.zeros((nseries, width), np.int64)
+    vals = np.zeros((nseries, width), np.float64)
+    counts = np.zeros(nseries, np.uint32)
+    for i, (t, v) in enumerate(series):
+        ts[i, :len(t)] = t
+        vals[i, :len(t)] = v
+        counts[i] = len(t)
+    o_out, o_wts = oracle.rollup_batch(ts, vals, counts, oracle.METRIC_TIMER,
+                                       window, 2, aggs, eps=eps, every=every)
+    streams = [oracle.encode_series(ts[i, :counts[i]], vals[i, :counts[i]],
+                                    start_ns=int(ts[i, 0]))
+               for i in range(nseries)]
+    blob, offsets, lens = pack_streams(streams)
+    d_blob = torch.from_numpy(blob).to("cuda:0")
+    d_off = torch.from_numpy(offsets.astype(np.int64)).to("cuda:0")
+    d_lens = torch.from_numpy(lens.astype(np.int32)).to("cuda:0")
+    out = torch.empty((nseries, 2, len(aggs)), dtype=torch.float64,
+                      device="cuda:0")
+    wts = torch.empty((nseries, 2), dtype=torch.int64, device="cuda:0")
+    errs = torch.empty(nseries, dtype=torch.int32, device="cuda:0")
+    engine.rollup_batch_dev(d_blob, d_off, d_lens, engine.METRIC_TIMER,
+                            window, 2, aggs, out, wts, errs,
+                            eps=eps, every=every)
+    torch.cuda.synchronize()
+    assert np.all(errs.cpu().numpy() == 0)
+    g = out.cpu().numpy()
+    for i in range(nseries):
+        assert np.array_equal(g[i].view(np.uint64), o_out[i].view(np.uint64)), \
+            (i, g[i], o_out[i])
+    # invalid options are rejected, not approximated
+    import pytest as _pytest
+    with _pytest.raises(engine.M3GpuError):
+        engine.rollup_batch_dev(d_blob, d_off, d_lens, engine.METRIC_TIMER,
+                                window, 2, aggs, out, wts, errs, every=2000)
