@@ -17,6 +17,8 @@ from .engine import (  # noqa: F401
     CL_ERRORS,
     CommitLog,
     commitlog_bootstrap_dev,
+    commitlog_bootstrap_dir_dev,
+    commitlog_read_dir,
     FS_ERRORS,
     FilesetVolume,
     fileset_ingest_dev,

@@ -690,3 +690,67 @@ def parse_unaggregated(buf):
         return out
     finally:
         L.m3gpu_unagg_close(h)
+
+
+def commitlog_read_dir(dirpath):
+    """Read every commit log file in a directory in log-index order and
+    merge series ACROSS files by (namespace, id) — unique_index is only
+    unique within one file (writer.go seen-bitset is per file). Returns
+    the same series-dict list as CommitLog.series(), with points in
+    (file index, log order)."""
+    import glob as _glob
+    files = []
+    for p in sorted(_glob.glob(os.path.join(str(dirpath), "commitlog-*.db"))):
+        cl = CommitLog(p)
+        files.append((cl.index, p, cl))
+    files.sort(key=lambda t: t[0])
+    merged = {}
+    order = []
+    for _, _, cl in files:
+        try:
+            for m in cl.series():
+                key = (m["namespace"], m["id"])
+                if key not in merged:
+                    merged[key] = m
+                    order.append(key)
+                else:
+                    g = merged[key]
+                    g["ts"] = np.concatenate([g["ts"], m["ts"]])
+                    g["vals"] = np.concatenate([g["vals"], m["vals"]])
+                    g["units"] = np.concatenate([g["units"], m["units"]])
+                    base = len(g["annotations"])
+                    g["annotations"].extend(
+                        (base + pi, b) for pi, b in m["annotations"])
+        finally:
+            cl.close()
+    return [merged[k] for k in order]
+
+
+def commitlog_bootstrap_dir_dev(torch, dirpath, int_optimized=True,
+                                device="cuda:0"):
+    """Bootstrap from a DIRECTORY of commit logs (the reference commitlog
+    bootstrapper's whole-volume job): merge per series across files, then
+    batch-encode blocks with the HIP encoder."""
+    meta = commitlog_read_dir(dirpath)
+    if not meta:
+        return [], None, None, None
+    n = len(meta)
+    width = max(len(m["ts"]) for m in meta)
+    ts = np.zeros((n, width), np.int64)
+    vals = np.zeros((n, width), np.float64)
+    counts = np.zeros(n, np.int32)
+    for i, m in enumerate(meta):
+        k = len(m["ts"])
+        ts[i, :k] = m["ts"]
+        vals[i, :k] = m["vals"]
+        counts[i] = k
+    d_ts = torch.from_numpy(ts).to(device)
+    d_vals = torch.from_numpy(vals).to(device)
+    d_counts = torch.from_numpy(counts).to(device)
+    out_stride = (24 * width + 32 + 15) & ~15
+    d_bytes = torch.zeros((n, out_stride), dtype=torch.uint8, device=device)
+    d_lens = torch.empty(n, dtype=torch.int32, device=device)
+    d_errs = torch.empty(n, dtype=torch.int32, device=device)
+    encode_batch_dev(d_ts, d_vals, d_counts, d_bytes, d_lens, d_errs,
+                     int_optimized=int_optimized)
+    return meta, d_bytes, d_lens, d_errs

@@ -163,3 +163,29 @@ def test_commitlog_empty(tmp_path):
         assert cl.index == 4
         assert cl.num_entries == 0
         assert cl.num_series == 0
+
+
+def test_commitlog_directory_merge(tmp_path):
+    """Multi-file bootstrap: series identified by (namespace, id) across
+    files (unique_index is per-file), files consumed in log-index order."""
+    rng = np.random.default_rng(101)
+    t0 = START
+    # file index 1: series A(idx 0) + B(idx 1)
+    e1 = [(0, b"series.A", 0, t0 + 1 * 10**9, 1.0, 4, None, None),
+          (1, b"series.B", 0, t0 + 2 * 10**9, 2.0, 4, None, None),
+          (0, b"series.A", 0, t0 + 3 * 10**9, 3.0, 4, None, None)]
+    # file index 2: series A has a DIFFERENT unique index; C is new
+    e2 = [(7, b"series.A", 0, t0 + 4 * 10**9, 4.0, 4, None, None),
+          (9, b"series.C", 0, t0 + 5 * 10**9, 5.0, 4, None, None)]
+    clw.write_commitlog(str(tmp_path / "commitlog-0-5.db"), e2, index=2)
+    clw.write_commitlog(str(tmp_path / "commitlog-0-4.db"), e1, index=1)
+    from m3_amd.engine import commitlog_read_dir
+    meta = commitlog_read_dir(tmp_path)
+    by_id = {m["id"]: m for m in meta}
+    assert set(by_id) == {b"series.A", b"series.B", b"series.C"}
+    # A merged across files in index order (1 then 2)
+    assert by_id[b"series.A"]["ts"].tolist() == [t0 + 1 * 10**9,
+                                                 t0 + 3 * 10**9,
+                                                 t0 + 4 * 10**9]
+    assert by_id[b"series.A"]["vals"].tolist() == [1.0, 3.0, 4.0]
+    assert by_id[b"series.C"]["vals"].tolist() == [5.0]
