@@ -201,3 +201,88 @@ def test_ingest_parsers_never_crash(data, which):
                 _e.FilesetVolume(td, 1427162400 * 10**9).close()
             except _e.M3GpuError:
                 pass
+
+
+@settings(max_examples=150, deadline=None)
+@given(st.data())
+def test_merge_matches_python_model(data):
+    """oracle_merge (MultiReaderIterator restatement) invariants vs an
+    independent python model: output = sorted unique union of replica
+    timestamps, and every value comes from a replica holding that
+    timestamp. (WHICH replica wins a tie depends on the reference's
+    swap-with-tail values order — pinned separately by the transcribed
+    reference merge tests, not re-modeled here.)"""
+    R = data.draw(st.integers(1, 4))
+    stride = 16
+    t0 = 1700000000 * 10**9
+    reps = []
+    for r in range(R):
+        n = data.draw(st.integers(0, stride))
+        times = sorted(data.draw(st.sets(st.integers(0, 30),
+                                         min_size=n, max_size=n)))
+        vals = [data.draw(st.integers(0, 1000)) / 4.0 for _ in times]
+        reps.append((times, vals))
+    ts = np.zeros((R, 1, stride), np.int64)
+    vals = np.zeros((R, 1, stride), np.float64)
+    counts = np.zeros((R, 1), np.uint32)
+    for r, (times, vs) in enumerate(reps):
+        ts[r, 0, :len(times)] = [t0 + t * 10**9 for t in times]
+        vals[r, 0, :len(times)] = vs
+        counts[r, 0] = len(times)
+    o_ts, o_vals, o_counts, o_errs = oracle.merge_batch(
+        ts, vals, counts, out_stride=stride * R)
+    assert o_errs[0] == 0
+    candidates = {}
+    for times, vs in reps:
+        for t, v in zip(times, vs):
+            candidates.setdefault(t, set()).add(v)
+    exp_times = sorted(candidates)
+    n = int(o_counts[0])
+    assert n == len(exp_times)
+    for i in range(n):
+        t = int((o_ts[0, i] - t0) // 10**9)
+        assert t == exp_times[i]
+        assert float(o_vals[0, i]) in candidates[t]
+
+
+@settings(max_examples=120, deadline=None)
+@given(st.data())
+def test_unagg_valid_messages_roundtrip(data):
+    """Any VALID unaggregated message stream the oracle writer produces
+    parses back field-for-field (property form of test_unagg)."""
+    from oracle import unagg_writer as uw
+    from m3_amd import engine as _e
+    if not _engine_available():
+        return
+    msgs = []
+    expect = []
+    for _ in range(data.draw(st.integers(1, 8))):
+        kind = data.draw(st.integers(0, 2))
+        mid = data.draw(st.binary(min_size=1, max_size=40))
+        md = data.draw(st.binary(min_size=0, max_size=30))
+        if kind == 0:
+            v = data.draw(st.integers(-2**62, 2**62))
+            msgs.append(uw.with_metadatas(1, uw.counter(mid, v), md))
+            expect.append(("counter", mid, v, None, md))
+        elif kind == 1:
+            vs = data.draw(st.lists(st.floats(allow_nan=False,
+                                              allow_infinity=False,
+                                              width=64),
+                                    min_size=0, max_size=20))
+            msgs.append(uw.with_metadatas(2, uw.batch_timer(mid, vs), md))
+            expect.append(("batch_timer", mid, None, vs, md))
+        else:
+            v = data.draw(st.floats(allow_nan=False, allow_infinity=False,
+                                    width=64))
+            msgs.append(uw.with_metadatas(3, uw.gauge(mid, v), md))
+            expect.append(("gauge", mid, None, [v] if v != 0.0 else [], md))
+    out = _e.parse_unaggregated(uw.encode_stream(msgs))
+    assert len(out) == len(expect)
+    for m, (t, mid, cv, vs, md) in zip(out, expect):
+        assert m["type"] == t
+        assert m["id"] == mid
+        assert m["metadatas"] == md
+        if cv is not None:
+            assert m["counter_value"] == cv
+        else:
+            assert m["values"].tolist() == list(vs)
