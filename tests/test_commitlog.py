@@ -189,3 +189,20 @@ def test_commitlog_directory_merge(tmp_path):
                                                  t0 + 4 * 10**9]
     assert by_id[b"series.A"]["vals"].tolist() == [1.0, 3.0, 4.0]
     assert by_id[b"series.C"]["vals"].tolist() == [5.0]
+
+
+def test_commitlog_empty_chunk_tolerated(tmp_path):
+    """A zero-payload chunk (valid header, size 0) contributes nothing and
+    parsing continues with the next chunk."""
+    rec = clw.uvarint(len(clw.encode_log_info(5))) + clw.encode_log_info(5)
+    size0 = struct.pack("<I", 0)
+    empty_chunk = size0 + struct.pack("<I", zlib.adler32(size0)) + \
+        struct.pack("<I", zlib.adler32(b""))
+    size1 = struct.pack("<I", len(rec))
+    chunk1 = size1 + struct.pack("<I", zlib.adler32(size1)) + \
+        struct.pack("<I", zlib.adler32(rec)) + rec
+    path = tmp_path / "commitlog-0-9.db"
+    path.write_bytes(empty_chunk + chunk1 + empty_chunk)
+    with CommitLog(path) as cl:
+        assert cl.index == 5
+        assert cl.num_entries == 0

@@ -104,3 +104,14 @@ def test_unagg_unknown_fields_skipped():
 
 def test_unagg_empty_stream():
     assert parse_unaggregated(b"") == []
+
+
+def test_unagg_large_batch_timer():
+    """Packed-doubles payload beyond one length byte (500 values = 4000 B)
+    and a large id exercise multi-byte varint lengths."""
+    vals = np.arange(500, dtype=np.float64) * 0.5
+    mid = b"x" * 300  # bin16 id
+    msg = uw.with_metadatas(2, uw.batch_timer(mid, vals))
+    m = parse_unaggregated(uw.encode_stream([msg]))[0]
+    assert m["id"] == mid
+    assert np.array_equal(m["values"], vals)
