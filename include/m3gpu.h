@@ -187,6 +187,12 @@ int m3gpu_regather_dev(
     const uint64_t* d_dst_offsets, uint32_t nseries, uint8_t* d_dst,
     void* hip_stream);
 
+/* Largest bucket size for which the CKMS stream with the given eps
+ * provably never merges for this agg set (quantiles extracted, sorted
+ * unique): up to this depth the wave-tier rollup's quantiles are exact
+ * order statistics. For tests and host-side sizing. */
+int m3gpu_ckms_exact_cap(const int32_t* agg_types, int naggs, double eps);
+
 /* ======================= fileset volume reader =======================
  * Native reader for the reference's dbnode fileset volumes (persist/fs
  * read.go:145-457 + msgpack/decoder.go + digest): open a volume from a

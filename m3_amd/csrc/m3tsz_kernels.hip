@@ -2300,6 +2300,69 @@ int m3gpu_regather_dev(
     return M3GPU_OK;
 }
 
+static double agg_quantile_of_host(int32_t t) {
+    switch (t) {
+    case M3GPU_AGG_MEDIAN: case M3GPU_AGG_P50: return 0.5;
+    case M3GPU_AGG_P10: return 0.1;
+    case M3GPU_AGG_P20: return 0.2;
+    case M3GPU_AGG_P25: return 0.25;
+    case M3GPU_AGG_P30: return 0.3;
+    case M3GPU_AGG_P40: return 0.4;
+    case M3GPU_AGG_P60: return 0.6;
+    case M3GPU_AGG_P70: return 0.7;
+    case M3GPU_AGG_P75: return 0.75;
+    case M3GPU_AGG_P80: return 0.8;
+    case M3GPU_AGG_P90: return 0.9;
+    case M3GPU_AGG_P95: return 0.95;
+    case M3GPU_AGG_P99: return 0.99;
+    case M3GPU_AGG_P999: return 0.999;
+    case M3GPU_AGG_P9999: return 0.9999;
+    default: return -1.0;
+    }
+}
+
+/* Exported standalone form of the plan builder's no-compression cap (the
+ * exact simulation of stream.go:362-385 thresholds; see the inline copy
+ * in m3gpu_rollup_batch_dev_opts). For tests and host-side sizing. */
+extern "C" int m3gpu_ckms_exact_cap(const int32_t* agg_types, int naggs,
+                                    double eps) {
+    double qs[MAX_AGGS];
+    int nq = 0;
+    for (int i = 0; i < naggs && i < MAX_AGGS; i++) {
+        double q = agg_quantile_of_host(agg_types[i]);
+        if (q >= 0) qs[nq++] = q;
+    }
+    for (int i = 1; i < nq; i++) {
+        double k = qs[i];
+        int j = i - 1;
+        while (j >= 0 && qs[j] > k) { qs[j + 1] = qs[j]; j--; }
+        qs[j + 1] = k;
+    }
+    int m = 0;
+    for (int i = 0; i < nq; i++)
+        if (m == 0 || qs[m - 1] != qs[i]) qs[m++] = qs[i];
+    nq = m;
+    int cap = QCAP;
+    double eps2 = 2.0 * eps;
+    for (int v = 4; v <= QCAP + 1; v++) {
+        bool merges = false;
+        for (int mr = 0; mr <= v && !merges; mr++) {
+            long long thr = LLONG_MAX;
+            for (int i = 0; i < nq; i++) {
+                long long qmin;
+                if (mr >= (long long)(qs[i] * (double)v))
+                    qmin = (long long)(eps2 * (double)mr / qs[i]);
+                else
+                    qmin = (long long)(eps2 * (double)(v - mr) / (1.0 - qs[i]));
+                if (qmin < thr) thr = qmin;
+            }
+            if (nq > 0 && thr >= 2) merges = true;
+        }
+        if (merges) { cap = v - 1; break; }
+    }
+    return cap;
+}
+
 int m3gpu_rollup_batch_dev_opts(
     const uint8_t* d_blobs, const uint64_t* d_offsets, const uint32_t* d_lens,
     uint32_t nseries, int int_optimized, uint8_t default_unit,

@@ -60,3 +60,64 @@ def test_product_path_never_imports_oracle():
          "sorted(m for m in sys.modules if m.startswith('oracle'))" % REPO],
         capture_output=True, text=True)
     assert r.returncode == 0, r.stderr
+
+
+def test_ckms_exact_cap_matches_python_simulation():
+    """The host planner's no-compression cap (m3gpu_ckms_exact_cap) vs an
+    independent python simulation of the reference's compress thresholds
+    (stream.go:362-385, int64 truncation), for several quantile sets and
+    eps values. Cross-checked against oracle list lengths: at n = cap the
+    post-flush CKMS sample list still holds every value (no merge); the
+    cap is the last such n for the default sets probed."""
+    import ctypes
+    import numpy as np
+    import oracle
+    from m3_amd import engine
+
+    L = engine.lib()
+    L.m3gpu_ckms_exact_cap.restype = ctypes.c_int
+    L.m3gpu_ckms_exact_cap.argtypes = [ctypes.POINTER(ctypes.c_int32),
+                                       ctypes.c_int, ctypes.c_double]
+
+    def cap_c(aggs, eps):
+        a = np.asarray([engine.M3GPU_AGG[x] for x in aggs], dtype=np.int32)
+        return L.m3gpu_ckms_exact_cap(
+            a.ctypes.data_as(ctypes.POINTER(ctypes.c_int32)), len(a), eps)
+
+    QMAP = dict(median=0.5, p50=0.5, p10=0.1, p25=0.25, p75=0.75, p90=0.9,
+                p95=0.95, p99=0.99, p999=0.999, p9999=0.9999)
+
+    def cap_py(aggs, eps, qcap=512):
+        qs = sorted({QMAP[a] for a in aggs if a in QMAP})
+        if not qs:
+            return qcap
+        eps2 = 2.0 * eps
+        for v in range(4, qcap + 2):
+            for mr in range(v + 1):
+                thr = min(
+                    int(eps2 * mr / q) if mr >= int(q * v)
+                    else int(eps2 * (v - mr) / (1.0 - q))
+                    for q in qs)
+                if thr >= 2:
+                    return v - 1
+        return qcap
+
+    sets = [
+        (["median", "p95", "p99", "count", "sum"], 1e-3),
+        (["median", "p50", "p95"], 1e-3),
+        (["p9999"], 1e-3),
+        (["median"], 1e-2),
+        (["p99"], 1e-2),
+        (["count", "sum"], 1e-3),  # no quantiles -> QCAP
+    ]
+    for aggs, eps in sets:
+        assert cap_c(aggs, eps) == cap_py(aggs, eps), (aggs, eps)
+
+    # semantic cross-check vs the oracle: at n == cap a CKMS stream keeps
+    # every sample (list length n); beyond it merges begin within a few n
+    aggs, eps = ["median", "p95", "p99"], 1e-3
+    cap = cap_c(aggs, eps)
+    qs = sorted({QMAP[a] for a in aggs})
+    rng = np.random.default_rng(7)
+    vals = rng.random(cap) * 100
+    assert oracle.ckms_list_len(vals, qs, eps=eps) == cap
