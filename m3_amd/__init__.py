@@ -13,6 +13,10 @@ Fails loudly if the HIP engine is missing: there is NO CPU fallback here —
 the oracle/ package is test infrastructure only and is never imported by
 this product path.
 """
+from .iterators import (  # noqa: F401
+    BatchIterators,
+    SliceReaderIterator,
+)
 from .engine import (  # noqa: F401
     CL_ERRORS,
     CommitLog,

@@ -1,0 +1,69 @@
+"""ReaderIterator-shaped wrappers (m3_amd/iterators.py) over decoded
+batches: Next/Current/Err semantics per dbnode/encoding/types.go:197-203."""
+import numpy as np
+import pytest
+
+import oracle
+from m3_amd.iterators import BatchIterators, SliceReaderIterator
+
+START = 1427162462 * 10**9
+
+
+def test_iterator_semantics():
+    rng = np.random.default_rng(103)
+    nseries, npts = 8, 25
+    ts = START + np.cumsum(rng.integers(1, 60, (nseries, npts)), axis=1) * 10**9
+    vals = np.round(rng.random((nseries, npts)) * 100, 2)
+    counts = rng.integers(1, npts + 1, nseries)
+    its = BatchIterators(ts, vals, counts)
+    assert len(its) == nseries
+    for i, it in enumerate(its):
+        got = []
+        while it.Next():
+            t, v, u = it.Current()
+            got.append((t, v))
+            assert u == 1
+        assert it.Err() is None
+        assert got == [(int(ts[i, j]), float(vals[i, j]))
+                       for j in range(counts[i])]
+        assert not it.Next()  # exhausted stays exhausted
+        it.Close()
+
+
+def test_iterator_current_before_next():
+    it = SliceReaderIterator(np.array([1]), np.array([2.0]), 1)
+    with pytest.raises(RuntimeError):
+        it.Current()
+    assert it.Next()
+    assert it.Current() == (1, 2.0, 1)
+
+
+def test_iterator_sticky_error():
+    """A per-series error makes the iterator yield nothing and report the
+    mapped error string — like a sticky it.Err() on one reference
+    iterator."""
+    ts = np.zeros((1, 4), np.int64)
+    vals = np.zeros((1, 4), np.float64)
+    its = BatchIterators(ts, vals, np.array([4]), errs=np.array([1]))
+    it = its.iterator(0)
+    assert not it.Next()
+    assert it.Err() == "eof"
+
+
+def test_iterator_over_oracle_decode():
+    """End-to-end CPU check: encode with the oracle, decode with the
+    oracle, iterate with the reference-shaped wrapper."""
+    rng = np.random.default_rng(107)
+    npts = 40
+    ts = START + np.cumsum(rng.integers(1, 30, npts)) * 10**9
+    vals = np.round(rng.random(npts) * 10, 1)
+    blob = oracle.encode_series(ts, vals, start_ns=int(ts[0]))
+    from m3_amd.engine import pack_streams
+    b, off, lens = pack_streams([blob])
+    o_ts, o_vals, o_counts = oracle.decode_batch(b, off, stride=npts + 4)
+    it = BatchIterators(o_ts, o_vals, o_counts).iterator(0)
+    out = []
+    while it.Next():
+        t, v, _ = it.Current()
+        out.append((t, v))
+    assert out == list(zip(ts.tolist(), vals.tolist()))
