@@ -281,6 +281,12 @@ int m3gpu_unagg_metric(int handle, int64_t i, int32_t* union_type,
                        const uint8_t** annotation, int64_t* annotation_len,
                        const uint8_t** metadatas, int64_t* metadatas_len);
 int m3gpu_unagg_values(int handle, int64_t i, double* out);
+/* Field-separated wrapper metadata (StagedMetadatas field 2, StoragePolicy
+ * field 3, ...): `metadatas` above is a concatenation and cannot be split
+ * back into the individual reference protos. */
+int64_t m3gpu_unagg_metadata_count(int handle, int64_t i);
+int m3gpu_unagg_metadata(int handle, int64_t i, int64_t j, int32_t* field,
+                         const uint8_t** bytes, int64_t* len);
 
 enum {
     M3GPU_UA_ERR_TRUNCATED = -121,

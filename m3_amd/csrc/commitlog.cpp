@@ -208,6 +208,9 @@ static bool cl_uvarint(const std::vector<uint8_t>& s, size_t* pos, uint64_t* out
     for (int i = 0; i < 10; i++) {
         if (*pos >= s.size()) return false;
         uint8_t b = s[(*pos)++];
+        /* Go binary.ReadUvarint overflow rule: the 10th byte may only
+         * contribute bit 63, i.e. must be <= 1 */
+        if (i == 9 && b > 1) return false;
         if (b < 0x80) {
             *out = v | ((uint64_t)b << shift);
             return true;
@@ -269,7 +272,8 @@ int m3gpu_commitlog_open(const char* path) {
             delete file;
             return cl_fail(M3GPU_CL_ERR_TRUNCATED, "truncated record size");
         }
-        if (sp + rec_len > stream.size()) {
+        /* subtraction form: sp + rec_len wraps for rec_len near 2^64 */
+        if (rec_len > stream.size() - sp) {
             delete file;
             return cl_fail(M3GPU_CL_ERR_TRUNCATED, "truncated record");
         }

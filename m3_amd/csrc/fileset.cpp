@@ -247,6 +247,11 @@ struct FsEntry {
 static int fs_parse_entries(const uint8_t* buf, size_t len, int64_t count,
                             std::vector<FsEntry>* out) {
     MsgRd rd{buf, len};
+    /* `count` comes from the (digest-valid but possibly corrupt) info file:
+     * bound it against the index file size before reserving — each entry
+     * occupies at least ~8 bytes, so count > len is always a schema error
+     * and an unbounded reserve() would throw past the C ABI. */
+    if (count < 0 || (uint64_t)count > (uint64_t)len) return M3GPU_FS_ERR_SCHEMA;
     out->reserve((size_t)count);
     for (int64_t e = 0; e < count; e++) {
         size_t entry_start = rd.pos;

@@ -56,6 +56,9 @@ struct UaRd {
         for (int i = 0; i < 10; i++) {
             if (pos >= n) { err = M3GPU_UA_ERR_TRUNCATED; return false; }
             uint8_t b = p[pos++];
+            /* 10th byte may only contribute bit 63 (Go binary.ReadUvarint
+             * overflow rule): anything above 1 overflows uint64 */
+            if (i == 9 && b > 1) { err = M3GPU_UA_ERR_PROTO; return false; }
             v |= (uint64_t)(b & 0x7f) << shift;
             if (!(b & 0x80)) { *out = v; return true; }
             shift += 7;
@@ -74,7 +77,8 @@ struct UaRd {
     bool bytes(const uint8_t** out, uint64_t* len) {
         uint64_t l;
         if (!u64(&l)) return false;
-        if (pos + l > n) { err = M3GPU_UA_ERR_TRUNCATED; return false; }
+        /* subtraction form: pos + l wraps for l near 2^64 (pos <= n always) */
+        if (l > n - pos) { err = M3GPU_UA_ERR_TRUNCATED; return false; }
         *out = p + pos;
         *len = l;
         pos += l;
@@ -96,6 +100,14 @@ struct UaRd {
     }
 };
 
+/* one retained wrapper-level length-delimited field (StagedMetadatas,
+ * StoragePolicy, ...), kept with its field number so downstream rule
+ * matching can recover the individual reference protos */
+struct UaMetaField {
+    uint32_t field;
+    std::vector<uint8_t> bytes;
+};
+
 struct UaMetric {
     int32_t union_type;
     int32_t metric_type;    /* metricpb.MetricType for timed metrics */
@@ -104,7 +116,8 @@ struct UaMetric {
     int64_t counter_value = 0;    /* exact int64 for counters */
     int64_t time_nanos = 0;       /* client_time (untimed) or time (timed) */
     std::vector<uint8_t> annotation;
-    std::vector<uint8_t> metadatas; /* opaque metadatas/policy bytes */
+    std::vector<uint8_t> metadatas;    /* concatenation (back-compat view) */
+    std::vector<UaMetaField> meta_fields; /* field-separated segments */
 };
 
 struct UaBatch {
@@ -182,9 +195,14 @@ static bool ua_parse_union(UaRd& rd, UaMetric& m, bool timed) {
             if (!rd.bytes(&b, &bl)) return false;
             UaRd sub{b, (size_t)bl};
             if (!ua_parse_metric(sub, m, timed)) { rd.err = sub.err; return false; }
-        } else if (wt == 2) { /* metadatas / storage policy: keep opaque */
+        } else if (wt == 2) { /* metadatas / storage policy: keep opaque,
+                               * field-separated (plus the legacy concat) */
             if (!rd.bytes(&b, &bl)) return false;
             m.metadatas.insert(m.metadatas.end(), b, b + bl);
+            UaMetaField mf;
+            mf.field = f;
+            mf.bytes.assign(b, b + bl);
+            m.meta_fields.push_back(std::move(mf));
         } else {
             if (!rd.skip(wt)) return false;
         }
@@ -216,7 +234,7 @@ int m3gpu_unagg_parse(const uint8_t* buf, uint64_t len) {
         int64_t size = (int64_t)(uv >> 1);
         if (uv & 1) size = ~size;
         if (size <= 0) { delete batch; return ua_fail(M3GPU_UA_ERR_SIZE, "non-positive size"); }
-        if (pos + (uint64_t)size > len) {
+        if ((uint64_t)size > len - pos) {
             delete batch;
             return ua_fail(M3GPU_UA_ERR_TRUNCATED, "message body");
         }
@@ -309,6 +327,30 @@ int m3gpu_unagg_metric(int h, int64_t i, int32_t* union_type,
     if (annotation_len) *annotation_len = (int64_t)m.annotation.size();
     if (metadatas) *metadatas = m.metadatas.data();
     if (metadatas_len) *metadatas_len = (int64_t)m.metadatas.size();
+    return 0;
+}
+
+/* Field-separated wrapper metadata access (StagedMetadatas field 2,
+ * StoragePolicy field 3, RoutingPolicy, unknown retained fields): the
+ * concatenated `metadatas` view cannot be split back into protos. */
+int64_t m3gpu_unagg_metadata_count(int h, int64_t i) {
+    UaBatch* b = ua_get(h);
+    if (!b) return M3GPU_UA_ERR_BADHANDLE;
+    if (i < 0 || (size_t)i >= b->metrics.size()) return M3GPU_UA_ERR_BADHANDLE;
+    return (int64_t)b->metrics[(size_t)i].meta_fields.size();
+}
+
+int m3gpu_unagg_metadata(int h, int64_t i, int64_t j, int32_t* field,
+                         const uint8_t** bytes, int64_t* len) {
+    UaBatch* b = ua_get(h);
+    if (!b) return M3GPU_UA_ERR_BADHANDLE;
+    if (i < 0 || (size_t)i >= b->metrics.size()) return M3GPU_UA_ERR_BADHANDLE;
+    const UaMetric& m = b->metrics[(size_t)i];
+    if (j < 0 || (size_t)j >= m.meta_fields.size()) return M3GPU_UA_ERR_BADHANDLE;
+    const UaMetaField& mf = m.meta_fields[(size_t)j];
+    if (field) *field = (int32_t)mf.field;
+    if (bytes) *bytes = mf.bytes.data();
+    if (len) *len = (int64_t)mf.bytes.size();
     return 0;
 }
 

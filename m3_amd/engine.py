@@ -655,6 +655,11 @@ def parse_unaggregated(buf):
                                          P(c_i64), P(c_vp), P(c_i64)]
         L.m3gpu_unagg_values.restype = c_int
         L.m3gpu_unagg_values.argtypes = [c_int, c_i64, P(c_f64)]
+        L.m3gpu_unagg_metadata_count.restype = c_i64
+        L.m3gpu_unagg_metadata_count.argtypes = [c_int, c_i64]
+        L.m3gpu_unagg_metadata.restype = c_int
+        L.m3gpu_unagg_metadata.argtypes = [c_int, c_i64, c_i64, P(c_i32),
+                                           P(c_vp), P(c_i64)]
         L._ua_configured = True
     b = np.frombuffer(bytes(buf), dtype=np.uint8)
     h = L.m3gpu_unagg_parse(b.ctypes.data_as(P(c_u8)), b.nbytes)
@@ -677,7 +682,18 @@ def parse_unaggregated(buf):
             vals = np.empty(nv.value, np.float64)
             if nv.value:
                 L.m3gpu_unagg_values(h, i, vals.ctypes.data_as(P(c_f64)))
+            meta_fields = []
+            for j in range(L.m3gpu_unagg_metadata_count(h, i)):
+                fnum, fp, fl = c_i32(), c_vp(), c_i64()
+                rc = L.m3gpu_unagg_metadata(
+                    h, i, j, ctypes.byref(fnum), ctypes.byref(fp),
+                    ctypes.byref(fl))
+                _check(rc, "m3gpu_unagg_metadata")
+                meta_fields.append(
+                    (fnum.value,
+                     ctypes.string_at(fp, fl.value) if fl.value else b""))
             out.append(dict(
+                metadata_fields=meta_fields,
                 type=UA_TYPES.get(ut.value, ut.value),
                 metric_type=mt.value,
                 id=ctypes.string_at(idp, idl.value) if idl.value else b"",

@@ -206,3 +206,25 @@ def test_commitlog_empty_chunk_tolerated(tmp_path):
     with CommitLog(path) as cl:
         assert cl.index == 5
         assert cl.num_entries == 0
+
+
+def test_commitlog_maximal_varint_record_len(tmp_path):
+    """A checksum-valid chunk whose record-length uvarint encodes ~2^64
+    must fail with a clean truncation error, not wrap the bounds check
+    and read past the payload (commitlog.cpp ClRd overflow fix)."""
+    payload = clw.uvarint((1 << 64) - 1)  # 10-byte maximal uvarint
+    size = struct.pack("<I", len(payload))
+    chunk = size + struct.pack("<I", zlib.adler32(size)) + \
+        struct.pack("<I", zlib.adler32(payload)) + payload
+    path = tmp_path / "commitlog-0-0.db"
+    path.write_bytes(chunk)
+    with pytest.raises(M3GpuError, match="truncated"):
+        CommitLog(path)
+    # a uvarint whose 10th byte exceeds 1 overflows uint64 entirely
+    payload = b"\xff" * 9 + b"\x02"
+    size = struct.pack("<I", len(payload))
+    chunk = size + struct.pack("<I", zlib.adler32(size)) + \
+        struct.pack("<I", zlib.adler32(payload)) + payload
+    path.write_bytes(chunk)
+    with pytest.raises(M3GpuError, match="truncated"):
+        CommitLog(path)

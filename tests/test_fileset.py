@@ -206,3 +206,27 @@ def test_fileset_entry_checksum_mismatch(tmp_path):
     # info still says 2 entries; first entry fails its checksum
     with pytest.raises(M3GpuError, match="entry_checksum|msgpack"):
         FilesetVolume(str(tmp_path), BLOCK_START)
+
+
+def test_fileset_huge_entries_count_rejected(tmp_path):
+    """A digest-valid volume whose info file claims entries=2^62 must
+    return a schema error instead of letting an unbounded reserve()
+    throw past the C ABI (fileset.cpp fs_parse_entries bound)."""
+    rng = np.random.default_rng(17)
+    series, _ = make_series(rng, 3)
+    fsw.write_volume(str(tmp_path), BLOCK_START, series)
+
+    def p(suffix):
+        return tmp_path / f"fileset-{BLOCK_START}-0-{suffix}.db"
+
+    info = fsw.encode_index_info(BLOCK_START, 2 * 3600 * 10**9, 1 << 62,
+                                 1, 64, 3)
+    p("info").write_bytes(info)
+    # recompute digests + checkpoint so only the count is wrong
+    names = ("info", "index", "summaries", "bloomfilter", "data")
+    dig = b"".join(struct.pack("<I", fsw.digest32(p(s).read_bytes()))
+                   for s in names)
+    p("digest").write_bytes(dig)
+    p("checkpoint").write_bytes(struct.pack("<I", fsw.digest32(dig)))
+    with pytest.raises(M3GpuError, match="schema"):
+        FilesetVolume(str(tmp_path), BLOCK_START)

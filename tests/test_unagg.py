@@ -115,3 +115,43 @@ def test_unagg_large_batch_timer():
     m = parse_unaggregated(uw.encode_stream([msg]))[0]
     assert m["id"] == mid
     assert np.array_equal(m["values"], vals)
+
+
+def test_unagg_maximal_varint_length_rejected():
+    """A bytes-field length encoded as a maximal 10-byte varint (~2^64)
+    must fail with a clean truncation/proto error, not wrap the bounds
+    check and abort the process (unagg.cpp UaRd::bytes overflow fix)."""
+    # counter payload: field 1 (id, wt=2) with length 2^64-1
+    max_len = uw.pv_uvarint((1 << 64) - 1)
+    assert len(max_len) == 10 and max_len[-1] == 1
+    bad_counter = uw.pv_uvarint(1 << 3 | 2) + max_len  # no body follows
+    msg = uw.with_metadatas(1, bad_counter)
+    with pytest.raises(M3GpuError, match="truncated|proto"):
+        parse_unaggregated(uw.encode_stream([msg]))
+    # 10th byte > 1 exceeds uint64 (Go binary.ReadUvarint overflow rule)
+    over = b"\xff" * 9 + b"\x02"
+    bad2 = uw.pv_uvarint(1 << 3 | 2) + over
+    with pytest.raises(M3GpuError, match="truncated|proto"):
+        parse_unaggregated(uw.encode_stream([uw.with_metadatas(1, bad2)]))
+    # same class inside skip() of an unknown length-delimited field
+    base = uw.counter(b"c", 7) + uw.pv_uvarint(9 << 3 | 2) + max_len
+    with pytest.raises(M3GpuError, match="truncated|proto"):
+        parse_unaggregated(uw.encode_stream([uw.with_metadatas(1, base)]))
+
+
+def test_unagg_metadata_fields_separated():
+    """StagedMetadatas (field 2) and StoragePolicy (field 3) of a
+    TimedMetricWithStoragePolicy come back as separate tagged segments,
+    not only as one unsplittable concatenation."""
+    sp = b"\x08\x0a\x10\x3c"  # opaque fake StoragePolicy proto bytes
+    md = b"\x0a\x05hello"
+    payload = (uw.pv_uvarint(1 << 3 | 2) +
+               uw.pv_uvarint(len(uw.timed_metric(2, b"tm", 5, 1.5))) +
+               uw.timed_metric(2, b"tm", 5, 1.5) +
+               uw.pv_uvarint(2 << 3 | 2) + uw.pv_uvarint(len(md)) + md +
+               uw.pv_uvarint(3 << 3 | 2) + uw.pv_uvarint(len(sp)) + sp)
+    msg = (uw.pv_varint_field(1, 7) + uw.pv_uvarint(8 << 3 | 2) +
+           uw.pv_uvarint(len(payload)) + payload)
+    m = parse_unaggregated(uw.encode_stream([msg]))[0]
+    assert m["metadatas"] == md + sp  # legacy concat view unchanged
+    assert m["metadata_fields"] == [(2, md), (3, sp)]
