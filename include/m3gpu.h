@@ -21,9 +21,11 @@
  *  - Encoded streams are the reference's on-disk M3TSZ block format
  *    (src/dbnode/encoding/m3tsz, incl. the EOS-marker tail), bit-exact.
  *  - Packed stream layout: blobs[offsets[i] .. offsets[i]+lens[i]) is series
- *    i's stream. offsets[i] MUST be 16-byte aligned (pad between streams with
- *    zero bytes; the total buffer padded to a multiple of 16) so stream
- *    refills are single aligned 16B loads. lens[] are true stream lengths.
+ *    i's stream. The kernels REQUIRE offsets[i] to be 8-byte aligned and
+ *    every stream zero-padded to an 8-byte boundary (refills are aligned
+ *    u64 loads; the zero pad reproduces reader64's zero-filled tail word).
+ *    pack_streams/compact emit 16-byte alignment, which satisfies this.
+ *    lens[] are true stream lengths.
  *  - Time units use the reference's xtime.Unit byte values
  *    (src/x/time/unit.go:30-42): 1=s, 2=ms, 3=us, 4=ns.
  */

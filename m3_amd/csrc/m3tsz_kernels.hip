@@ -19,8 +19,9 @@
  *      rollup: bucket values are staged to LDS, quantiles picked by a
  *        rank-select over lanes (no sort network needed), reproducing the
  *        reference CKMS walk exactly (see ckms_small_n semantics below).
- *  - Streams are decoded from 8-byte-aligned offsets (layout contract in
- *    m3gpu.h) so every refill is one aligned dwordx2 load.
+ *  - Streams are decoded from 8-byte-aligned offsets, zero-padded to an
+ *    8-byte boundary (layout contract in m3gpu.h; pack_streams emits 16B
+ *    alignment, which satisfies it) so every refill is an aligned u64 load.
  *  - Wave64 only; no CUDA shims, no hipified code.
  *
  * Compile: hipcc --offload-arch=gfx950 -O3 -ffp-contract=off (bit-exact f64:
@@ -91,85 +92,131 @@ __device__ __forceinline__ double go_modf(double v, double* ip) {
 /* istream.go:73-115 over reader64.go:40-80, with the m3gpu.h layout
  * contract: stream starts 8B-aligned, buffer zero-padded to 8B. */
 
-struct BitReader {
-    /* Register bit window with a cheap common path (bits served from `cur`)
-     * and a two-deep word pipeline: `nxt` is the word the reference reader
-     * would fetch next (so EOF semantics match istream.go:73-115 over
-     * reader64.go:40-80 exactly), `pf` is prefetched a further word ahead —
-     * its load is issued ~128 bits before first use, hiding refill latency
-     * under parsing. The blob's zero padding reproduces reader64's
-     * zero-filled partial tail word. */
-    const uint64_t* words;
-    int64_t len;        /* true byte length */
-    int64_t index;      /* byte index of the next word to prefetch */
-    uint64_t cur;       /* left-aligned buffered bits */
-    uint32_t rem;       /* valid bits in cur */
-    uint64_t nxt;       /* next word (left-aligned) */
-    uint32_t nxt_bits;
-    uint64_t pf;        /* prefetched word after nxt */
-    uint32_t pf_bits;
+/* Per-lane LDS input ring geometry (decode kernel): 16 buffered words per
+ * lane, rows padded to 17 so that both the per-lane ds_read_b64 pulls and
+ * the refill ds_write_b64 bursts are bank-conflict-free (b64 reads bank on
+ * (a/4)%64 in 2x32 lane groups, writes on (a/4)%32 in 4x16 groups; the
+ * 17-word row stride makes lane bases land on distinct bank pairs). */
+#define IN_WORDS 16
+#define IN_STRIDE 17
 
-    __device__ __forceinline__ void prefetch() {
-        if (index < len) {
-            pf = __builtin_bswap64(words[index >> 3]);
-            int64_t avail = len - index;
-            pf_bits = avail >= 8 ? 64 : (uint32_t)(8 * avail);
-            index += 8;
+struct BitReader {
+    /* 128-bit register lookahead (`a` = next bits, `b` = following word,
+     * both byte-swapped to stream bit order; `p` = bits of `a` already
+     * consumed): peek64() is branch-free, consume() crosses at most one
+     * word per call (every field is <= 64 bits). EOF semantics match
+     * istream.go:73-115 over reader64.go:40-80: `bits_left` counts the
+     * un-consumed TRUE stream bits, reads past it fail per series; the
+     * blob's zero padding reproduces reader64's zero-filled partial tail
+     * word, so the window itself can always hold 128 physical bits.
+     *
+     * Words arrive one of two ways:
+     *  - LDS ring (decode kernel, `lds` set): refill() tops the per-lane
+     *    ring up to 16 words once per 8-point tile — each lane issues up
+     *    to 8 back-to-back loads of CONSECUTIVE words of its own stream,
+     *    so the per-lane 64B line is fetched once and consumed fully
+     *    (L1-amortized), and the long-latency global gathers leave the
+     *    per-point dependency chain entirely: the parser's word pulls are
+     *    cheap ds_read_b64s.
+     *  - direct (`lds` null, rollup/merge kernels): a 1-deep register
+     *    prefetch (`pfw`) beyond the window, i.e. 3 words of lookahead,
+     *    keeps the refill load issued ~128 bits before first use. */
+    const uint64_t* words;
+    uint64_t a, b;
+    uint32_t p;
+    int64_t bits_left;  /* un-consumed stream bits (from the true byte len) */
+    uint32_t wnext;     /* next word index to pull into the window */
+    uint32_t wtotal;    /* ceil(len/8) */
+    uint64_t* lds;      /* this lane's ring base, or nullptr */
+    uint32_t rfill;     /* words fetched into the ring so far */
+    uint64_t pfw;       /* direct mode: 1-deep prefetched word */
+
+    __device__ __forceinline__ uint64_t next_word() {
+        uint64_t w;
+        if (lds) {
+            /* ring hit is the steady state (refill() runs per tile); a
+             * mid-tile underrun (adversarially dense streams consuming
+             * >16 words in 8 points) falls back to a direct load —
+             * correct, just slower */
+            if (wnext < rfill) w = lds[wnext & (IN_WORDS - 1)];
+            else w = (wnext < wtotal) ? __builtin_bswap64(words[wnext]) : 0;
         } else {
-            pf = 0;
-            pf_bits = 0;
+            w = pfw;
+            uint32_t nn = wnext + 1;
+            pfw = (nn < wtotal) ? __builtin_bswap64(words[nn]) : 0;
         }
+        wnext++;
+        return w;
     }
-    __device__ __forceinline__ void init(const uint8_t* base, uint64_t off, uint32_t l) {
+
+    /* Top the ring up to IN_WORDS words (whole-word granularity; streams
+     * are zero-padded to 8B, §m3gpu.h layout). Call once per tile with all
+     * lanes converged; `active=false` lanes stop fetching. Two batches of
+     * up to 8 loads each: loads issue back-to-back into registers, then
+     * write to LDS — one memory round-trip per tile, not per word. */
+    __device__ __forceinline__ void refill(bool active) {
+        if (!lds) return;
+        uint32_t used = rfill - wnext;
+        uint32_t need = active ? (IN_WORDS - used) : 0;
+        uint32_t avail = wtotal - rfill;
+        if (need > avail) need = avail;
+        uint64_t v[8];
+#pragma unroll
+        for (uint32_t i = 0; i < 8; i++)
+            if (i < need) v[i] = __builtin_bswap64(words[rfill + i]);
+#pragma unroll
+        for (uint32_t i = 0; i < 8; i++)
+            if (i < need) lds[(rfill + i) & (IN_WORDS - 1)] = v[i];
+        if (__any(need > 8)) {
+            uint64_t v2[8];
+#pragma unroll
+            for (uint32_t i = 0; i < 8; i++)
+                if (i + 8 < need) v2[i] = __builtin_bswap64(words[rfill + i + 8]);
+#pragma unroll
+            for (uint32_t i = 0; i < 8; i++)
+                if (i + 8 < need) lds[(rfill + i + 8) & (IN_WORDS - 1)] = v2[i];
+        }
+        rfill += need;
+    }
+
+    __device__ __forceinline__ void init(const uint8_t* base, uint64_t off,
+                                         uint32_t l, uint64_t* ring) {
         words = (const uint64_t*)(base + off);
-        len = l;
-        index = 0;
-        cur = 0;
-        rem = 0;
-        prefetch();
-        nxt = pf;
-        nxt_bits = pf_bits;
-        prefetch();
+        wtotal = (l + 7) >> 3;
+        bits_left = (int64_t)l * 8;
+        lds = ring;
+        wnext = 0;
+        rfill = 0;
+        if (ring) refill(true);
+        else pfw = wtotal ? __builtin_bswap64(words[0]) : 0;
+        a = next_word();
+        b = next_word();
+        p = 0;
+    }
+
+    /* next 64 physical bits (zero-padded past the stream tail); branch-free */
+    __device__ __forceinline__ uint64_t peek64() const {
+        return (a << p) | (p ? (b >> (64 - p)) : 0);
+    }
+    /* advance n <= 64 bits already validated against bits_left */
+    __device__ __forceinline__ void consume(uint32_t n) {
+        bits_left -= n;
+        p += n;
+        if (p >= 64) {
+            a = b;
+            b = next_word();
+            p -= 64;
+        }
     }
     __device__ __forceinline__ int read_bits(uint32_t n, uint64_t* out) {
-        if (n <= rem) {
-            *out = n ? (cur >> (64 - n)) : 0;
-            cur = (n >= 64) ? 0 : (cur << n);
-            rem -= n;
-            return 0;
-        }
-        uint32_t need = n - rem;
-        if (nxt_bits < need) return M3GPU_SERIES_EOF;
-        uint64_t res = rem ? (cur >> (64 - rem)) : 0;
-        res = ((need >= 64) ? 0 : (res << need)) | (nxt >> (64 - need));
-        cur = (need >= 64) ? 0 : (nxt << need);
-        rem = nxt_bits - need;
-        nxt = pf;
-        nxt_bits = pf_bits;
-        prefetch();
-        *out = res;
+        if (bits_left < (int64_t)n) return M3GPU_SERIES_EOF;
+        *out = n ? (peek64() >> (64 - n)) : 0;
+        consume(n);
         return 0;
     }
-    /* consume n bits already validated by a successful peek_bits(n) */
-    __device__ __forceinline__ void consume(uint32_t n) {
-        if (n <= rem) {
-            cur = (n >= 64) ? 0 : (cur << n);
-            rem -= n;
-            return;
-        }
-        uint32_t need = n - rem;
-        cur = (need >= 64) ? 0 : (nxt << need);
-        rem = nxt_bits - need;
-        nxt = pf;
-        nxt_bits = pf_bits;
-        prefetch();
-    }
     __device__ __forceinline__ int peek_bits(uint32_t n, uint64_t* out) {
-        if (n <= rem) { *out = n ? (cur >> (64 - n)) : 0; return 0; }
-        uint32_t need = n - rem;
-        if (nxt_bits < need) return M3GPU_SERIES_EOF;
-        uint64_t res = rem ? (cur >> (64 - rem)) : 0;
-        *out = ((need >= 64) ? 0 : (res << need)) | (nxt >> (64 - need));
+        if (bits_left < (int64_t)n) return M3GPU_SERIES_EOF;
+        *out = n ? (peek64() >> (64 - n)) : 0;
         return 0;
     }
 };
@@ -181,6 +228,7 @@ struct BitReader {
 struct Decoder {
     BitReader r;
     int64_t prev_time, prev_time_delta;
+    int64_t unit_ns; /* cached UNIT_NS_D[time_unit] (0 when unit invalid) */
     double int_val;
     uint64_t prev_float_bits, prev_xor;
     uint8_t time_unit, scheme_unit, mult, sig;
@@ -188,12 +236,17 @@ struct Decoder {
     bool have_scheme, tu_changed, done, is_float;
     bool int_optimized;
 
+    __device__ __forceinline__ void set_unit(uint8_t tu) {
+        time_unit = tu;
+        unit_ns = unit_valid(tu) ? UNIT_NS_D[tu] : 0;
+    }
+
     __device__ __forceinline__ void init(const uint8_t* base, uint64_t off, uint32_t len,
-                         bool intopt, uint8_t dunit) {
-        r.init(base, off, len);
+                         bool intopt, uint8_t dunit, uint64_t* ring = nullptr) {
+        r.init(base, off, len, ring);
         prev_time = 0; prev_time_delta = 0;
         int_val = 0; prev_float_bits = 0; prev_xor = 0;
-        time_unit = 0; scheme_unit = 0; mult = 0; sig = 0;
+        time_unit = 0; unit_ns = 0; scheme_unit = 0; mult = 0; sig = 0;
         default_unit = dunit;
         have_scheme = false; tu_changed = false; done = false; is_float = false;
         int_optimized = intopt;
@@ -209,7 +262,7 @@ struct Decoder {
             tu_changed = true;
             if (scheme_default_bits(tu)) { have_scheme = true; scheme_unit = tu; }
         }
-        time_unit = tu;
+        set_unit(tu);
         return 0;
     }
 
@@ -371,7 +424,7 @@ struct Decoder {
         int64_t nt = (int64_t)nt_bits;
         if (time_unit == 0 && unit_valid(default_unit) &&
             nt % UNIT_NS_D[default_unit] == 0) {
-            time_unit = default_unit;
+            set_unit(default_unit);
         }
         if (scheme_default_bits(time_unit)) { have_scheme = true; scheme_unit = time_unit; }
         int64_t dod;
@@ -578,59 +631,75 @@ struct Decoder {
         return 0;
     }
 
+    /* Branch-minimized fast path: ONE branch-free 64-bit peek classifies
+     * the timestamp field (dod==0 | bucket 0-2) arithmetically — no
+     * per-bucket branching — and the int-mode value field (no-update diff
+     * | repeat) with pure select chains, then a single consume advances
+     * the window. Per-lane divergence survives only at: the int/float
+     * mode split, the rare escapes (markers, default buckets, sig/mode
+     * updates, near-EOS) which return -1000 and fall to the stepwise
+     * path (bit-identical semantics), and the window crossing inside
+     * consume(). Grammar consequence (i), SURVEY.md Appendix A. */
     __device__ __forceinline__ int next_fused(int64_t* t, double* v) {
-        uint64_t w;
-        if (tu_changed || !have_scheme || prev_time == 0 ||
-            r.peek_bits(64, &w) != 0)
+        if (tu_changed || !have_scheme || prev_time == 0 || r.bits_left < 64)
             return -1000;
-        uint32_t c1;
-        int64_t dod;
-        if (!(w >> 63)) {
-            c1 = 1;
-            dod = 0;
-        } else {
-            if ((w >> 55) == MARKER_OPCODE) return -1000; /* marker */
-            uint32_t top4 = (uint32_t)(w >> 60);
-            uint32_t L = __builtin_clz(~(top4 << 28)); /* leading ones, 1..4 */
-            if (L >= 4) return -1000; /* default bucket: stepwise */
-            uint32_t vb = (L == 1) ? 7 : (L == 2) ? 9 : 12;
-            uint32_t ob = L + 1;
-            dod = sign_extend((w << ob) >> (64 - vb), vb) * UNIT_NS_D[time_unit];
-            c1 = ob + vb;
+        const uint64_t w = r.peek64();
+        /* ---- timestamp: '0' => dod 0; '1'^L 0 + {7,9,12}b => bucket ---- */
+        const uint64_t b0 = w >> 63;
+        const uint32_t L = (uint32_t)__builtin_clzll(~w | 1); /* leading 1s */
+        if (L >= 4 || (w >> 55) == MARKER_OPCODE)
+            return -1000; /* default bucket / EOS|annotation|timeunit marker */
+        const uint32_t vb = (0xC970u >> (L * 4)) & 0xFu; /* {0,7,9,12} */
+        const uint32_t tsbits = b0 ? (L + 1 + vb) : 1;
+        const uint32_t vbs = b0 ? vb : 7; /* shift-safe stand-in when dod==0 */
+        int64_t dod = sign_extend((w << (L + 1)) >> (64 - vbs), vbs) * unit_ns;
+        dod = b0 ? dod : 0; /* unit_ns==0 swallows invalid units (:271-274) */
+        const uint64_t w2 = w << tsbits; /* tsbits <= 16 here */
+
+        if (int_optimized && !is_float) {
+            /* int mode: '1' sign+sig diff | '01' repeat | '00...' stepwise */
+            const uint64_t v0 = w2 >> 63;
+            if (!v0 && ((w2 >> 62) & 1) == 0)
+                return -1000; /* mode/sig/mult update */
+            if (v0 && (sig > 45u || tsbits + 2u + sig > 64u))
+                return -1000; /* wide sig: field may exceed the peek */
+            const uint32_t nb = v0 ? (2u + sig) : 2u;
+            /* shifts masked &63: on repeat lanes sig may be 64 (slow-width
+             * state) and the field value is select-discarded, but the
+             * shift amounts must stay defined */
+            const uint32_t sg = sig & 63u;
+            const uint64_t field = (w2 << 1) >> ((63u - sg) & 63u);
+            const uint64_t sbit = (field >> sg) & 1;
+            const uint64_t mag = field ^ (sbit << sg);
+            const double add = (sbit ? 1.0 : -1.0) * (double)mag;
+            /* repeat must leave int_val EXACTLY unchanged (adding 0.0 would
+             * flip -0.0; DESIGN.md §6 repeat quirk) — select, don't add */
+            if (v0) int_val += add;
+            r.consume(tsbits + nb);
+            prev_time_delta += dod;
+            prev_time += prev_time_delta;
+            *t = prev_time;
+            *v = (mult == 0) ? int_val : int_val / exp10_table(mult);
+            return 1;
         }
-        uint64_t w2 = w << c1;
-        if (!int_optimized) { /* pure float stream: XOR field directly */
-            int rx = fused_xor(w2, c1);
-            if (rx == -1000) return -1000;
-            if (rx) return rx; /* negative error, same point as stepwise */
-        } else if (is_float) {
-            /* float mode: '1' + XOR | '01' repeat | '00...' stepwise */
-            if (w2 >> 63) {
-                int rx = fused_xor(w2 << 1, c1 + 1);
-                if (rx == -1000) return -1000;
-                if (rx) return rx;
-            } else if ((w2 >> 62) == 0x1) {
-                r.consume(c1 + 2);
-            } else {
-                return -1000;
-            }
-        } else if (w2 >> 63) { /* int mode, opcodeNoUpdate: sign + sig diff */
-            if (sig > 45 || c1 + 2 + sig > 64) return -1000;
-            uint64_t bits = (w2 << 1) >> (63 - sig);
-            r.consume(c1 + 2 + sig);
-            double sgn = -1.0;
-            if ((bits >> sig) == 1) { sgn = 1.0; bits ^= (1ULL << sig); }
-            int_val += sgn * (double)bits;
-        } else if ((w2 >> 62) == 0x1) { /* repeat */
-            r.consume(c1 + 2);
+        /* float value (pure float stream, or int-optimized float mode) */
+        int rx;
+        if (!int_optimized) {
+            rx = fused_xor(w2, tsbits);
+        } else if (w2 >> 63) { /* '1' + XOR */
+            rx = fused_xor(w2 << 1, tsbits + 1);
+        } else if ((w2 >> 62) & 1) { /* '01' repeat */
+            r.consume(tsbits + 2);
+            rx = 0;
         } else {
-            return -1000; /* '00': mode/sig/mult update -> stepwise */
+            return -1000; /* '00...' mode change: stepwise */
         }
+        if (rx == -1000) return -1000; /* nothing consumed */
+        if (rx) return rx; /* negative error, same EOF point as stepwise */
         prev_time_delta += dod;
         prev_time += prev_time_delta;
         *t = prev_time;
-        if (!int_optimized || is_float) *v = bits2f(prev_float_bits);
-        else *v = (mult == 0) ? int_val : int_val / exp10_table(mult);
+        *v = bits2f(prev_float_bits);
         return 1;
     }
 
@@ -661,14 +730,24 @@ struct Decoder {
  * one point per active lane per iteration. The VLC parser state lives in
  * per-lane VGPRs; data-dependent branches diverge only where lanes disagree
  * (host-side batches group similar series for lane coherence, but any order
- * is correct). Decoded points stage through an LDS tile and flush as
- * line-coalesced row segments every DEC_TILE points:
- *   flush step j: lane l stores row (l>>3)+8j, point (l&7) — 8 consecutive
- *   8B addresses per row = full 64B line utilization for ts[] and val[]. */
+ * is correct).
+ *
+ * Memory structure (v6):
+ *  - INPUT through a per-lane LDS ring (BitReader::refill): the global
+ *    gathers leave the per-point chain — once per 8-point tile each lane
+ *    bursts up to 8 loads of consecutive words of its own stream (per-lane
+ *    64B lines fetched once, L1-amortized), and the parser pulls words
+ *    with cheap conflict-free ds_read_b64s.
+ *  - OUTPUT accumulated in per-lane REGISTERS (8 ts + 8 val slots, indexed
+ *    by the unrolled tile counter) and flushed as 4+4 back-to-back 16B
+ *    stores into the lane's own rows: consecutive stores complete each
+ *    64B line, so write-combining holds WRITE_SIZE at the algorithmic
+ *    16 B/pt with no LDS transpose and no cross-lane shuffles. */
 
 #define DEC_TILE 8
 
-__global__ void __launch_bounds__(BLOCK_THREADS)
+__global__ void __launch_bounds__(BLOCK_THREADS, 4) /* cap at 128 VGPRs:
+    the kernel sits ~1 VGPR over the 4-waves/SIMD granule; the hint trades it */
 k_decode_batch(const uint8_t* __restrict__ blobs,
                const uint64_t* __restrict__ offsets,
                const uint32_t* __restrict__ lens,
@@ -679,26 +758,19 @@ k_decode_batch(const uint8_t* __restrict__ blobs,
                uint32_t stride) {
     const uint32_t lane = threadIdx.x & (WAVE - 1);
     const uint32_t wave = __builtin_amdgcn_readfirstlane(threadIdx.x >> 6);
-    const uint32_t s_base = blockIdx.x * BLOCK_THREADS + wave * WAVE;
-    const uint32_t slot = s_base + lane;
+    const uint32_t slot = blockIdx.x * BLOCK_THREADS + wave * WAVE + lane;
     /* optional scheduling permutation (e.g. length-sorted): waves then get
      * 64 similar-cost streams, removing intra-wave and intra-CU skew.
      * Purely a schedule: outputs still land in series order. */
     const uint32_t series = (perm && slot < nseries) ? (uint32_t)perm[slot] : slot;
 
-    /* unpadded tiles with an XOR column swizzle (col ^ (row & 7)): bank
-     * conflict-free for both per-lane access and the coalesced tile pass,
-     * and the exact 32 KB/block admits 5 blocks/CU (5 waves/SIMD). */
-    __shared__ int64_t ts_tile_all[WAVES_PER_BLOCK][WAVE][DEC_TILE];
-    __shared__ double val_tile_all[WAVES_PER_BLOCK][WAVE][DEC_TILE];
-    int64_t (*ts_tile)[DEC_TILE] = ts_tile_all[wave];
-    double (*val_tile)[DEC_TILE] = val_tile_all[wave];
-
+    __shared__ uint64_t ring_all[WAVES_PER_BLOCK][WAVE][IN_STRIDE];
+    uint64_t* ring = ring_all[wave][lane];
 
     const bool in_range = slot < nseries;
     Decoder d;
-    if (in_range)
-        d.init(blobs, offsets[series], lens[series], int_optimized != 0, default_unit);
+    d.init(blobs, in_range ? offsets[series] : 0, in_range ? lens[series] : 0,
+           int_optimized != 0, default_unit, ring);
 
     bool running = in_range;
     int err = 0;
@@ -706,37 +778,14 @@ k_decode_batch(const uint8_t* __restrict__ blobs,
     uint32_t k = 0;
 
     const bool discard = out_ts == nullptr; /* parse-only diagnostic */
-    auto flush = [&](uint32_t base_pt) {
-        if (discard) return;
-        __builtin_amdgcn_wave_barrier();
-        const uint32_t p = lane & (DEC_TILE - 1);
-        const uint32_t r0 = lane / DEC_TILE;
-        const uint32_t pt = base_pt + p;
-        if (__all(cnt >= base_pt + DEC_TILE)) {
-            /* all 64 rows full: unconditional stores, no per-row counts */
-            for (uint32_t j = 0; j < DEC_TILE; j++) {
-                uint32_t r = r0 + j * (WAVE / DEC_TILE);
-                uint32_t pc = p ^ (r & 7);
-                uint64_t row = (uint64_t)__shfl((int)series, (int)r);
-                out_ts[row * stride + pt] = ts_tile[r][pc];
-                out_vals[row * stride + pt] = val_tile[r][pc];
-            }
-        } else {
-            for (uint32_t j = 0; j < DEC_TILE; j++) {
-                uint32_t r = r0 + j * (WAVE / DEC_TILE);
-                uint32_t pc = p ^ (r & 7);
-                uint32_t c = (uint32_t)__shfl((int)cnt, (int)r);
-                uint64_t row = (uint64_t)__shfl((int)series, (int)r);
-                if (pt < c) {
-                    out_ts[row * stride + pt] = ts_tile[r][pc];
-                    out_vals[row * stride + pt] = val_tile[r][pc];
-                }
-            }
-        }
-        __builtin_amdgcn_wave_barrier();
-    };
+    const bool paired = (stride & 1) == 0;  /* 16B-store eligible rows */
 
-    while (__any(running)) {
+    /* manually unrolled tile (the optimizer refuses to unroll across the
+     * parser's control flow, and a non-unrolled tq[j] would spill the
+     * accumulators to scratch) */
+    int64_t t0, t1, t2, t3, t4, t5, t6, t7;
+    double v0, v1, v2, v3, v4, v5, v6, v7;
+    auto step = [&](int64_t& tj, double& vj) {
         if (running) {
             int64_t t;
             double v;
@@ -748,18 +797,58 @@ k_decode_batch(const uint8_t* __restrict__ blobs,
                 err = M3GPU_SERIES_CAPACITY;
                 running = false;
             } else {
-                if (!discard) {
-                    uint32_t col = (k & (DEC_TILE - 1)) ^ (lane & 7);
-                    ts_tile[lane][col] = t;
-                    val_tile[lane][col] = v;
-                }
+                tj = t;
+                vj = v;
                 cnt++;
             }
         }
-        k++;
-        if ((k & (DEC_TILE - 1)) == 0) flush(k - DEC_TILE);
+    };
+    auto pair_t = [](int64_t x, int64_t y) {
+        longlong2 p2;
+        p2.x = x;
+        p2.y = y;
+        return p2;
+    };
+    auto pair_v = [](double x, double y) {
+        double2 p2;
+        p2.x = x;
+        p2.y = y;
+        return p2;
+    };
+
+    while (__any(running)) {
+        d.r.refill(running);
+        step(t0, v0); step(t1, v1); step(t2, v2); step(t3, v3);
+        step(t4, v4); step(t5, v5); step(t6, v6); step(t7, v7);
+        if (!discard) {
+            const uint32_t produced = (cnt > k) ? (cnt - k) : 0;
+            if (produced) {
+                int64_t* tp = out_ts + (uint64_t)series * stride + k;
+                double* vp = out_vals + (uint64_t)series * stride + k;
+                if (produced == DEC_TILE && paired) {
+                    *(longlong2*)(tp + 0) = pair_t(t0, t1);
+                    *(longlong2*)(tp + 2) = pair_t(t2, t3);
+                    *(longlong2*)(tp + 4) = pair_t(t4, t5);
+                    *(longlong2*)(tp + 6) = pair_t(t6, t7);
+                    *(double2*)(vp + 0) = pair_v(v0, v1);
+                    *(double2*)(vp + 2) = pair_v(v2, v3);
+                    *(double2*)(vp + 4) = pair_v(v4, v5);
+                    *(double2*)(vp + 6) = pair_v(v6, v7);
+                } else {
+                    /* ragged tail: per-point stores */
+                    tp[0] = t0; vp[0] = v0;
+                    if (produced > 1) { tp[1] = t1; vp[1] = v1; }
+                    if (produced > 2) { tp[2] = t2; vp[2] = v2; }
+                    if (produced > 3) { tp[3] = t3; vp[3] = v3; }
+                    if (produced > 4) { tp[4] = t4; vp[4] = v4; }
+                    if (produced > 5) { tp[5] = t5; vp[5] = v5; }
+                    if (produced > 6) { tp[6] = t6; vp[6] = v6; }
+                    if (produced > 7) { tp[7] = t7; vp[7] = v7; }
+                }
+            }
+        }
+        k += DEC_TILE;
     }
-    if (k & (DEC_TILE - 1)) flush(k & ~(uint32_t)(DEC_TILE - 1));
 
     if (in_range) {
         out_counts[series] = cnt;
