@@ -146,7 +146,7 @@ def main():
         # wave's 64 streams are HBM neighbors; scheduling perm becomes
         # identity and is dropped
         new_lens = d_lens[d_perm.long()]
-        aligned = ((new_lens.to(torch.int64) + 15) // 16) * 16
+        aligned = ((new_lens.to(torch.int64) + 63) // 64) * 64
         # n+1 offsets: the end sentinel is needed by the oracle-side
         # cpu_baseline slice (the kernels only use offsets[series])
         new_off = torch.zeros(aligned.numel() + 1, dtype=torch.int64,

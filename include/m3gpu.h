@@ -24,7 +24,8 @@
  *    i's stream. The kernels REQUIRE offsets[i] to be 8-byte aligned and
  *    every stream zero-padded to an 8-byte boundary (refills are aligned
  *    u64 loads; the zero pad reproduces reader64's zero-filled tail word).
- *    pack_streams/compact emit 16-byte alignment, which satisfies this.
+ *    pack_streams and the fileset/regather packers emit 64-byte
+ *    alignment (one HBM line per stream chunk), which satisfies this.
  *    lens[] are true stream lengths.
  *  - Time units use the reference's xtime.Unit byte values
  *    (src/x/time/unit.go:30-42): 1=s, 2=ms, 3=us, 4=ns.

@@ -5,8 +5,9 @@
  * digest/{digest.go,buffer.go,writer.go}, fs.go:27-51 naming,
  * files.go:1729-1745 path construction) for bulk ingestion on the MI355X
  * host: open a volume, validate every digest, parse the info and index
- * files, and repack the data blocks into the 16-byte-aligned batch layout
- * the decode kernels consume (include/m3gpu.h).
+ * files, and repack the data blocks into the 64-byte-aligned batch layout
+ * the decode kernels consume (include/m3gpu.h; 8B required, 64B = one
+ * HBM line per stream chunk for the decode ring's aligned refills).
  *
  * Scope notes:
  *  - flush-type volumes with both current (fileset-<t>-<v>-<suffix>.db) and
@@ -521,7 +522,7 @@ int64_t m3gpu_fileset_packed_size(int handle) {
     FsVolume* v = fs_get(handle);
     if (!v) return M3GPU_FS_ERR_BADHANDLE;
     int64_t total = 0;
-    for (const FsEntry& e : v->entries) total += (e.size + 15) & ~15ll;
+    for (const FsEntry& e : v->entries) total += (e.size + 63) & ~63ll;
     return total;
 }
 
@@ -535,7 +536,7 @@ int m3gpu_fileset_pack(int handle, uint8_t* blob, uint64_t blob_cap,
     uint64_t off = 0;
     for (size_t i = 0; i < v->entries.size(); i++) {
         const FsEntry& e = v->entries[i];
-        uint64_t aligned = ((uint64_t)e.size + 15) & ~15ull;
+        uint64_t aligned = ((uint64_t)e.size + 63) & ~63ull;
         if (off + aligned > blob_cap) return M3GPU_FS_ERR_CAPACITY;
         memcpy(blob + off, v->data.data() + e.offset, (size_t)e.size);
         if (aligned > (uint64_t)e.size)

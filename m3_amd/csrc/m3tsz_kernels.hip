@@ -92,13 +92,14 @@ __device__ __forceinline__ double go_modf(double v, double* ip) {
 /* istream.go:73-115 over reader64.go:40-80, with the m3gpu.h layout
  * contract: stream starts 8B-aligned, buffer zero-padded to 8B. */
 
-/* Per-lane LDS input ring geometry (decode kernel): 16 buffered words per
- * lane, rows padded to 17 so that both the per-lane ds_read_b64 pulls and
- * the refill ds_write_b64 bursts are bank-conflict-free (b64 reads bank on
- * (a/4)%64 in 2x32 lane groups, writes on (a/4)%32 in 4x16 groups; the
- * 17-word row stride makes lane bases land on distinct bank pairs). */
-#define IN_WORDS 16
-#define IN_STRIDE 17
+/* Per-lane LDS input ring geometry (decode kernel): 8 buffered words per
+ * lane (one 64B chunk = one HBM line), rows padded to 9 so that both the
+ * per-lane ds_read_b64 pulls and the refill ds_write_b64 bursts are
+ * bank-conflict-free (b64 reads bank on (a/4)%64 in 2x32 lane groups,
+ * writes on (a/4)%32 in 4x16 groups; the 9-word row stride makes lane
+ * bases land on distinct bank pairs). */
+#define IN_WORDS 8
+#define IN_STRIDE 9
 
 struct BitReader {
     /* 128-bit register lookahead (`a` = next bits, `b` = following word,
@@ -122,19 +123,23 @@ struct BitReader {
      *    prefetch (`pfw`) beyond the window, i.e. 3 words of lookahead,
      *    keeps the refill load issued ~128 bits before first use. */
     const uint64_t* words;
-    uint64_t a, b;
+    uint64_t a, b, c;   /* c: ring-word prefetch — the ds_read latency sits
+                         * between one crossing and the NEXT, not in the
+                         * peek chain */
     uint32_t p;
     int64_t bits_left;  /* un-consumed stream bits (from the true byte len) */
     uint32_t wnext;     /* next word index to pull into the window */
     uint32_t wtotal;    /* ceil(len/8) */
     uint64_t* lds;      /* this lane's ring base, or nullptr */
     uint32_t rfill;     /* words fetched into the ring so far */
-    uint64_t pfw;       /* direct mode: 1-deep prefetched word */
+    uint64_t pend[IN_WORDS]; /* deferred refill: loads issued at tile start */
+    uint32_t pend_n;
+    uint64_t pfw;       /* direct mode: extra prefetched word */
 
     __device__ __forceinline__ uint64_t next_word() {
         uint64_t w;
         if (lds) {
-            /* ring hit is the steady state (refill() runs per tile); a
+            /* ring hit is the steady state (refill runs per tile); a
              * mid-tile underrun (adversarially dense streams consuming
              * >16 words in 8 points) falls back to a direct load —
              * correct, just slower */
@@ -149,34 +154,38 @@ struct BitReader {
         return w;
     }
 
-    /* Top the ring up to IN_WORDS words (whole-word granularity; streams
-     * are zero-padded to 8B, §m3gpu.h layout). Call once per tile with all
-     * lanes converged; `active=false` lanes stop fetching. Two batches of
-     * up to 8 loads each: loads issue back-to-back into registers, then
-     * write to LDS — one memory round-trip per tile, not per word. */
-    __device__ __forceinline__ void refill(bool active) {
+    /* Deferred per-tile ring top-up, CHUNK-granular: one refill = the next
+     * whole 64B line of this lane's stream (8 consecutive u64 loads issued
+     * back-to-back — the first fetches the line, the rest hit L1 within
+     * the burst; with 64B-aligned stream packing every line is fetched
+     * exactly once, no inter-tile L1 retention needed). refill_issue()
+     * starts the loads at TILE START only when the whole ring is free;
+     * refill_commit() writes them to the ring at TILE END, after the
+     * tile's output stores have issued — gfx9 vmcnt retires in issue
+     * order, so the commit's counted wait isolates these loads and never
+     * drains the in-flight output stores. The loads' latency hides under
+     * the whole tile's parsing. */
+    __device__ __forceinline__ void refill_issue(bool active) {
         if (!lds) return;
+        if (rfill < wnext) rfill = wnext; /* resync: underrun consumed
+                                           * [rfill, wnext) directly */
         uint32_t used = rfill - wnext;
-        uint32_t need = active ? (IN_WORDS - used) : 0;
-        uint32_t avail = wtotal - rfill;
-        if (need > avail) need = avail;
-        uint64_t v[8];
+        pend_n = 0;
+        if (active && used == 0 && rfill < wtotal) {
+            uint32_t n = wtotal - rfill;
+            pend_n = n > IN_WORDS ? IN_WORDS : n;
 #pragma unroll
-        for (uint32_t i = 0; i < 8; i++)
-            if (i < need) v[i] = __builtin_bswap64(words[rfill + i]);
-#pragma unroll
-        for (uint32_t i = 0; i < 8; i++)
-            if (i < need) lds[(rfill + i) & (IN_WORDS - 1)] = v[i];
-        if (__any(need > 8)) {
-            uint64_t v2[8];
-#pragma unroll
-            for (uint32_t i = 0; i < 8; i++)
-                if (i + 8 < need) v2[i] = __builtin_bswap64(words[rfill + i + 8]);
-#pragma unroll
-            for (uint32_t i = 0; i < 8; i++)
-                if (i + 8 < need) lds[(rfill + i + 8) & (IN_WORDS - 1)] = v2[i];
+            for (uint32_t i = 0; i < IN_WORDS; i++)
+                if (i < pend_n) pend[i] = __builtin_bswap64(words[rfill + i]);
         }
-        rfill += need;
+    }
+    __device__ __forceinline__ void refill_commit() {
+        if (!lds) return;
+#pragma unroll
+        for (uint32_t i = 0; i < IN_WORDS; i++)
+            if (i < pend_n) lds[(rfill + i) & (IN_WORDS - 1)] = pend[i];
+        rfill += pend_n;
+        pend_n = 0;
     }
 
     __device__ __forceinline__ void init(const uint8_t* base, uint64_t off,
@@ -187,10 +196,16 @@ struct BitReader {
         lds = ring;
         wnext = 0;
         rfill = 0;
-        if (ring) refill(true);
-        else pfw = wtotal ? __builtin_bswap64(words[0]) : 0;
+        pend_n = 0;
+        if (ring) {
+            refill_issue(true);
+            refill_commit();
+        } else {
+            pfw = wtotal ? __builtin_bswap64(words[0]) : 0;
+        }
         a = next_word();
         b = next_word();
+        c = next_word();
         p = 0;
     }
 
@@ -204,7 +219,8 @@ struct BitReader {
         p += n;
         if (p >= 64) {
             a = b;
-            b = next_word();
+            b = c;
+            c = next_word();
             p -= 64;
         }
     }
@@ -679,7 +695,11 @@ struct Decoder {
             prev_time_delta += dod;
             prev_time += prev_time_delta;
             *t = prev_time;
-            *v = (mult == 0) ? int_val : int_val / exp10_table(mult);
+            /* the f64 divide is ~25 VALU ops: branch it so mult==0 waves
+             * (the common case) never pay for a select-discarded divide */
+            double outv = int_val;
+            if (mult != 0) outv = int_val / exp10_table(mult);
+            *v = outv;
             return 1;
         }
         /* float value (pure float stream, or int-optimized float mode) */
@@ -716,7 +736,8 @@ struct Decoder {
         if (err) return -err;
         *t = prev_time;
         if (!int_optimized || is_float) *v = bits2f(prev_float_bits);
-        else *v = (mult == 0) ? int_val : int_val / exp10_table(mult);
+        else if (mult != 0) *v = int_val / exp10_table(mult);
+        else *v = int_val;
         return 1;
     }
     __device__ __forceinline__ double exp10_table(uint8_t m) {
@@ -746,8 +767,9 @@ struct Decoder {
 
 #define DEC_TILE 8
 
-__global__ void __launch_bounds__(BLOCK_THREADS, 4) /* cap at 128 VGPRs:
-    the kernel sits ~1 VGPR over the 4-waves/SIMD granule; the hint trades it */
+__global__ void __launch_bounds__(BLOCK_THREADS, 3) /* 3 waves/SIMD: jointly
+    pinned by the 50.4 KB/block LDS (ring + tiles -> 3 blocks/CU) and the
+    136-168 VGPR allocation granule */
 k_decode_batch(const uint8_t* __restrict__ blobs,
                const uint64_t* __restrict__ offsets,
                const uint32_t* __restrict__ lens,
@@ -764,8 +786,16 @@ k_decode_batch(const uint8_t* __restrict__ blobs,
      * Purely a schedule: outputs still land in series order. */
     const uint32_t series = (perm && slot < nseries) ? (uint32_t)perm[slot] : slot;
 
+    /* input ring + output tiles: 4x(4.5 + 8) KB/block -> 3 blocks/CU.
+     * Output tiles use the XOR column swizzle (col ^ (lane & 7)): bank
+     * conflict-free for both the per-lane staging write and the
+     * transposed flush read. */
     __shared__ uint64_t ring_all[WAVES_PER_BLOCK][WAVE][IN_STRIDE];
+    __shared__ int64_t ts_tile_all[WAVES_PER_BLOCK][WAVE][DEC_TILE];
+    __shared__ double val_tile_all[WAVES_PER_BLOCK][WAVE][DEC_TILE];
     uint64_t* ring = ring_all[wave][lane];
+    int64_t (*ts_tile)[DEC_TILE] = ts_tile_all[wave];
+    double (*val_tile)[DEC_TILE] = val_tile_all[wave];
 
     const bool in_range = slot < nseries;
     Decoder d;
@@ -778,14 +808,41 @@ k_decode_batch(const uint8_t* __restrict__ blobs,
     uint32_t k = 0;
 
     const bool discard = out_ts == nullptr; /* parse-only diagnostic */
-    const bool paired = (stride & 1) == 0;  /* 16B-store eligible rows */
 
-    /* manually unrolled tile (the optimizer refuses to unroll across the
-     * parser's control flow, and a non-unrolled tq[j] would spill the
-     * accumulators to scratch) */
-    int64_t t0, t1, t2, t3, t4, t5, t6, t7;
-    double v0, v1, v2, v3, v4, v5, v6, v7;
-    auto step = [&](int64_t& tj, double& vj) {
+    /* transposed flush: lane l stores row (l>>3)+8j, point (l&7) — 8
+     * consecutive 8B addresses per row = full 64B-line utilization for
+     * ts[] and val[] (WRITE_SIZE stays at the algorithmic 16 B/pt) */
+    auto flush = [&](uint32_t base_pt) {
+        if (discard) return;
+        __builtin_amdgcn_wave_barrier();
+        const uint32_t p = lane & (DEC_TILE - 1);
+        const uint32_t r0 = lane / DEC_TILE;
+        const uint32_t pt = base_pt + p;
+        if (__all(cnt >= base_pt + DEC_TILE)) {
+            /* all 64 rows full: unconditional stores, no per-row counts */
+            for (uint32_t j = 0; j < DEC_TILE; j++) {
+                uint32_t rr = r0 + j * (WAVE / DEC_TILE);
+                uint32_t pc = p ^ (rr & 7);
+                uint64_t row = (uint64_t)__shfl((int)series, (int)rr);
+                out_ts[row * stride + pt] = ts_tile[rr][pc];
+                out_vals[row * stride + pt] = val_tile[rr][pc];
+            }
+        } else {
+            for (uint32_t j = 0; j < DEC_TILE; j++) {
+                uint32_t rr = r0 + j * (WAVE / DEC_TILE);
+                uint32_t pc = p ^ (rr & 7);
+                uint32_t cc = (uint32_t)__shfl((int)cnt, (int)rr);
+                uint64_t row = (uint64_t)__shfl((int)series, (int)rr);
+                if (pt < cc) {
+                    out_ts[row * stride + pt] = ts_tile[rr][pc];
+                    out_vals[row * stride + pt] = val_tile[rr][pc];
+                }
+            }
+        }
+        __builtin_amdgcn_wave_barrier();
+    };
+
+    auto step = [&](uint32_t j) {
         if (running) {
             int64_t t;
             double v;
@@ -797,56 +854,22 @@ k_decode_batch(const uint8_t* __restrict__ blobs,
                 err = M3GPU_SERIES_CAPACITY;
                 running = false;
             } else {
-                tj = t;
-                vj = v;
+                if (!discard) {
+                    uint32_t col = j ^ (lane & 7);
+                    ts_tile[lane][col] = t;
+                    val_tile[lane][col] = v;
+                }
                 cnt++;
             }
         }
     };
-    auto pair_t = [](int64_t x, int64_t y) {
-        longlong2 p2;
-        p2.x = x;
-        p2.y = y;
-        return p2;
-    };
-    auto pair_v = [](double x, double y) {
-        double2 p2;
-        p2.x = x;
-        p2.y = y;
-        return p2;
-    };
 
     while (__any(running)) {
-        d.r.refill(running);
-        step(t0, v0); step(t1, v1); step(t2, v2); step(t3, v3);
-        step(t4, v4); step(t5, v5); step(t6, v6); step(t7, v7);
-        if (!discard) {
-            const uint32_t produced = (cnt > k) ? (cnt - k) : 0;
-            if (produced) {
-                int64_t* tp = out_ts + (uint64_t)series * stride + k;
-                double* vp = out_vals + (uint64_t)series * stride + k;
-                if (produced == DEC_TILE && paired) {
-                    *(longlong2*)(tp + 0) = pair_t(t0, t1);
-                    *(longlong2*)(tp + 2) = pair_t(t2, t3);
-                    *(longlong2*)(tp + 4) = pair_t(t4, t5);
-                    *(longlong2*)(tp + 6) = pair_t(t6, t7);
-                    *(double2*)(vp + 0) = pair_v(v0, v1);
-                    *(double2*)(vp + 2) = pair_v(v2, v3);
-                    *(double2*)(vp + 4) = pair_v(v4, v5);
-                    *(double2*)(vp + 6) = pair_v(v6, v7);
-                } else {
-                    /* ragged tail: per-point stores */
-                    tp[0] = t0; vp[0] = v0;
-                    if (produced > 1) { tp[1] = t1; vp[1] = v1; }
-                    if (produced > 2) { tp[2] = t2; vp[2] = v2; }
-                    if (produced > 3) { tp[3] = t3; vp[3] = v3; }
-                    if (produced > 4) { tp[4] = t4; vp[4] = v4; }
-                    if (produced > 5) { tp[5] = t5; vp[5] = v5; }
-                    if (produced > 6) { tp[6] = t6; vp[6] = v6; }
-                    if (produced > 7) { tp[7] = t7; vp[7] = v7; }
-                }
-            }
-        }
+        d.r.refill_issue(running);
+        step(0); step(1); step(2); step(3);
+        step(4); step(5); step(6); step(7);
+        flush(k);
+        d.r.refill_commit();
         k += DEC_TILE;
     }
 

@@ -106,10 +106,12 @@ def _raise_series_errors(errs, what):
 
 def pack_streams(streams):
     """Pack a list of encoded streams into (blob, offsets, lens) with the
-    16-byte-aligned zero-padded layout the C ABI requires."""
+    64-byte-aligned zero-padded layout (one stream chunk = one HBM line,
+    so the decode ring's chunk refills are line-aligned; the C ABI itself
+    requires only 8B)."""
     n = len(streams)
     lens = np.fromiter((len(s) for s in streams), dtype=np.uint32, count=n)
-    padded = (lens.astype(np.uint64) + 15) & ~np.uint64(15)
+    padded = (lens.astype(np.uint64) + 63) & ~np.uint64(63)
     offsets = np.zeros(n + 1, dtype=np.uint64)
     np.cumsum(padded, out=offsets[1:])
     blob = np.zeros(int(offsets[-1]), dtype=np.uint8)

@@ -86,7 +86,7 @@ def encode_on_device(torch, nseries, npts, chunk=65536, device="cuda:0",
             raise engine.M3GpuError("encode errors while building workload")
         lens = d_lens_t[:n].cpu().numpy().astype(np.uint32)
         lens_all[s0:s0 + n] = lens
-        padded = (lens.astype(np.int64) + 15) & ~15
+        padded = (lens.astype(np.int64) + 63) & ~63
         offs = np.zeros(n + 1, dtype=np.int64)
         np.cumsum(padded, out=offs[1:])
         d_blob_chunk = torch.zeros(int(offs[-1]), dtype=torch.uint8, device=device)

@@ -59,7 +59,7 @@ def test_fileset_roundtrip(tmp_path):
         for sid, blob, tags in series:
             assert by_id[sid][4] == (tags or b"")
         blob_arr, offsets, lens = v.pack()
-        assert np.all(offsets % 16 == 0)
+        assert np.all(offsets % 64 == 0)
         for i, (sid, size, off, ck, tags) in enumerate(ents):
             got = bytes(blob_arr[int(offsets[i]):int(offsets[i]) + int(lens[i])])
             assert got == raw[sid][2], sid
