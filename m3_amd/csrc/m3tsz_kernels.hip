@@ -98,8 +98,12 @@ __device__ __forceinline__ double go_modf(double v, double* ip) {
  * bank-conflict-free (b64 reads bank on (a/4)%64 in 2x32 lane groups,
  * writes on (a/4)%32 in 4x16 groups; the 9-word row stride makes lane
  * bases land on distinct bank pairs). */
-#define IN_WORDS 8
-#define IN_STRIDE 9
+#ifndef IN_WORDS
+#define IN_WORDS 4
+#endif
+#ifndef IN_STRIDE
+#define IN_STRIDE 4
+#endif
 
 struct BitReader {
     /* 128-bit register lookahead (`a` = next bits, `b` = following word,
@@ -662,35 +666,79 @@ struct Decoder {
         const uint64_t w = r.peek64();
         /* ---- timestamp: '0' => dod 0; '1'^L 0 + {7,9,12}b => bucket ---- */
         const uint64_t b0 = w >> 63;
-        const uint32_t L = (uint32_t)__builtin_clzll(~w | 1); /* leading 1s */
-        if (L >= 4 || (w >> 55) == MARKER_OPCODE)
-            return -1000; /* default bucket / EOS|annotation|timeunit marker */
-        const uint32_t vb = (0xC970u >> (L * 4)) & 0xFu; /* {0,7,9,12} */
-        const uint32_t tsbits = b0 ? (L + 1 + vb) : 1;
-        const uint32_t vbs = b0 ? vb : 7; /* shift-safe stand-in when dod==0 */
-        int64_t dod = sign_extend((w << (L + 1)) >> (64 - vbs), vbs) * unit_ns;
-        dod = b0 ? dod : 0; /* unit_ns==0 swallows invalid units (:271-274) */
+        uint32_t tsbits = 1;
+        int64_t dod = 0;
+        if (b0) { /* branched: regular-cadence waves (dod==0 = a single
+                   * 0 bit, the dominant shape) skip the bucket block */
+            const uint32_t L = (uint32_t)__builtin_clzll(~w); /* leading 1s */
+            if (L >= 4 || (w >> 55) == MARKER_OPCODE)
+                return -1000; /* default bucket / EOS|annot|timeunit marker */
+            const uint32_t vb = (0xC970u >> (L * 4)) & 0xFu; /* {7,9,12} */
+            tsbits = L + 1 + vb;
+            /* unit_ns==0 swallows invalid units (:271-274) */
+            dod = sign_extend((w << (L + 1)) >> (64 - vb), vb) * unit_ns;
+        }
         const uint64_t w2 = w << tsbits; /* tsbits <= 16 here */
 
         if (int_optimized && !is_float) {
-            /* int mode: '1' sign+sig diff | '01' repeat | '00...' stepwise */
+            /* int mode: '1' sign+sig diff | '01' repeat | '000' sig/mult
+             * update + diff (encoder.go:200-250) | '001' float switch ->
+             * stepwise. Covering the update opcode here matters: adaptive
+             * sig tracking (int_sig_bits_tracker.go) re-widths every few
+             * points on random-walk data, and each stepwise escape drags
+             * the whole wave through the big slow path. */
             const uint64_t v0 = w2 >> 63;
-            if (!v0 && ((w2 >> 62) & 1) == 0)
-                return -1000; /* mode/sig/mult update */
-            if (v0 && (sig > 45u || tsbits + 2u + sig > 64u))
-                return -1000; /* wide sig: field may exceed the peek */
-            const uint32_t nb = v0 ? (2u + sig) : 2u;
-            /* shifts masked &63: on repeat lanes sig may be 64 (slow-width
-             * state) and the field value is select-discarded, but the
-             * shift amounts must stay defined */
-            const uint32_t sg = sig & 63u;
-            const uint64_t field = (w2 << 1) >> ((63u - sg) & 63u);
-            const uint64_t sbit = (field >> sg) & 1;
-            const uint64_t mag = field ^ (sbit << sg);
-            const double add = (sbit ? 1.0 : -1.0) * (double)mag;
-            /* repeat must leave int_val EXACTLY unchanged (adding 0.0 would
-             * flip -0.0; DESIGN.md §6 repeat quirk) — select, don't add */
-            if (v0) int_val += add;
+            const uint64_t v1 = (w2 >> 62) & 1;
+            uint32_t nb;
+            if (v0) { /* opcodeNoUpdate: sign + sig diff */
+                if (sig > 45u || tsbits + 2u + sig > 64u)
+                    return -1000; /* wide sig: field may exceed the peek */
+                nb = 2u + sig;
+                const uint64_t field = (w2 << 1) >> (63u - sig);
+                const uint64_t sbit = (field >> sig) & 1;
+                const uint64_t mag = field ^ (sbit << sig);
+                int_val += (sbit ? 1.0 : -1.0) * (double)mag;
+            } else if (v1) {
+                /* repeat: int_val EXACTLY unchanged (adding 0.0 would flip
+                 * -0.0; DESIGN.md §6 repeat quirk) */
+                nb = 2;
+            } else if ((w2 >> 61) & 1) {
+                return -1000; /* '001' -> float-mode switch: stepwise */
+            } else {
+                /* '000' + readIntSigMult + readIntValDiff
+                 * (iterator.go:178-219) from the same peek */
+                const uint64_t u = w2 << 3;
+                uint8_t nsig = sig;
+                uint32_t off;
+                if (u >> 63) { /* opcodeUpdateSig */
+                    const uint64_t z = (u >> 62) & 1;
+                    nsig = z ? (uint8_t)(((u >> 56) & 0x3f) + 1) : 0;
+                    off = z ? 8u : 2u;
+                } else {
+                    off = 1u;
+                }
+                const uint64_t m = u << off;
+                uint8_t nmult = mult;
+                uint32_t moff;
+                if (m >> 63) { /* opcodeUpdateMult */
+                    nmult = (uint8_t)((m >> 60) & 0x7);
+                    moff = 4u;
+                    if (nmult > MAX_MULT)
+                        return -1000; /* stepwise raises invalid_mult */
+                } else {
+                    moff = 1u;
+                }
+                if (nsig > 45u || tsbits + 3u + off + moff + 1u + nsig > 64u)
+                    return -1000;
+                const uint64_t f = m << moff;
+                const uint64_t sbit = f >> 63;
+                const uint64_t mag =
+                    nsig ? ((f << 1) >> ((64u - nsig) & 63u)) : 0;
+                int_val += (sbit ? 1.0 : -1.0) * (double)mag;
+                sig = nsig;
+                mult = nmult;
+                nb = 3u + off + moff + 1u + nsig;
+            }
             r.consume(tsbits + nb);
             prev_time_delta += dod;
             prev_time += prev_time_delta;
@@ -786,16 +834,13 @@ k_decode_batch(const uint8_t* __restrict__ blobs,
      * Purely a schedule: outputs still land in series order. */
     const uint32_t series = (perm && slot < nseries) ? (uint32_t)perm[slot] : slot;
 
-    /* input ring + output tiles: 4x(4.5 + 8) KB/block -> 3 blocks/CU.
-     * Output tiles use the XOR column swizzle (col ^ (lane & 7)): bank
-     * conflict-free for both the per-lane staging write and the
-     * transposed flush read. */
+    /* input ring + output tile. The tile is AoS — (ts, val) PAIRS, one
+     * ds_write_b128 per point instead of two b64s — with a pair-granular
+     * XOR column swizzle (col ^ (lane & 7)). */
     __shared__ uint64_t ring_all[WAVES_PER_BLOCK][WAVE][IN_STRIDE];
-    __shared__ int64_t ts_tile_all[WAVES_PER_BLOCK][WAVE][DEC_TILE];
-    __shared__ double val_tile_all[WAVES_PER_BLOCK][WAVE][DEC_TILE];
+    __shared__ longlong2 tile_all[WAVES_PER_BLOCK][WAVE][DEC_TILE];
     uint64_t* ring = ring_all[wave][lane];
-    int64_t (*ts_tile)[DEC_TILE] = ts_tile_all[wave];
-    double (*val_tile)[DEC_TILE] = val_tile_all[wave];
+    longlong2 (*tile)[DEC_TILE] = tile_all[wave];
 
     const bool in_range = slot < nseries;
     Decoder d;
@@ -809,9 +854,10 @@ k_decode_batch(const uint8_t* __restrict__ blobs,
 
     const bool discard = out_ts == nullptr; /* parse-only diagnostic */
 
-    /* transposed flush: lane l stores row (l>>3)+8j, point (l&7) — 8
-     * consecutive 8B addresses per row = full 64B-line utilization for
-     * ts[] and val[] (WRITE_SIZE stays at the algorithmic 16 B/pt) */
+    /* transposed flush: lane l covers row (l>>3)+8j, point (l&7) — one
+     * ds_read_b128 pulls the (ts, val) pair, then two 8B stores whose
+     * addresses are consecutive across the 8 lanes of a row = full
+     * 64B-line utilization (WRITE_SIZE stays at the algorithmic 16 B/pt) */
     auto flush = [&](uint32_t base_pt) {
         if (discard) return;
         __builtin_amdgcn_wave_barrier();
@@ -824,8 +870,9 @@ k_decode_batch(const uint8_t* __restrict__ blobs,
                 uint32_t rr = r0 + j * (WAVE / DEC_TILE);
                 uint32_t pc = p ^ (rr & 7);
                 uint64_t row = (uint64_t)__shfl((int)series, (int)rr);
-                out_ts[row * stride + pt] = ts_tile[rr][pc];
-                out_vals[row * stride + pt] = val_tile[rr][pc];
+                longlong2 pr = tile[rr][pc];
+                out_ts[row * stride + pt] = pr.x;
+                out_vals[row * stride + pt] = __longlong_as_double(pr.y);
             }
         } else {
             for (uint32_t j = 0; j < DEC_TILE; j++) {
@@ -834,15 +881,18 @@ k_decode_batch(const uint8_t* __restrict__ blobs,
                 uint32_t cc = (uint32_t)__shfl((int)cnt, (int)rr);
                 uint64_t row = (uint64_t)__shfl((int)series, (int)rr);
                 if (pt < cc) {
-                    out_ts[row * stride + pt] = ts_tile[rr][pc];
-                    out_vals[row * stride + pt] = val_tile[rr][pc];
+                    longlong2 pr = tile[rr][pc];
+                    out_ts[row * stride + pt] = pr.x;
+                    out_vals[row * stride + pt] = __longlong_as_double(pr.y);
                 }
             }
         }
         __builtin_amdgcn_wave_barrier();
     };
 
-    auto step = [&](uint32_t j) {
+    /* capacity is tile-hoisted: the per-point check only runs on the rare
+     * tile that could cross `stride` */
+    auto step = [&](uint32_t j, bool check_cap) {
         if (running) {
             int64_t t;
             double v;
@@ -850,14 +900,16 @@ k_decode_batch(const uint8_t* __restrict__ blobs,
             if (rstat <= 0) {
                 err = -rstat;
                 running = false;
-            } else if (cnt >= stride) {
+            } else if (check_cap && cnt >= stride) {
                 err = M3GPU_SERIES_CAPACITY;
                 running = false;
             } else {
                 if (!discard) {
                     uint32_t col = j ^ (lane & 7);
-                    ts_tile[lane][col] = t;
-                    val_tile[lane][col] = v;
+                    longlong2 pr;
+                    pr.x = t;
+                    pr.y = __double_as_longlong(v);
+                    tile[lane][col] = pr;
                 }
                 cnt++;
             }
@@ -866,10 +918,19 @@ k_decode_batch(const uint8_t* __restrict__ blobs,
 
     while (__any(running)) {
         d.r.refill_issue(running);
-        step(0); step(1); step(2); step(3);
-        step(4); step(5); step(6); step(7);
-        flush(k);
+        /* NOT unrolled: the fully-inlined parser is ~10k instructions —
+         * eight copies would blow the instruction cache */
+        /* wave-uniform capacity flag: cnt <= k + j, so the per-point
+         * check matters only on a tile that could cross `stride` */
+        const bool check_cap = k + DEC_TILE > stride;
+#pragma unroll 1
+        for (uint32_t j = 0; j < DEC_TILE; j++) step(j, check_cap);
+        /* commit BEFORE the flush: the commit's counted vmcnt wait then
+         * covers only this tile's loads plus the PREVIOUS tile's stores
+         * (issued a whole tile ago, long retired) — never the 16 stores
+         * the flush is about to issue */
         d.r.refill_commit();
+        flush(k);
         k += DEC_TILE;
     }
 
