@@ -87,11 +87,18 @@ def main():
     world = int(os.environ.get("WORLD_SIZE", "1"))
     local_rank = int(os.environ.get("LOCAL_RANK", str(rank)))
     dist = None
+    ndev = max(torch.cuda.device_count(), 1)
+    backend = "nccl"
+    if world > ndev:
+        # oversubscription (more ranks than GPUs, e.g. a 2-rank rendezvous
+        # test on a 1-GPU box): RCCL refuses duplicate devices, so the
+        # barrier/MAX-reduce run over gloo; the compute still shares the GPU
+        backend = "gloo"
     if world > 1:
         import torch.distributed as dist_mod
         dist = dist_mod
-        dist.init_process_group("nccl")
-    device = f"cuda:{local_rank}"
+        dist.init_process_group(backend)
+    device = f"cuda:{local_rank % ndev}"
     torch.cuda.set_device(device)
 
     if args.nseries is None:
@@ -215,7 +222,8 @@ def main():
 
     elapsed = wall1 - wall0
     if dist:
-        t = torch.tensor([elapsed], device=device)
+        t = torch.tensor([elapsed],
+                         device=device if backend == "nccl" else "cpu")
         dist.all_reduce(t, op=dist.ReduceOp.MAX)
         elapsed = float(t.item())
 

@@ -1398,6 +1398,11 @@ struct RollupPlan {
      * (the calcQuantiles walk). Computed on the host; buckets beyond it
      * flag M3GPU_SERIES_BUCKET_OVERFLOW rather than approximate. */
     int32_t exact_cap;
+    /* Largest bucket size for which EVERY requested quantile resolves to
+     * the bucket MAX under the reference's walk (all-high quantile sets,
+     * e.g. p90+): the lane kernel then needs no value staging at all —
+     * running min/max suffice. 0 = extreme mode not applicable. */
+    int32_t extreme_cap;
     /* CKMS stream options (quantile/cm/options.go:30-32 defaults:
      * eps=1e-3, insertAndCompressEvery=1024); configurable via the
      * *_opts ABI entries. every is capped at 1024 by validation (the
@@ -1669,9 +1674,14 @@ k_compact(const uint8_t* __restrict__ src, uint32_t src_stride,
  * wave-per-series kernel (64-deep staging). */
 #define RQCAP_LANE 16
 
-template <bool WITH_Q, int METRIC>
-__global__ void __launch_bounds__(BLOCK_THREADS, 5) /* target 5 waves/SIMD:
-    the kernel sits a few VGPRs over the 96 granule; the hint trades them */
+#define RQ_NONE 0
+#define RQ_STAGED 1
+#define RQ_EXTREME 2
+
+template <int QMODE, int METRIC>
+__global__ void __launch_bounds__(BLOCK_THREADS, QMODE == RQ_STAGED ? 3 : 4)
+/* staged: 40KB LDS (qbuf + ring) pins 3 blocks/CU, so 3 waves/SIMD with no
+   spills; other modes target the 128-VGPR granule (4 waves/SIMD) */
 k_rollup_lane(const uint8_t* __restrict__ blobs,
               const uint64_t* __restrict__ offsets,
               const uint32_t* __restrict__ lens,
@@ -1684,14 +1694,18 @@ k_rollup_lane(const uint8_t* __restrict__ blobs,
     const uint32_t wave = __builtin_amdgcn_readfirstlane(threadIdx.x >> 6);
     const uint32_t series = blockIdx.x * BLOCK_THREADS + wave * WAVE + lane;
 
+    constexpr bool WITH_Q = QMODE == RQ_STAGED;
     __shared__ double qbuf_all[WITH_Q ? WAVES_PER_BLOCK : 1]
                               [WITH_Q ? WAVE : 1][WITH_Q ? RQCAP_LANE : 1];
     double* qrow = WITH_Q ? qbuf_all[wave][lane] : nullptr;
+    /* per-lane input ring, as in k_decode_batch */
+    __shared__ uint64_t rring_all[WAVES_PER_BLOCK][WAVE][IN_STRIDE];
+    uint64_t* rring = rring_all[wave][lane];
 
     const bool in_range = series < nseries;
     Decoder d;
-    if (in_range)
-        d.init(blobs, offsets[series], lens[series], int_optimized != 0, default_unit);
+    d.init(blobs, in_range ? offsets[series] : 0, in_range ? lens[series] : 0,
+           int_optimized != 0, default_unit, rring);
 
     double* out_row = out + (uint64_t)series * nbuckets * plan.naggs;
     int64_t* wts_row = out_window_ts ? out_window_ts + (uint64_t)series * nbuckets : nullptr;
@@ -1700,6 +1714,7 @@ k_rollup_lane(const uint8_t* __restrict__ blobs,
     bs.reset();
     int64_t base = 0;
     int64_t cur_bucket = -1;
+    int64_t cur_lo = 0, cur_hi = 0; /* current bucket ts bounds */
     uint32_t nq = 0;
     int err = 0;
     bool running = in_range;
@@ -1740,7 +1755,13 @@ k_rollup_lane(const uint8_t* __restrict__ blobs,
                 default: break;
                 }
             } else { /* timer.go:131-153; qrow holds the sorted bucket values */
-                if (WITH_Q && qi >= 0) {
+                if (QMODE == RQ_EXTREME && qi >= 0) {
+                    /* all-high quantile set: every requested quantile
+                     * resolves to the bucket max for n <= plan.extreme_cap
+                     * (host-verified against the walk AND quantilesFromBuf;
+                     * deeper buckets took the overflow path) */
+                    r = bs.count ? bs.fmax : 0.0;
+                } else if (WITH_Q && qi >= 0) {
                     if (nq == 0) r = 0.0; /* empty stream Quantile -> 0 */
                     else if (nq <= 3) { /* quantilesFromBuf :210-229 */
                         uint32_t idx = (uint32_t)(plan.qs[qi] * (double)nq);
@@ -1757,8 +1778,16 @@ k_rollup_lane(const uint8_t* __restrict__ blobs,
                     }
                 } else {
                     switch (t) {
-                    case M3GPU_AGG_MIN: r = (WITH_Q && nq) ? qrow[0] : 0.0; break;
-                    case M3GPU_AGG_MAX: r = (WITH_Q && nq) ? qrow[nq - 1] : 0.0; break;
+                    case M3GPU_AGG_MIN:
+                        r = (QMODE == RQ_EXTREME)
+                                ? (bs.count ? bs.fmin : 0.0)
+                                : ((WITH_Q && nq) ? qrow[0] : 0.0);
+                        break;
+                    case M3GPU_AGG_MAX:
+                        r = (QMODE == RQ_EXTREME)
+                                ? (bs.count ? bs.fmax : 0.0)
+                                : ((WITH_Q && nq) ? qrow[nq - 1] : 0.0);
+                        break;
                     case M3GPU_AGG_MEAN: r = bs.count ? bs.fsum / (double)bs.count : 0.0; break;
                     case M3GPU_AGG_COUNT: r = (double)bs.count; break;
                     case M3GPU_AGG_SUM: r = bs.fsum; break;
@@ -1784,6 +1813,9 @@ k_rollup_lane(const uint8_t* __restrict__ blobs,
     };
 
     while (__any(running)) {
+        d.r.refill_issue(running);
+#pragma unroll 1
+        for (uint32_t jt = 0; jt < DEC_TILE; jt++) {
         if (running) {
             int64_t t;
             double v;
@@ -1796,12 +1828,18 @@ k_rollup_lane(const uint8_t* __restrict__ blobs,
                 int64_t b = -1;
                 bool ok = true;
                 if (have) {
-                    if (cur_bucket < 0) base = (t / window_ns) * window_ns; /* Truncate */
-                    if (t < base) { err = M3GPU_SERIES_UNSORTED; running = false; ok = false; }
-                    else {
-                        b = (t - base) / window_ns;
-                        if (b >= (int64_t)nbuckets) { err = M3GPU_SERIES_CAPACITY; running = false; ok = false; }
-                        else if (b < cur_bucket) { err = M3GPU_SERIES_UNSORTED; running = false; ok = false; }
+                    if (cur_bucket >= 0 && t >= cur_lo && t < cur_hi) {
+                        /* same bucket: two compares instead of a 64-bit
+                         * division per point */
+                        b = cur_bucket;
+                    } else {
+                        if (cur_bucket < 0) base = (t / window_ns) * window_ns; /* Truncate */
+                        if (t < base) { err = M3GPU_SERIES_UNSORTED; running = false; ok = false; }
+                        else {
+                            b = (t - base) / window_ns;
+                            if (b >= (int64_t)nbuckets) { err = M3GPU_SERIES_CAPACITY; running = false; ok = false; }
+                            else if (b < cur_bucket) { err = M3GPU_SERIES_UNSORTED; running = false; ok = false; }
+                        }
                     }
                 }
                 if (ok && (!have || b != cur_bucket)) {
@@ -1817,6 +1855,8 @@ k_rollup_lane(const uint8_t* __restrict__ blobs,
                         bs.reset();
                         nq = 0;
                         cur_bucket = b;
+                        cur_lo = base + b * window_ns;
+                        cur_hi = cur_lo + window_ns;
                     }
                 }
                 if (ok) {
@@ -1837,30 +1877,69 @@ k_rollup_lane(const uint8_t* __restrict__ blobs,
                             bs.fsumsq += v * v;
                         }
                     } else { /* timer.go:56-75 */
-                        bs.count++;
-                        bs.fsum += v;
-                        bs.fsumsq += v * v;
-                        if (WITH_Q && plan.nq > 0) {
-                            if (nq >= RQCAP_LANE ||
-                                nq >= (uint32_t)plan.exact_cap) {
-                                err = M3GPU_SERIES_BUCKET_OVERFLOW;
-                                running = false;
-                            } else {
-                                /* insertion into this lane's sorted LDS row */
-                                uint32_t j = nq;
-                                while (j > 0 && qrow[j - 1] > v) { qrow[j] = qrow[j - 1]; j--; }
-                                qrow[j] = v;
-                                nq++;
+                        if (QMODE == RQ_EXTREME &&
+                            (isnan(v) ||
+                             bs.count >= (int64_t)plan.extreme_cap)) {
+                            /* NaN ordering and deep buckets keep the exact
+                             * staged semantics: retry on the wave tier */
+                            err = M3GPU_SERIES_BUCKET_OVERFLOW;
+                            running = false;
+                        } else {
+                            if (QMODE == RQ_EXTREME) {
+                                if (bs.count == 0) {
+                                    bs.fmin = v;
+                                    bs.fmax = v;
+                                } else {
+                                    if (v < bs.fmin) bs.fmin = v;
+                                    if (v > bs.fmax) bs.fmax = v;
+                                }
+                            }
+                            bs.count++;
+                            bs.fsum += v;
+                            bs.fsumsq += v * v;
+                            if (WITH_Q && plan.nq > 0) {
+                                if (nq >= RQCAP_LANE ||
+                                    nq >= (uint32_t)plan.exact_cap) {
+                                    err = M3GPU_SERIES_BUCKET_OVERFLOW;
+                                    running = false;
+                                    bs.count--; /* not accumulated */
+                                } else {
+                                    /* insertion into this lane's sorted row */
+                                    uint32_t j = nq;
+                                    while (j > 0 && qrow[j - 1] > v) { qrow[j] = qrow[j - 1]; j--; }
+                                    qrow[j] = v;
+                                    nq++;
+                                }
                             }
                         }
                     }
                 }
             }
         }
+        }
+        d.r.refill_commit();
     }
     if (in_range) out_errs[series] = err;
 }
 
+
+__global__ void __launch_bounds__(BLOCK_THREADS)
+k_count_errcode(const int32_t* __restrict__ errs, uint32_t n, int32_t code,
+                uint32_t* __restrict__ out) {
+    /* grid-stride count of one per-series error code: lets the rollup
+     * dispatch test for tier retries with a 4-byte D2H instead of pulling
+     * and scanning the whole error array every call */
+    uint32_t cnt = 0;
+    for (uint32_t i = blockIdx.x * blockDim.x + threadIdx.x; i < n;
+         i += gridDim.x * blockDim.x)
+        if (errs[i] == code) cnt++;
+    if (__any(cnt != 0)) {
+        cnt += __shfl_down(cnt, 32); cnt += __shfl_down(cnt, 16);
+        cnt += __shfl_down(cnt, 8); cnt += __shfl_down(cnt, 4);
+        cnt += __shfl_down(cnt, 2); cnt += __shfl_down(cnt, 1);
+        if ((threadIdx.x & (WAVE - 1)) == 0 && cnt) atomicAdd(out, cnt);
+    }
+}
 
 /* ===================== replica-deduplicating merge =====================
  * MultiReaderIterator semantics over one slice of R (<=4) replica rows of
@@ -2650,33 +2729,75 @@ int m3gpu_rollup_batch_dev_opts(
                 if (plan.qs[k] == q) { plan.qidx[i] = (int8_t)k; break; }
     }
 
+    /* extreme_cap: largest n (bounded by exact_cap) for which EVERY
+     * requested quantile resolves to sorted[n-1] — the bucket max — in
+     * BOTH regimes the lane kernel reproduces (quantilesFromBuf for n<=3,
+     * the calcQuantiles walk with the k_{i-1}+1 shift above). All-high
+     * quantile sets (p90+, e.g. the sum/min/max/p99 rollup) then need no
+     * per-point value staging at all: running min/max suffice. */
+    plan.extreme_cap = 0;
+    if (nq > 0) {
+        int capn = plan.exact_cap < QCAP ? plan.exact_cap : QCAP;
+        for (int n = 1; n <= capn; n++) {
+            bool allmax = true;
+            if (n <= 3) {
+                for (int i = 0; i < nq && allmax; i++) {
+                    int idx = (int)(qs[i] * (double)n);
+                    if (idx >= n) idx = n - 1;
+                    if (idx != n - 1) allmax = false;
+                }
+            } else {
+                int kk = 0;
+                for (int i = 0; i < nq; i++) {
+                    int rank = (int)ceil(qs[i] * (double)n);
+                    kk = (i == 0) ? rank : ((rank > kk + 1) ? rank : kk + 1);
+                    int kc = kk > n ? n : kk;
+                    if (kc != n) { allmax = false; break; }
+                }
+            }
+            if (!allmax) break;
+            plan.extreme_cap = n;
+        }
+    }
+
     hipStream_t s = (hipStream_t)hip_stream;
     bool with_q = plan.nq > 0 && metric_type == M3GPU_METRIC_TIMER;
+    /* deeper-than-cap buckets overflow to the wave tier; only use extreme
+     * mode when the cap covers normal windows (a lone p99 with the
+     * production-default eps has exact_cap 13, which bounds it) */
+    bool extreme = with_q && plan.extreme_cap >= 8;
     /* metric type is a template parameter: dead per-metric BucketState
      * fields drop out of the register file */
-    if (with_q)
-        hipLaunchKernelGGL((m3::k_rollup_lane<true, M3GPU_METRIC_TIMER>),
+    if (extreme)
+        hipLaunchKernelGGL((m3::k_rollup_lane<RQ_EXTREME, M3GPU_METRIC_TIMER>),
+                           dim3(grid_lane(nseries)),
+                           dim3(BLOCK_THREADS), 0, s,
+                           d_blobs, d_offsets, d_lens, nseries, int_optimized,
+                           default_unit, window_ns, nbuckets, plan,
+                           d_out, d_out_window_ts, d_out_errs);
+    else if (with_q)
+        hipLaunchKernelGGL((m3::k_rollup_lane<RQ_STAGED, M3GPU_METRIC_TIMER>),
                            dim3(grid_lane(nseries)),
                            dim3(BLOCK_THREADS), 0, s,
                            d_blobs, d_offsets, d_lens, nseries, int_optimized,
                            default_unit, window_ns, nbuckets, plan,
                            d_out, d_out_window_ts, d_out_errs);
     else if (metric_type == M3GPU_METRIC_COUNTER)
-        hipLaunchKernelGGL((m3::k_rollup_lane<false, M3GPU_METRIC_COUNTER>),
+        hipLaunchKernelGGL((m3::k_rollup_lane<RQ_NONE, M3GPU_METRIC_COUNTER>),
                            dim3(grid_lane(nseries)),
                            dim3(BLOCK_THREADS), 0, s,
                            d_blobs, d_offsets, d_lens, nseries, int_optimized,
                            default_unit, window_ns, nbuckets, plan,
                            d_out, d_out_window_ts, d_out_errs);
     else if (metric_type == M3GPU_METRIC_GAUGE)
-        hipLaunchKernelGGL((m3::k_rollup_lane<false, M3GPU_METRIC_GAUGE>),
+        hipLaunchKernelGGL((m3::k_rollup_lane<RQ_NONE, M3GPU_METRIC_GAUGE>),
                            dim3(grid_lane(nseries)),
                            dim3(BLOCK_THREADS), 0, s,
                            d_blobs, d_offsets, d_lens, nseries, int_optimized,
                            default_unit, window_ns, nbuckets, plan,
                            d_out, d_out_window_ts, d_out_errs);
     else
-        hipLaunchKernelGGL((m3::k_rollup_lane<false, M3GPU_METRIC_TIMER>),
+        hipLaunchKernelGGL((m3::k_rollup_lane<RQ_NONE, M3GPU_METRIC_TIMER>),
                            dim3(grid_lane(nseries)),
                            dim3(BLOCK_THREADS), 0, s,
                            d_blobs, d_offsets, d_lens, nseries, int_optimized,
@@ -2684,10 +2805,22 @@ int m3gpu_rollup_batch_dev_opts(
                            d_out, d_out_window_ts, d_out_errs);
     HIP_TRY(hipGetLastError());
     if (with_q) {
-        /* buckets deeper than RQCAP_LANE overflow the per-lane staging:
-         * retry exactly those series on the wave-per-series kernel (64-deep
-         * staging). Requires a stream sync to inspect the error flags. */
+        /* buckets deeper than the lane tier's capacity overflow to the
+         * wave-per-series kernel. The common case is zero overflows: test
+         * with a device-side count + 4-byte D2H instead of pulling and
+         * scanning the whole error array every call. */
+        static uint32_t* d_ovf = nullptr;
+        if (!d_ovf) HIP_TRY(hipMalloc(&d_ovf, sizeof(uint32_t)));
+        HIP_TRY(hipMemsetAsync(d_ovf, 0, sizeof(uint32_t), s));
+        hipLaunchKernelGGL(m3::k_count_errcode, dim3(256), dim3(BLOCK_THREADS),
+                           0, s, d_out_errs, nseries,
+                           M3GPU_SERIES_BUCKET_OVERFLOW, d_ovf);
+        HIP_TRY(hipGetLastError());
+        uint32_t h_ovf = 0;
+        HIP_TRY(hipMemcpyAsync(&h_ovf, d_ovf, sizeof(uint32_t),
+                               hipMemcpyDeviceToHost, s));
         HIP_TRY(hipStreamSynchronize(s));
+        if (h_ovf == 0) return M3GPU_OK;
         int32_t* h_errs = (int32_t*)malloc(nseries * sizeof(int32_t));
         if (!h_errs) { snprintf(g_err, sizeof(g_err), "oom"); return M3GPU_ERR_HIP; }
         hipError_t ce = hipMemcpy(h_errs, d_out_errs, nseries * sizeof(int32_t),
