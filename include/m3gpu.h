@@ -97,6 +97,21 @@ int m3gpu_decode_batch_dev_perm(
     int64_t* d_out_ts, double* d_out_vals, uint32_t* d_out_counts,
     int32_t* d_out_errs, uint32_t stride, void* hip_stream);
 
+/* Annotation-capturing variant (materializes Current()'s third return,
+ * iterator.go:226-231 / timestamp_iterator.go:327-356): d_out_ann is a
+ * per-series region of ann_stride bytes (4-aligned, >= 16) laid out as
+ *   [u32 n_events][n_events x {u32 point, u32 off, u32 len}][...bytes]
+ * where `point` is the 0-based index of the first datapoint the
+ * annotation applies to (sticky until replaced), and off/len locate the
+ * bytes within the region (they grow from the tail). A region too small
+ * for a series' annotations flags M3GPU_SERIES_CAPACITY on that series. */
+int m3gpu_decode_batch_dev_ann(
+    const uint8_t* d_blobs, const uint64_t* d_offsets, const uint32_t* d_lens,
+    uint32_t nseries, int int_optimized, uint8_t default_unit,
+    int64_t* d_out_ts, double* d_out_vals, uint32_t* d_out_counts,
+    int32_t* d_out_errs, uint32_t stride,
+    uint8_t* d_out_ann, uint32_t ann_stride, void* hip_stream);
+
 int m3gpu_decode_batch(
     const uint8_t* blobs, uint64_t blobs_len,
     const uint64_t* offsets, const uint32_t* lens,

@@ -261,8 +261,16 @@ struct Decoder {
         unit_ns = unit_valid(tu) ? UNIT_NS_D[tu] : 0;
     }
 
+    /* optional annotation capture (m3gpu_decode_batch_dev_ann): events
+     * {point, off, len} grow from the front of the per-series region,
+     * annotation bytes from the tail; collision flags CAPACITY */
+    uint8_t* ann_region;
+    uint32_t ann_stride_, ann_events, ann_tail, points;
+    int ann_err;
+
     __device__ __forceinline__ void init(const uint8_t* base, uint64_t off, uint32_t len,
-                         bool intopt, uint8_t dunit, uint64_t* ring = nullptr) {
+                         bool intopt, uint8_t dunit, uint64_t* ring = nullptr,
+                         uint8_t* ann = nullptr, uint32_t ann_stride = 0) {
         r.init(base, off, len, ring);
         prev_time = 0; prev_time_delta = 0;
         int_val = 0; prev_float_bits = 0; prev_xor = 0;
@@ -270,6 +278,12 @@ struct Decoder {
         default_unit = dunit;
         have_scheme = false; tu_changed = false; done = false; is_float = false;
         int_optimized = intopt;
+        ann_region = ann;
+        ann_stride_ = ann_stride;
+        ann_events = 0;
+        ann_tail = ann_stride;
+        points = 0;
+        ann_err = 0;
     }
 
     /* timestamp_iterator.go:115-135 */
@@ -306,20 +320,45 @@ struct Decoder {
         return M3GPU_SERIES_ANNOTATION;
     }
 
-    /* timestamp_iterator.go:327-356 (annotation bytes are skipped — the bulk
-     * decode surface does not return annotations) */
+    /* timestamp_iterator.go:327-356. Without a capture region the
+     * annotation bytes are skipped; with one (the _ann decode surface)
+     * each annotation-set event is recorded as {point, off, len} so the
+     * caller can materialize Current()'s sticky PrevAnt
+     * (iterator.go:226-231) by carrying events forward. */
     __device__ __forceinline__ int skip_annotation() {
         int64_t alen;
         int err = read_varint(&alen);
         if (err) return err;
         alen += 1;
         if (alen <= 0) return M3GPU_SERIES_ANNOTATION;
+        uint8_t* dst = nullptr;
+        if (ann_region && !ann_err) {
+            uint32_t ev_end = 4 + (ann_events + 1) * 12;
+            if (alen > (int64_t)ann_tail || ev_end > ann_tail - (uint32_t)alen) {
+                ann_err = M3GPU_SERIES_CAPACITY; /* region full: flag, keep
+                                                  * decoding values */
+            } else {
+                ann_tail -= (uint32_t)alen;
+                dst = ann_region + ann_tail;
+                uint32_t* ev = (uint32_t*)(ann_region + 4 + ann_events * 12);
+                ev[0] = points; /* index of the point this starts applying to */
+                ev[1] = ann_tail;
+                ev[2] = (uint32_t)alen;
+                ann_events++;
+            }
+        }
         for (int64_t i = 0; i < alen; i++) {
             uint64_t b;
             err = r.read_bits(8, &b);
             if (err) return err;
+            if (dst) dst[i] = (uint8_t)b;
         }
         return 0;
+    }
+    __device__ __forceinline__ void ann_finish(int* err_io) {
+        if (!ann_region) return;
+        *(uint32_t*)ann_region = ann_events;
+        if (ann_err && *err_io == 0) *err_io = ann_err;
     }
 
     /* timestamp_iterator.go:307-325 */
@@ -775,7 +814,10 @@ struct Decoder {
     __device__ __forceinline__ int next(int64_t* t, double* v) {
         if (done) return 0;
         int f = next_fused(t, v);
-        if (f != -1000) return f;
+        if (f != -1000) {
+            points += (f == 1);
+            return f;
+        }
         bool first;
         int err = read_timestamp(&first);
         if (err) return -err;
@@ -786,6 +828,7 @@ struct Decoder {
         if (!int_optimized || is_float) *v = bits2f(prev_float_bits);
         else if (mult != 0) *v = int_val / exp10_table(mult);
         else *v = int_val;
+        points++;
         return 1;
     }
     __device__ __forceinline__ double exp10_table(uint8_t m) {
@@ -825,7 +868,8 @@ k_decode_batch(const uint8_t* __restrict__ blobs,
                uint32_t nseries, int int_optimized, uint8_t default_unit,
                int64_t* __restrict__ out_ts, double* __restrict__ out_vals,
                uint32_t* __restrict__ out_counts, int32_t* __restrict__ out_errs,
-               uint32_t stride) {
+               uint32_t stride,
+               uint8_t* __restrict__ out_ann, uint32_t ann_stride) {
     const uint32_t lane = threadIdx.x & (WAVE - 1);
     const uint32_t wave = __builtin_amdgcn_readfirstlane(threadIdx.x >> 6);
     const uint32_t slot = blockIdx.x * BLOCK_THREADS + wave * WAVE + lane;
@@ -845,7 +889,10 @@ k_decode_batch(const uint8_t* __restrict__ blobs,
     const bool in_range = slot < nseries;
     Decoder d;
     d.init(blobs, in_range ? offsets[series] : 0, in_range ? lens[series] : 0,
-           int_optimized != 0, default_unit, ring);
+           int_optimized != 0, default_unit, ring,
+           (out_ann && in_range) ? out_ann + (uint64_t)series * ann_stride
+                                 : nullptr,
+           ann_stride);
 
     bool running = in_range;
     int err = 0;
@@ -935,6 +982,7 @@ k_decode_batch(const uint8_t* __restrict__ blobs,
     }
 
     if (in_range) {
+        d.ann_finish(&err);
         out_counts[series] = cnt;
         out_errs[series] = err;
     }
@@ -2500,7 +2548,30 @@ int m3gpu_decode_batch_dev_perm(
                        dim3(BLOCK_THREADS), 0, s,
                        d_blobs, d_offsets, d_lens, d_perm, nseries,
                        int_optimized, default_unit, d_out_ts, d_out_vals,
-                       d_out_counts, d_out_errs, stride);
+                       d_out_counts, d_out_errs, stride, (uint8_t*)NULL, 0);
+    HIP_TRY(hipGetLastError());
+    return M3GPU_OK;
+}
+
+int m3gpu_decode_batch_dev_ann(
+    const uint8_t* d_blobs, const uint64_t* d_offsets, const uint32_t* d_lens,
+    uint32_t nseries, int int_optimized, uint8_t default_unit,
+    int64_t* d_out_ts, double* d_out_vals, uint32_t* d_out_counts,
+    int32_t* d_out_errs, uint32_t stride,
+    uint8_t* d_out_ann, uint32_t ann_stride, void* hip_stream) {
+    if (!nseries) return M3GPU_OK;
+    if (ann_stride % 4 || ann_stride < 16) {
+        snprintf(g_err, sizeof(g_err),
+                 "ann_stride must be 4-byte aligned and >= 16");
+        return M3GPU_ERR_BADARG;
+    }
+    hipStream_t s = (hipStream_t)hip_stream;
+    hipLaunchKernelGGL(m3::k_decode_batch, dim3(grid_lane(nseries)),
+                       dim3(BLOCK_THREADS), 0, s,
+                       d_blobs, d_offsets, d_lens, (const int32_t*)NULL,
+                       nseries, int_optimized, default_unit, d_out_ts,
+                       d_out_vals, d_out_counts, d_out_errs, stride,
+                       d_out_ann, ann_stride);
     HIP_TRY(hipGetLastError());
     return M3GPU_OK;
 }

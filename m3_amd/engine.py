@@ -65,6 +65,10 @@ def lib():
         L.m3gpu_decode_batch_dev_perm.argtypes = [c_vp, c_vp, c_vp, c_vp, c_u32,
                                                   c_int, c_u8, c_vp, c_vp, c_vp,
                                                   c_vp, c_u32, c_vp]
+        L.m3gpu_decode_batch_dev_ann.restype = c_int
+        L.m3gpu_decode_batch_dev_ann.argtypes = [c_vp, c_vp, c_vp, c_u32,
+                                                 c_int, c_u8, c_vp, c_vp, c_vp,
+                                                 c_vp, c_u32, c_vp, c_u32, c_vp]
         L.m3gpu_decode_batch.restype = c_int
         L.m3gpu_decode_batch.argtypes = [P(c_u8), c_u64, P(c_u64), P(c_u32), c_u32,
                                          c_int, c_u8, P(c_i64), P(c_f64), P(c_u32),
@@ -222,6 +226,54 @@ def decode_batch_dev(d_blob, d_offsets, d_lens, out_ts, out_vals, out_counts,
         _dev_ptr(out_vals), _dev_ptr(out_counts), _dev_ptr(out_errs), stride,
         _torch_stream())
     _check(rc, "m3gpu_decode_batch_dev")
+
+
+def decode_batch_dev_ann(d_blob, d_offsets, d_lens, out_ts, out_vals,
+                         out_counts, out_errs, out_ann, int_optimized=True,
+                         default_unit=1):
+    """Annotation-capturing decode (materializes ReaderIterator.Current()'s
+    third return, iterator.go:226-231). out_ann: uint8 [nseries, ann_stride]
+    device tensor; each series' region holds
+    [u32 n_events][n_events x {u32 point, u32 off, u32 len}][...bytes]
+    (bytes grow from the tail). Decode with parse_ann_region / iterate with
+    decoded_ann_per_point."""
+    nseries = d_lens.numel()
+    stride = out_ts.shape[1]
+    rc = lib().m3gpu_decode_batch_dev_ann(
+        _dev_ptr(d_blob), _dev_ptr(d_offsets), _dev_ptr(d_lens), nseries,
+        1 if int_optimized else 0, default_unit, _dev_ptr(out_ts),
+        _dev_ptr(out_vals), _dev_ptr(out_counts), _dev_ptr(out_errs), stride,
+        _dev_ptr(out_ann), out_ann.shape[1], _torch_stream())
+    _check(rc, "m3gpu_decode_batch_dev_ann")
+
+
+def parse_ann_region(region):
+    """Parse one series' annotation region (host uint8 array) into a list of
+    (point_idx, bytes) set-events in stream order."""
+    region = np.asarray(region, dtype=np.uint8)
+    n = int(region[:4].view(np.uint32)[0])
+    out = []
+    for i in range(n):
+        ev = region[4 + i * 12: 4 + i * 12 + 12].view(np.uint32)
+        point, off, ln = int(ev[0]), int(ev[1]), int(ev[2])
+        out.append((point, bytes(region[off:off + ln])))
+    return out
+
+
+def decoded_ann_per_point(region, npts):
+    """Materialize the sticky PrevAnt view (iterator.go:226-231): the
+    annotation returned with each of npts datapoints (None until the first
+    set-event)."""
+    events = parse_ann_region(region)
+    out = [None] * npts
+    cur = None
+    j = 0
+    for p in range(npts):
+        while j < len(events) and events[j][0] <= p:
+            cur = events[j][1]
+            j += 1
+        out[p] = cur
+    return out
 
 
 def encode_batch_dev(d_ts, d_vals, d_counts, out_bytes, out_lens, out_errs,

@@ -119,6 +119,9 @@ def encode_series(ts_ns, vals, units=None, annotations=None, start_ns=None,
         ann_off = np.asarray(offs, dtype=np.int32)
         ann_bytes = np.frombuffer(bytes(blob), dtype=np.uint8) if blob else np.zeros(1, np.uint8)
     cap = 32 + n * 24
+    if annotations is not None:
+        # annotation markers: 11-bit marker + varint len + bytes each
+        cap += sum(16 + len(a) for a in annotations if a)
     out = np.zeros(cap, dtype=np.uint8)
     L = lib()
     args = [ts.ctypes.data_as(P(c_i64)), v.ctypes.data_as(P(c_f64)),

@@ -882,3 +882,96 @@ def test_rollup_nondefault_ckms_options(torch, engine):
     with _pytest.raises(engine.M3GpuError):
         engine.rollup_batch_dev(d_blob, d_off, d_lens, engine.METRIC_TIMER,
                                 window, 2, aggs, out, wts, errs, every=2000)
+
+
+def test_decode_annotations(torch, engine):
+    """Annotation-returning bulk decode (m3gpu_decode_batch_dev_ann) vs the
+    oracle: the reference's roundtrip annotation shapes
+    (roundtrip_test.go:135-148 — repeated "foo"/"bar" runs, dedupe via
+    xxhash) plus long and binary annotations. Events must reproduce the
+    sticky PrevAnt view (iterator.go:226-231) byte-for-byte."""
+    rng = np.random.default_rng(1234)
+    start = 1427162462 * 10**9
+    streams, ann_truth = [], []
+    cases = []
+    # repeated foo/bar runs (reference roundtrip shapes)
+    cases.append([b"foo"] * 10 + [b"bar"] * 10 + [b"foo"] * 5)
+    # sparse: annotation only at points 0, 7, 8
+    sparse = [None] * 25
+    sparse[0], sparse[7], sparse[8] = b"first", b"mid", b"mid2"
+    cases.append(sparse)
+    # none at all
+    cases.append([None] * 30)
+    # binary + long annotations, changing every point
+    cases.append([bytes(rng.integers(0, 256, 40, dtype=np.uint8).tolist())
+                  for _ in range(12)])
+    for anns in cases:
+        n = len(anns)
+        ts = start + np.arange(n) * 10**9
+        vals = np.round(rng.random(n) * 100, 2)
+        streams.append(oracle.encode_series(ts, vals, annotations=anns,
+                                            start_ns=start))
+        dec = oracle.decode_series(streams[-1], with_annotations=True)
+        # oracle gives per-point NEW annotations; carry forward = PrevAnt
+        cur, ref = None, []
+        for a in dec["annotations"]:
+            if a is not None:
+                cur = a
+            ref.append(cur)
+        ann_truth.append((dec["ts"], dec["vals"], ref))
+
+    from m3_amd.engine import (pack_streams, decode_batch_dev_ann,
+                               decoded_ann_per_point)
+    blob, offsets, lens = pack_streams(streams)
+    dev = "cuda:0"
+    d_blob = torch.from_numpy(blob).to(dev)
+    d_off = torch.from_numpy(offsets[:-1].astype(np.int64)).to(dev)
+    d_lens = torch.from_numpy(lens.astype(np.int32)).to(dev)
+    ns, stride = len(streams), 64
+    out_ts = torch.empty((ns, stride), dtype=torch.int64, device=dev)
+    out_vals = torch.empty((ns, stride), dtype=torch.float64, device=dev)
+    out_counts = torch.empty(ns, dtype=torch.int32, device=dev)
+    out_errs = torch.empty(ns, dtype=torch.int32, device=dev)
+    out_ann = torch.zeros((ns, 1024), dtype=torch.uint8, device=dev)
+    decode_batch_dev_ann(d_blob, d_off, d_lens, out_ts, out_vals, out_counts,
+                         out_errs, out_ann)
+    torch.cuda.synchronize()
+    assert torch.all(out_errs == 0).item()
+    h_ann = out_ann.cpu().numpy()
+    for i, (ref_ts, ref_vals, ref_anns) in enumerate(ann_truth):
+        n = int(out_counts[i].item())
+        assert n == len(ref_ts)
+        assert np.array_equal(out_ts[i, :n].cpu().numpy(), ref_ts)
+        assert np.array_equal(out_vals[i, :n].cpu().numpy().view(np.uint64),
+                              np.asarray(ref_vals).view(np.uint64))
+        got = decoded_ann_per_point(h_ann[i], n)
+        assert got == ref_anns, f"series {i}"
+
+
+def test_decode_annotations_region_overflow(torch, engine):
+    """A region too small for a series' annotations flags CAPACITY on that
+    series only; values still decode."""
+    rng = np.random.default_rng(55)
+    start = 1427162462 * 10**9
+    anns = [bytes([65 + (i % 26)]) * 50 for i in range(20)]  # 1000B of anns
+    ts = start + np.arange(20) * 10**9
+    vals = rng.random(20)
+    big = oracle.encode_series(ts, vals, annotations=anns, start_ns=start)
+    small = oracle.encode_series(ts, vals, start_ns=start)
+    from m3_amd.engine import pack_streams, decode_batch_dev_ann
+    blob, offsets, lens = pack_streams([big, small])
+    dev = "cuda:0"
+    d_blob = torch.from_numpy(blob).to(dev)
+    d_off = torch.from_numpy(offsets[:-1].astype(np.int64)).to(dev)
+    d_lens = torch.from_numpy(lens.astype(np.int32)).to(dev)
+    out_ts = torch.empty((2, 32), dtype=torch.int64, device=dev)
+    out_vals = torch.empty((2, 32), dtype=torch.float64, device=dev)
+    out_counts = torch.empty(2, dtype=torch.int32, device=dev)
+    out_errs = torch.empty(2, dtype=torch.int32, device=dev)
+    out_ann = torch.zeros((2, 128), dtype=torch.uint8, device=dev)  # too small
+    decode_batch_dev_ann(d_blob, d_off, d_lens, out_ts, out_vals, out_counts,
+                         out_errs, out_ann)
+    torch.cuda.synchronize()
+    assert int(out_errs[0].item()) == 6  # capacity
+    assert int(out_errs[1].item()) == 0
+    assert int(out_counts[0].item()) == 20  # values still decoded
