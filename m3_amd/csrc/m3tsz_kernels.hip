@@ -169,15 +169,25 @@ struct BitReader {
      * order, so the commit's counted wait isolates these loads and never
      * drains the in-flight output stores. The loads' latency hides under
      * the whole tile's parsing. */
+#ifndef RF_LOW
+#define RF_LOW 3 /* refill when <= this many ring words remain. Topping up
+                  * every tile (word-granular bursts, L1-amortized) measured
+                  * faster than whole-chunk-only refills (RF_LOW 0): the ring
+                  * then never dries mid-tile, so the parser's underrun
+                  * fallback (an in-chain global gather whose wait also
+                  * drains in-flight output stores) stays off the hot path */
+#endif
     __device__ __forceinline__ void refill_issue(bool active) {
         if (!lds) return;
         if (rfill < wnext) rfill = wnext; /* resync: underrun consumed
                                            * [rfill, wnext) directly */
         uint32_t used = rfill - wnext;
         pend_n = 0;
-        if (active && used == 0 && rfill < wtotal) {
+        if (active && used <= RF_LOW && rfill < wtotal) {
             uint32_t n = wtotal - rfill;
-            pend_n = n > IN_WORDS ? IN_WORDS : n;
+            uint32_t room = IN_WORDS - used;
+            if (n > room) n = room;
+            pend_n = n;
 #pragma unroll
             for (uint32_t i = 0; i < IN_WORDS; i++)
                 if (i < pend_n) pend[i] = __builtin_bswap64(words[rfill + i]);
