@@ -11,7 +11,10 @@ Semantics mirrored:
   - Next() advances and reports whether a point is available
     (iterator.go:81-100: first Next positions on the first point);
   - Current() returns (timestamp_ns, value, unit) for the current point
-    and is only valid after a true Next() (types.go:200);
+    and is only valid after a true Next() (types.go:200); CurrentAnnotation()
+    gives the sticky annotation the reference returns as Current()'s third
+    value (iterator.go:226-231) when the iterator was built with the
+    annotation-set events of m3gpu_decode_batch_dev_ann;
   - Err() returns the sticky per-series error (the engine's
     M3GPU_SERIES_* code mapped to a string, or None);
   - Close() releases references (pooling is the caller's concern here).
@@ -24,13 +27,18 @@ from .engine import SERIES_ERRORS
 class SliceReaderIterator:
     """`encoding.ReaderIterator` over one decoded series row."""
 
-    def __init__(self, ts, vals, count, err=0, unit=1):
+    def __init__(self, ts, vals, count, err=0, unit=1, ann_events=None):
         self._ts = ts
         self._vals = vals
         self._n = int(count)
         self._err = int(err)
         self._unit = unit
         self._i = -1
+        # annotation-set events [(point_idx, bytes)] in stream order
+        # (engine.parse_ann_region); carried forward like PrevAnt
+        self._ann_events = ann_events or []
+        self._ann_j = 0
+        self._ann = None
 
     def Next(self):
         if self._err != 0:
@@ -38,12 +46,23 @@ class SliceReaderIterator:
         if self._i + 1 >= self._n:
             return False
         self._i += 1
+        while (self._ann_j < len(self._ann_events) and
+               self._ann_events[self._ann_j][0] <= self._i):
+            self._ann = self._ann_events[self._ann_j][1]
+            self._ann_j += 1
         return True
 
     def Current(self):
         if self._i < 0 or self._i >= self._n:
             raise RuntimeError("Current() before a successful Next()")
         return int(self._ts[self._i]), float(self._vals[self._i]), self._unit
+
+    def CurrentAnnotation(self):
+        """The reference Current()'s third return (sticky PrevAnt,
+        iterator.go:226-231): bytes or None."""
+        if self._i < 0 or self._i >= self._n:
+            raise RuntimeError("CurrentAnnotation() before a successful Next()")
+        return self._ann
 
     def Err(self):
         if self._err == 0:
@@ -62,20 +81,27 @@ class BatchIterators:
     plus per-series counts/errs, e.g. the (moved-to-host) outputs of
     `decode_batch_dev` or `fileset_ingest_dev`."""
 
-    def __init__(self, ts, vals, counts, errs=None, unit=1):
+    def __init__(self, ts, vals, counts, errs=None, unit=1, ann_regions=None):
         self._ts = np.asarray(ts)
         self._vals = np.asarray(vals)
         self._counts = np.asarray(counts)
         self._errs = np.asarray(errs) if errs is not None else None
         self._unit = unit
+        # optional [nseries, ann_stride] uint8 regions from
+        # m3gpu_decode_batch_dev_ann
+        self._ann = np.asarray(ann_regions) if ann_regions is not None else None
 
     def __len__(self):
         return len(self._counts)
 
     def iterator(self, i):
         err = int(self._errs[i]) if self._errs is not None else 0
+        events = None
+        if self._ann is not None:
+            from .engine import parse_ann_region
+            events = parse_ann_region(self._ann[i])
         return SliceReaderIterator(self._ts[i], self._vals[i],
-                                   self._counts[i], err, self._unit)
+                                   self._counts[i], err, self._unit, events)
 
     def __iter__(self):
         for i in range(len(self)):

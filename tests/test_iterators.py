@@ -67,3 +67,21 @@ def test_iterator_over_oracle_decode():
         t, v, _ = it.Current()
         out.append((t, v))
     assert out == list(zip(ts.tolist(), vals.tolist()))
+
+
+def test_iterator_annotations_sticky():
+    """CurrentAnnotation() mirrors the reference's sticky PrevAnt
+    (iterator.go:226-231) from annotation-set events."""
+    from m3_amd.iterators import SliceReaderIterator
+    ts = np.arange(6, dtype=np.int64)
+    vals = np.arange(6, dtype=np.float64)
+    events = [(0, b"a"), (2, b"b"), (3, b"c")]
+    it = SliceReaderIterator(ts, vals, 6, ann_events=events)
+    seen = []
+    while it.Next():
+        seen.append(it.CurrentAnnotation())
+    assert seen == [b"a", b"a", b"b", b"c", b"c", b"c"]
+    # no events: always None
+    it = SliceReaderIterator(ts, vals, 3)
+    it.Next()
+    assert it.CurrentAnnotation() is None
