@@ -112,6 +112,15 @@ int m3gpu_decode_batch_dev_ann(
     int32_t* d_out_errs, uint32_t stride,
     uint8_t* d_out_ann, uint32_t ann_stride, void* hip_stream);
 
+/* host-pointer convenience form of the annotation-capturing decode */
+int m3gpu_decode_batch_ann(
+    const uint8_t* blobs, uint64_t blobs_len,
+    const uint64_t* offsets, const uint32_t* lens,
+    uint32_t nseries, int int_optimized, uint8_t default_unit,
+    int64_t* out_ts, double* out_vals, uint32_t* out_counts,
+    int32_t* out_errs, uint32_t stride,
+    uint8_t* out_ann, uint32_t ann_stride);
+
 int m3gpu_decode_batch(
     const uint8_t* blobs, uint64_t blobs_len,
     const uint64_t* offsets, const uint32_t* lens,

@@ -24,6 +24,7 @@ package m3gpu
 import "C"
 
 import (
+	"encoding/binary"
 	"errors"
 	"unsafe"
 )
@@ -160,4 +161,68 @@ func RollupBatch(
 		return nil, nil, nil, nil, lastError()
 	}
 	return out, windowTs, errs, nil
+}
+
+// AnnotationEvent is one annotation-set event from the annotation-capturing
+// decode: the annotation starts applying at Point (0-based) and stays the
+// Current() third return (iterator.go:226-231 sticky PrevAnt) until the
+// next event.
+type AnnotationEvent struct {
+	Point uint32
+	Bytes []byte
+}
+
+// DecodeBatchAnn decodes like DecodeBatch and additionally captures every
+// annotation-set event per series (m3gpu_decode_batch_ann; region layout in
+// m3gpu.h). annStride sizes the per-series region (4-aligned, >= 16); a
+// series whose annotations overflow it flags SeriesCapacity but still
+// decodes its values.
+func DecodeBatchAnn(
+	blobs []byte, offsets []uint64, lens []uint32,
+	intOptimized bool, defaultUnit byte, stride uint32, annStride uint32,
+) (ts []int64, vals []float64, counts []uint32, errs []SeriesError,
+	anns [][]AnnotationEvent, err error) {
+	n := uint32(len(lens))
+	if n == 0 {
+		return nil, nil, nil, nil, nil, nil
+	}
+	ts = make([]int64, uint64(n)*uint64(stride))
+	vals = make([]float64, uint64(n)*uint64(stride))
+	counts = make([]uint32, n)
+	errs = make([]SeriesError, n)
+	region := make([]byte, uint64(n)*uint64(annStride))
+	intOpt := C.int(0)
+	if intOptimized {
+		intOpt = 1
+	}
+	rc := C.m3gpu_decode_batch_ann(
+		(*C.uint8_t)(unsafe.Pointer(&blobs[0])), C.uint64_t(len(blobs)),
+		(*C.uint64_t)(unsafe.Pointer(&offsets[0])),
+		(*C.uint32_t)(unsafe.Pointer(&lens[0])),
+		C.uint32_t(n), intOpt, C.uint8_t(defaultUnit),
+		(*C.int64_t)(unsafe.Pointer(&ts[0])),
+		(*C.double)(unsafe.Pointer(&vals[0])),
+		(*C.uint32_t)(unsafe.Pointer(&counts[0])),
+		(*C.int32_t)(unsafe.Pointer(&errs[0])), C.uint32_t(stride),
+		(*C.uint8_t)(unsafe.Pointer(&region[0])), C.uint32_t(annStride))
+	if rc != 0 {
+		return nil, nil, nil, nil, nil, lastError()
+	}
+	anns = make([][]AnnotationEvent, n)
+	le := binary.LittleEndian
+	for i := uint32(0); i < n; i++ {
+		r := region[uint64(i)*uint64(annStride) : uint64(i+1)*uint64(annStride)]
+		nev := le.Uint32(r[0:4])
+		evs := make([]AnnotationEvent, 0, nev)
+		for j := uint32(0); j < nev; j++ {
+			ev := r[4+j*12 : 4+j*12+12]
+			off, ln := le.Uint32(ev[4:8]), le.Uint32(ev[8:12])
+			evs = append(evs, AnnotationEvent{
+				Point: le.Uint32(ev[0:4]),
+				Bytes: append([]byte(nil), r[off:off+ln]...),
+			})
+		}
+		anns[i] = evs
+	}
+	return ts, vals, counts, errs, anns, nil
 }

@@ -3054,6 +3054,55 @@ int m3gpu_decode_batch(
     return rc;
 }
 
+int m3gpu_decode_batch_ann(
+    const uint8_t* blobs, uint64_t blobs_len,
+    const uint64_t* offsets, const uint32_t* lens,
+    uint32_t nseries, int int_optimized, uint8_t default_unit,
+    int64_t* out_ts, double* out_vals, uint32_t* out_counts,
+    int32_t* out_errs, uint32_t stride,
+    uint8_t* out_ann, uint32_t ann_stride) {
+    /* host-pointer convenience form of m3gpu_decode_batch_dev_ann (what a
+     * cgo ReaderIterator shim calls for annotation-bearing blocks) */
+    uint8_t* d_blobs = nullptr;
+    uint64_t* d_offsets = nullptr;
+    uint32_t* d_lens = nullptr;
+    int64_t* d_ts = nullptr;
+    double* d_vals = nullptr;
+    uint32_t* d_counts = nullptr;
+    int32_t* d_errs = nullptr;
+    uint8_t* d_ann = nullptr;
+    uint64_t npts = (uint64_t)nseries * stride;
+    uint64_t ann_total = (uint64_t)nseries * ann_stride;
+    int rc = M3GPU_OK;
+    HIP_TRY(hipMalloc(&d_blobs, blobs_len));
+    HIP_TRY(hipMalloc(&d_offsets, (nseries + 1) * sizeof(uint64_t)));
+    HIP_TRY(hipMalloc(&d_lens, nseries * sizeof(uint32_t)));
+    HIP_TRY(hipMalloc(&d_ts, npts * sizeof(int64_t)));
+    HIP_TRY(hipMalloc(&d_vals, npts * sizeof(double)));
+    HIP_TRY(hipMalloc(&d_counts, nseries * sizeof(uint32_t)));
+    HIP_TRY(hipMalloc(&d_errs, nseries * sizeof(int32_t)));
+    HIP_TRY(hipMalloc(&d_ann, ann_total));
+    HIP_TRY(hipMemset(d_ann, 0, ann_total));
+    HIP_TRY(hipMemcpy(d_blobs, blobs, blobs_len, hipMemcpyHostToDevice));
+    HIP_TRY(hipMemcpy(d_offsets, offsets, (nseries + 1) * sizeof(uint64_t), hipMemcpyHostToDevice));
+    HIP_TRY(hipMemcpy(d_lens, lens, nseries * sizeof(uint32_t), hipMemcpyHostToDevice));
+    rc = m3gpu_decode_batch_dev_ann(d_blobs, d_offsets, d_lens, nseries,
+                                    int_optimized, default_unit, d_ts, d_vals,
+                                    d_counts, d_errs, stride, d_ann,
+                                    ann_stride, nullptr);
+    if (rc == M3GPU_OK) {
+        HIP_TRY(hipMemcpy(out_ts, d_ts, npts * sizeof(int64_t), hipMemcpyDeviceToHost));
+        HIP_TRY(hipMemcpy(out_vals, d_vals, npts * sizeof(double), hipMemcpyDeviceToHost));
+        HIP_TRY(hipMemcpy(out_counts, d_counts, nseries * sizeof(uint32_t), hipMemcpyDeviceToHost));
+        HIP_TRY(hipMemcpy(out_errs, d_errs, nseries * sizeof(int32_t), hipMemcpyDeviceToHost));
+        HIP_TRY(hipMemcpy(out_ann, d_ann, ann_total, hipMemcpyDeviceToHost));
+    }
+    (void)hipFree(d_blobs); (void)hipFree(d_offsets); (void)hipFree(d_lens);
+    (void)hipFree(d_ts); (void)hipFree(d_vals); (void)hipFree(d_counts);
+    (void)hipFree(d_errs); (void)hipFree(d_ann);
+    return rc;
+}
+
 int m3gpu_encode_batch(
     const int64_t* ts, const double* vals, const uint32_t* counts,
     uint32_t nseries, uint32_t stride, int int_optimized, uint8_t unit,

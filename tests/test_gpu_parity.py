@@ -975,3 +975,48 @@ def test_decode_annotations_region_overflow(torch, engine):
     assert int(out_errs[0].item()) == 6  # capacity
     assert int(out_errs[1].item()) == 0
     assert int(out_counts[0].item()) == 20  # values still decoded
+
+
+def test_decode_annotations_host_form(torch, engine):
+    """m3gpu_decode_batch_ann (host-pointer convenience form, the cgo
+    surface) produces the same values + annotation events as the _dev
+    form."""
+    import ctypes
+    from ctypes import POINTER as P, c_uint8, c_uint64, c_uint32, c_int64, \
+        c_double, c_int32
+    rng = np.random.default_rng(77)
+    start = 1427162462 * 10**9
+    anns = [b"x", None, b"yy", None, b"zzz"] * 4
+    ts = start + np.arange(20) * 10**9
+    vals = np.round(rng.random(20) * 10, 3)
+    stream = oracle.encode_series(ts, vals, annotations=anns, start_ns=start)
+    from m3_amd.engine import pack_streams, decoded_ann_per_point, lib
+    blob, offsets, lens = pack_streams([stream])
+    L = lib()
+    L.m3gpu_decode_batch_ann.restype = ctypes.c_int
+    L.m3gpu_decode_batch_ann.argtypes = [
+        P(c_uint8), c_uint64, P(c_uint64), P(c_uint32), c_uint32,
+        ctypes.c_int, c_uint8, P(c_int64), P(c_double), P(c_uint32),
+        P(c_int32), c_uint32, P(c_uint8), c_uint32]
+    stride, astride = 32, 256
+    o_ts = np.zeros((1, stride), np.int64)
+    o_vals = np.zeros((1, stride), np.float64)
+    o_counts = np.zeros(1, np.uint32)
+    o_errs = np.zeros(1, np.int32)
+    o_ann = np.zeros((1, astride), np.uint8)
+    rc = L.m3gpu_decode_batch_ann(
+        blob.ctypes.data_as(P(c_uint8)), blob.nbytes,
+        offsets.ctypes.data_as(P(c_uint64)), lens.ctypes.data_as(P(c_uint32)),
+        1, 1, 1,
+        o_ts.ctypes.data_as(P(c_int64)), o_vals.ctypes.data_as(P(c_double)),
+        o_counts.ctypes.data_as(P(c_uint32)), o_errs.ctypes.data_as(P(c_int32)),
+        stride, o_ann.ctypes.data_as(P(c_uint8)), astride)
+    assert rc == 0 and o_errs[0] == 0 and o_counts[0] == 20
+    assert np.array_equal(o_ts[0, :20], ts)
+    got = decoded_ann_per_point(o_ann[0], 20)
+    cur, ref = None, []
+    for a in anns:
+        if a is not None:
+            cur = a
+        ref.append(cur)
+    assert got == ref
