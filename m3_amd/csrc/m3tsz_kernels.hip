@@ -82,10 +82,15 @@ __device__ __forceinline__ int64_t go_f2i(double v) {
 __device__ __forceinline__ uint64_t f2bits(double v) { return __double_as_longlong(v); }
 __device__ __forceinline__ double bits2f(uint64_t b) { return __longlong_as_double((long long)b); }
 
-/* Go math.Modf: Modf(+-Inf) = (+-Inf, NaN) */
+/* Go math.Modf: Modf(+-Inf) = (+-Inf, NaN). The finite path is exact
+ * trunc-toward-zero + subtract (v - trunc(v) is exact in IEEE — Sterbenz
+ * for |v| >= 1, and trunc(v) == 0 below), one v_trunc_f64 + v_add_f64
+ * instead of the libm modf sequence. */
 __device__ __forceinline__ double go_modf(double v, double* ip) {
     if (isinf(v)) { *ip = v; return __longlong_as_double(0x7ff8000000000000LL); }
-    return modf(v, ip);
+    double i = trunc(v);
+    *ip = i;
+    return v - i;
 }
 
 /* ===================== device bit reader ===================== */
@@ -1062,7 +1067,20 @@ struct BitWriter {
     }
 };
 
-/* m3tsz.go:78-119 convertToIntFloat — bit-sensitive: -ffp-contract=off */
+/* m3tsz.go:78-119 convertToIntFloat — bit-sensitive: -ffp-contract=off.
+ * math.Nextafter on a strictly-positive finite double is a 1-ulp step,
+ * which for IEEE positives is an integer inc/dec of the bit pattern —
+ * exact, and much cheaper than the libm nextafter sequence inside this
+ * up-to-7-iteration loop (val > 0 is guaranteed here: val == 0 has
+ * r == 0 and returns first; toward-0 steps down, toward i+1 > val steps
+ * up). */
+__device__ __forceinline__ double go_next_down(double x) { /* x > 0 finite */
+    return __longlong_as_double(__double_as_longlong(x) - 1);
+}
+__device__ __forceinline__ double go_next_up(double x) { /* x > 0 finite */
+    return __longlong_as_double(__double_as_longlong(x) + 1);
+}
+
 __device__ __forceinline__ int convert_to_int_float(double v, uint8_t cur_max_mult,
                                     double* out_val, uint8_t* out_mult, bool* out_is_float) {
     const double mult_tab[7] = {1.0, 10.0, 100.0, 1000.0, 10000.0, 100000.0, 1000000.0};
@@ -1082,10 +1100,10 @@ __device__ __forceinline__ int convert_to_int_float(double v, uint8_t cur_max_mu
         r = go_modf(val, &i);
         if (r == 0) { *out_val = sign * i; *out_mult = m; *out_is_float = false; return 0; }
         else if (r < 0.1) {
-            if (nextafter(val, 0.0) <= i) { *out_val = sign * i; *out_mult = m; *out_is_float = false; return 0; }
+            if (go_next_down(val) <= i) { *out_val = sign * i; *out_mult = m; *out_is_float = false; return 0; }
         } else if (r > 0.9) {
             double nxt = i + 1;
-            if (nextafter(val, nxt) >= nxt) { *out_val = sign * nxt; *out_mult = m; *out_is_float = false; return 0; }
+            if (go_next_up(val) >= nxt) { *out_val = sign * nxt; *out_mult = m; *out_is_float = false; return 0; }
         }
     }
     *out_val = v; *out_mult = 0; *out_is_float = true;
