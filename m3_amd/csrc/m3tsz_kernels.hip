@@ -74,6 +74,19 @@ __device__ __forceinline__ int64_t sign_extend(uint64_t v, uint32_t nbits) {
     uint32_t sh = 64 - nbits;
     return ((int64_t)(v << sh)) >> sh;
 }
+/* 10^m for m in [0,6] as a pure VALU select chain (see exp10_table note:
+ * an indexed const array would cost a global load + vmcnt(0) per use) */
+__device__ __forceinline__ double exp10_sel(uint8_t m) {
+    double r = 1.0;
+    r = (m == 1) ? 10.0 : r;
+    r = (m == 2) ? 100.0 : r;
+    r = (m == 3) ? 1000.0 : r;
+    r = (m == 4) ? 10000.0 : r;
+    r = (m == 5) ? 100000.0 : r;
+    r = (m == 6) ? 1000000.0 : r;
+    return r;
+}
+
 /* Go float64->int64 (amd64 CVTTSD2SQ): out-of-range/NaN -> INT64_MIN */
 __device__ __forceinline__ int64_t go_f2i(double v) {
     if (!(v >= -9223372036854775808.0 && v < 9223372036854775808.0)) return INT64_MIN;
@@ -264,6 +277,8 @@ struct Decoder {
     BitReader r;
     int64_t prev_time, prev_time_delta;
     int64_t unit_ns; /* cached UNIT_NS_D[time_unit] (0 when unit invalid) */
+    double mult_pow;  /* cached 10^mult (mult changes rarely; keeping the
+                       * select chain out of the per-point value emit) */
     double int_val;
     uint64_t prev_float_bits, prev_xor;
     uint8_t time_unit, scheme_unit, mult, sig;
@@ -290,6 +305,7 @@ struct Decoder {
         prev_time = 0; prev_time_delta = 0;
         int_val = 0; prev_float_bits = 0; prev_xor = 0;
         time_unit = 0; unit_ns = 0; scheme_unit = 0; mult = 0; sig = 0;
+        mult_pow = 1.0;
         default_unit = dunit;
         have_scheme = false; tu_changed = false; done = false; is_float = false;
         int_optimized = intopt;
@@ -408,7 +424,7 @@ struct Decoder {
                 err = r.read_bits(vbits[i], &db);
                 if (err) return err;
                 if (!unit_valid(time_unit)) { *out = 0; return 0; }
-                *out = sign_extend(db, vbits[i]) * UNIT_NS_D[time_unit];
+                *out = sign_extend(db, vbits[i]) * unit_ns;
                 return 0;
             }
         }
@@ -417,7 +433,7 @@ struct Decoder {
         err = r.read_bits(dbits, &db);
         if (err) return err;
         if (!unit_valid(time_unit)) { *out = 0; return 0; }
-        *out = sign_extend(db, dbits) * UNIT_NS_D[time_unit];
+        *out = sign_extend(db, dbits) * unit_ns;
         return 0;
     }
 
@@ -485,7 +501,7 @@ struct Decoder {
             int err = r.read_bits(vb, &db);
             if (err) return err;
             if (!unit_valid(time_unit)) { *out = 0; return 0; } /* swallowed */
-            *out = sign_extend(db, vb) * UNIT_NS_D[time_unit];
+            *out = sign_extend(db, vb) * unit_ns;
             return 0;
         }
     }
@@ -596,6 +612,7 @@ struct Decoder {
             if (err) return err;
             mult = (uint8_t)m;
             if (mult > MAX_MULT) return M3GPU_SERIES_INVALID_MULT;
+            mult_pow = exp10_sel(mult);
         }
         return 0;
     }
@@ -790,7 +807,7 @@ struct Decoder {
                     nsig ? ((f << 1) >> ((64u - nsig) & 63u)) : 0;
                 int_val += (sbit ? 1.0 : -1.0) * (double)mag;
                 sig = nsig;
-                mult = nmult;
+                if (nmult != mult) { mult = nmult; mult_pow = exp10_sel(nmult); }
                 nb = 3u + off + moff + 1u + nsig;
             }
             r.consume(tsbits + nb);
@@ -800,7 +817,7 @@ struct Decoder {
             /* the f64 divide is ~25 VALU ops: branch it so mult==0 waves
              * (the common case) never pay for a select-discarded divide */
             double outv = int_val;
-            if (mult != 0) outv = int_val / exp10_table(mult);
+            if (mult != 0) outv = int_val / mult_pow;
             *v = outv;
             return 1;
         }
@@ -841,14 +858,17 @@ struct Decoder {
         if (err) return -err;
         *t = prev_time;
         if (!int_optimized || is_float) *v = bits2f(prev_float_bits);
-        else if (mult != 0) *v = int_val / exp10_table(mult);
+        else if (mult != 0) *v = int_val / mult_pow;
         else *v = int_val;
         points++;
         return 1;
     }
+    /* select chain, NOT an indexed array: a dynamically-indexed local
+     * const table lowers to a .rodata global_load whose vmcnt(0) wait
+     * (gfx9 vmcnt counts stores too) drains every in-flight output store
+     * right in the per-point value chain */
     __device__ __forceinline__ double exp10_table(uint8_t m) {
-        const double mult_tab[7] = {1.0, 10.0, 100.0, 1000.0, 10000.0, 100000.0, 1000000.0};
-        return mult_tab[m];
+        return exp10_sel(m);
     }
 };
 
@@ -873,9 +893,8 @@ struct Decoder {
 
 #define DEC_TILE 8
 
-__global__ void __launch_bounds__(BLOCK_THREADS, 3) /* 3 waves/SIMD: jointly
-    pinned by the 50.4 KB/block LDS (ring + tiles -> 3 blocks/CU) and the
-    136-168 VGPR allocation granule */
+__global__ void __launch_bounds__(BLOCK_THREADS, 4) /* hold the 128-VGPR
+    granule: ring + tiles = 40 KB/block -> 4 blocks/CU, 4 waves/SIMD */
 k_decode_batch(const uint8_t* __restrict__ blobs,
                const uint64_t* __restrict__ offsets,
                const uint32_t* __restrict__ lens,
@@ -1083,7 +1102,6 @@ __device__ __forceinline__ double go_next_up(double x) { /* x > 0 finite */
 
 __device__ __forceinline__ int convert_to_int_float(double v, uint8_t cur_max_mult,
                                     double* out_val, uint8_t* out_mult, bool* out_is_float) {
-    const double mult_tab[7] = {1.0, 10.0, 100.0, 1000.0, 10000.0, 100000.0, 1000000.0};
     const double MAXINT = 9223372036854775808.0;
     if (cur_max_mult == 0 && v < MAXINT) {
         double i, r;
@@ -1094,7 +1112,7 @@ __device__ __forceinline__ int convert_to_int_float(double v, uint8_t cur_max_mu
     double sign = 1.0;
     if (v < 0) sign = -1.0;
     for (uint8_t m = cur_max_mult; m <= MAX_MULT; m++) {
-        double val = v * mult_tab[m] * sign;
+        double val = v * exp10_sel(m) * sign;
         if (val >= 1e13) break;
         double i, r;
         r = go_modf(val, &i);
