@@ -414,6 +414,7 @@ struct Decoder {
         if (cb == 0) { *out = 0; return 0; }
         const uint32_t opcodes[3] = {0x2, 0x6, 0xe};
         const uint32_t vbits[3] = {7, 9, 12};
+#pragma unroll
         for (int i = 0; i < 3; i++) {
             uint64_t nxt;
             err = r.read_bits(1, &nxt);
@@ -1132,6 +1133,9 @@ __device__ __forceinline__ int convert_to_int_float(double v, uint8_t cur_max_mu
  * path has a fixed unit and no annotations (m3gpu.h contract). */
 struct Encoder {
     BitWriter w;
+    int64_t unit_ns; /* cached UNIT_NS_D[unit] — the bulk path has one
+                      * fixed unit; an indexed read per point would be a
+                      * .rodata global load + vmcnt(0) in the hot chain */
     int64_t prev_time, prev_time_delta;
     uint64_t prev_xor, prev_float_bits;
     double int_val;
@@ -1148,8 +1152,9 @@ struct Encoder {
         prev_xor = 0; prev_float_bits = 0;
         int_val = 0;
         /* initialTimeUnit (timestamp_encoder.go:248-259) */
-        time_unit = (unit_valid(default_unit) &&
-                     start_ns % UNIT_NS_D[default_unit] == 0) ? default_unit : 0;
+        unit_ns = unit_valid(default_unit) ? UNIT_NS_D[default_unit] : 0;
+        time_unit = (unit_valid(default_unit) && unit_ns != 0 &&
+                     start_ns % unit_ns == 0) ? default_unit : 0;
         max_mult = 0;
         num_sig_state = 0; cur_highest_lower_sig = 0; num_lower_sig = 0;
         has_written_first = false; is_float = false;
@@ -1165,8 +1170,7 @@ struct Encoder {
     /* timestamp_encoder.go:205-246 */
     __device__ __forceinline__ int write_dod_unchanged(int64_t prev_delta, int64_t cur_delta, uint8_t unit) {
         if (!unit_valid(unit)) return M3GPU_SERIES_NO_SCHEME;
-        int64_t u = UNIT_NS_D[unit];
-        int64_t dod = (cur_delta - prev_delta) / u;
+        int64_t dod = (cur_delta - prev_delta) / unit_ns;
         if (unit == 1 || unit == 2) {
             if ((int64_t)(int32_t)dod != dod) return M3GPU_SERIES_DOD_OVERFLOW;
         }
@@ -1176,6 +1180,9 @@ struct Encoder {
         const uint32_t opcodes[3] = {0x2, 0x6, 0xe};
         const uint32_t obits[3] = {2, 3, 4};
         const uint32_t vbits[3] = {7, 9, 12};
+        /* unrolled: rolled, the const tables become dynamically-indexed
+         * .rodata global loads per point */
+#pragma unroll
         for (int i = 0; i < 3; i++) {
             int64_t bmin = -((int64_t)1 << (vbits[i] - 1));
             int64_t bmax = ((int64_t)1 << (vbits[i] - 1)) - 1;
@@ -1201,6 +1208,7 @@ struct Encoder {
             write_marker(MARKER_TIMEUNIT);
             w.write_bits(unit, 8);
             time_unit = unit;
+            unit_ns = UNIT_NS_D[unit];
             tu_changed = true;
         }
         int64_t time_delta = cur_time - prev_time;
