@@ -1177,20 +1177,18 @@ struct Encoder {
         int dbits = scheme_default_bits(unit);
         if (!dbits) return M3GPU_SERIES_NO_SCHEME;
         if (dod == 0) { w.write_bits(0, 1); return 0; }
-        const uint32_t opcodes[3] = {0x2, 0x6, 0xe};
-        const uint32_t obits[3] = {2, 3, 4};
-        const uint32_t vbits[3] = {7, 9, 12};
-        /* unrolled: rolled, the const tables become dynamically-indexed
-         * .rodata global loads per point */
-#pragma unroll
-        for (int i = 0; i < 3; i++) {
-            int64_t bmin = -((int64_t)1 << (vbits[i] - 1));
-            int64_t bmax = ((int64_t)1 << (vbits[i] - 1)) - 1;
-            if (dod >= bmin && dod <= bmax) {
-                w.write_bits(opcodes[i], obits[i]);
-                w.write_bits((uint64_t)dod, vbits[i]);
-                return 0;
-            }
+        /* bucket select arithmetically (the table-lookup form vectorizes
+         * into per-lane .rodata global loads): dod fits b two's-complement
+         * bits iff bits(|folded|)+1 <= b */
+        const uint64_t m = (uint64_t)(dod < 0 ? ~dod : dod);
+        const uint32_t nb = (m ? 64u - (uint32_t)__builtin_clzll(m) : 0u) + 1u;
+        if (nb <= 12u) {
+            const uint32_t sel = (nb <= 7u) ? 0u : (nb <= 9u) ? 1u : 2u;
+            const uint32_t opc = (sel == 0) ? 0x2u : (sel == 1) ? 0x6u : 0xeu;
+            const uint32_t vb = (sel == 0) ? 7u : (sel == 1) ? 9u : 12u;
+            w.write_bits(opc, sel + 2u);
+            w.write_bits((uint64_t)dod, vb);
+            return 0;
         }
         w.write_bits(0xf, 4);
         w.write_bits((uint64_t)dod, dbits);
