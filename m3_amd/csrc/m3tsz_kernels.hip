@@ -1167,10 +1167,24 @@ struct Encoder {
         w.write_bits(marker, 2);
     }
 
+    /* unit_ns takes one of a few fixed values (unit.go:30-42); constant-
+     * divisor arms (the branch is wave-uniform) let the compiler emit
+     * magic-multiply sequences instead of a ~40-op runtime 64-bit
+     * division per encoded point */
+    __device__ __forceinline__ int64_t div_unit(int64_t x) const {
+        switch (unit_ns) {
+        case 1LL: return x;
+        case 1000LL: return x / 1000LL;
+        case 1000000LL: return x / 1000000LL;
+        case 1000000000LL: return x / 1000000000LL;
+        default: return x / unit_ns;
+        }
+    }
+
     /* timestamp_encoder.go:205-246 */
     __device__ __forceinline__ int write_dod_unchanged(int64_t prev_delta, int64_t cur_delta, uint8_t unit) {
         if (!unit_valid(unit)) return M3GPU_SERIES_NO_SCHEME;
-        int64_t dod = (cur_delta - prev_delta) / unit_ns;
+        int64_t dod = div_unit(cur_delta - prev_delta);
         if (unit == 1 || unit == 2) {
             if ((int64_t)(int32_t)dod != dod) return M3GPU_SERIES_DOD_OVERFLOW;
         }
