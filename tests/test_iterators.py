@@ -85,3 +85,24 @@ def test_iterator_annotations_sticky():
     it = SliceReaderIterator(ts, vals, 3)
     it.Next()
     assert it.CurrentAnnotation() is None
+
+
+def test_ann_region_helpers_cpu():
+    """parse_ann_region / decoded_ann_per_point are pure host logic: build
+    a region by hand (layout: m3gpu.h) and check both views."""
+    from m3_amd.engine import parse_ann_region, decoded_ann_per_point
+    region = np.zeros(128, np.uint8)
+    events = [(0, b"aa"), (3, b"bcd")]
+    tail = 128
+    evb = region[4:].view(np.uint32)
+    for j, (pt, data) in enumerate(events):
+        tail -= len(data)
+        region[tail:tail + len(data)] = np.frombuffer(data, np.uint8)
+        evb[j * 3:j * 3 + 3] = [pt, tail, len(data)]
+    region[:4].view(np.uint32)[0] = len(events)
+    assert parse_ann_region(region) == [(0, b"aa"), (3, b"bcd")]
+    assert decoded_ann_per_point(region, 5) == [b"aa", b"aa", b"aa", b"bcd",
+                                                b"bcd"]
+    empty = np.zeros(16, np.uint8)
+    assert parse_ann_region(empty) == []
+    assert decoded_ann_per_point(empty, 3) == [None, None, None]
