@@ -28,6 +28,7 @@
  * convertToIntFloat m3tsz.go:78-119, decode accumulation iterator.go:168-175).
  */
 #include <hip/hip_runtime.h>
+#include <atomic>
 #include <math.h>
 #include <string.h>
 #include <stdio.h>
@@ -2946,8 +2947,20 @@ int m3gpu_rollup_batch_dev_opts(
          * wave-per-series kernel. The common case is zero overflows: test
          * with a device-side count + 4-byte D2H instead of pulling and
          * scanning the whole error array every call. */
-        static uint32_t* d_ovf = nullptr;
-        if (!d_ovf) HIP_TRY(hipMalloc(&d_ovf, sizeof(uint32_t)));
+        /* process-lifetime scratch; atomics guard the one-time alloc
+         * (host entry points may be called from multiple threads) */
+        static std::atomic<uint32_t*> g_ovf{nullptr};
+        uint32_t* d_ovf = g_ovf.load(std::memory_order_acquire);
+        if (!d_ovf) {
+            uint32_t* fresh = nullptr;
+            HIP_TRY(hipMalloc(&fresh, sizeof(uint32_t)));
+            uint32_t* expected = nullptr;
+            if (!g_ovf.compare_exchange_strong(expected, fresh,
+                                               std::memory_order_acq_rel)) {
+                (void)hipFree(fresh); /* another thread won the race */
+            }
+            d_ovf = g_ovf.load(std::memory_order_acquire);
+        }
         HIP_TRY(hipMemsetAsync(d_ovf, 0, sizeof(uint32_t), s));
         hipLaunchKernelGGL(m3::k_count_errcode, dim3(256), dim3(BLOCK_THREADS),
                            0, s, d_out_errs, nseries,
