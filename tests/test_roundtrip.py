@@ -178,3 +178,43 @@ def test_batch_matches_series():
     for i in range(0, nseries, 17):
         single = oracle.encode_series(ts[i], vals[i], start_ns=int(ts[i, 0]))
         assert bytes(rows[i, :lens[i]]) == single
+
+
+def test_pack_streams_64b_alignment():
+    """pack_streams emits 64B-aligned offsets (one HBM line per stream
+    chunk — m3gpu.h layout note); the ABI minimum is 8B."""
+    from m3_amd.engine import pack_streams
+    streams = [b"\x01" * n for n in (1, 63, 64, 65, 129, 0, 7)]
+    blob, offsets, lens = pack_streams(streams)
+    assert np.all(offsets % 64 == 0)
+    assert int(offsets[-1]) == len(blob)
+    for i, s in enumerate(streams):
+        o = int(offsets[i])
+        assert bytes(blob[o:o + len(s)]) == s
+        # zero pad up to the next stream
+        assert not blob[o + len(s):int(offsets[i + 1])].any()
+
+
+def test_oracle_roundtrip_annotations_and_unit_changes():
+    """Oracle self-check: annotations interleaved with time-unit changes
+    round-trip exactly (annotation markers chain with timeunit markers —
+    tryReadMarker loops, timestamp_iterator.go:174-235)."""
+    rng = np.random.default_rng(31337)
+    start = 1427162462 * 10**9
+    n = 60
+    ts = start + np.cumsum(rng.integers(1, 50, n)) * 10**9
+    # switch units mid-stream: s -> ms -> s
+    units = np.full(n, 1, np.uint8)
+    units[20:40] = 2
+    ts[20:40] = (ts[20:40] // 10**6) * 10**6
+    vals = np.round(rng.random(n) * 100, 2)
+    anns = [None] * n
+    anns[0], anns[19], anns[20], anns[45] = b"a0", b"pre", b"at-switch", b"zz"
+    blob = oracle.encode_series(ts, vals, units=units, annotations=anns,
+                                start_ns=start)
+    dec = oracle.decode_series(blob, with_annotations=True)
+    assert np.array_equal(dec["ts"], ts)
+    assert np.array_equal(dec["vals"], vals)
+    assert np.array_equal(dec["units"], units)
+    got = [(i, a) for i, a in enumerate(dec["annotations"]) if a is not None]
+    assert got == [(0, b"a0"), (19, b"pre"), (20, b"at-switch"), (45, b"zz")]
