@@ -256,3 +256,32 @@ def test_unagg_handhex():
     assert m["counter_value"] == 42
     assert m["metadatas"] == md
     assert m["metadata_fields"] == [(2, md)]
+
+
+def test_commitlog_handhex_annotation_and_second_entry(tmp_path):
+    """Hand-hex continued: a second LogEntry for a registered series
+    carries only the index (no metadata re-send) plus an annotation
+    (encoder.go:435-445 field 7); the reader must attach both points to
+    the one series in log order."""
+    ts1 = BLOCK_START + 5 * 10**9
+    ts2 = ts1 + 10**9
+    log_info = root(T_LOG_INFO) + mparr(3) + mpu(0) + mpu(0) + mpu(3)
+    metadata = (root(T_LOG_METADATA) + mparr(3) +
+                mpb(b"cl.ann") + mpb(b"ns1") + mpu(2) + mpb(None))
+    e1 = (root(T_LOG_ENTRY) + mparr(7) + mpu(9) + mpu(0) + mpb(metadata) +
+          mpu(ts1) + mpf64(1.5) + mpu(4) + mpb(None))
+    e2 = (root(T_LOG_ENTRY) + mparr(7) + mpu(9) + mpu(0) + mpb(None) +
+          mpu(ts2) + mpf64(2.5) + mpu(4) + mpb(b"note"))
+    stream = b"".join(uvarint(len(r)) + r for r in (log_info, e1, e2))
+    path = tmp_path / "commitlog-0-3.db"
+    path.write_bytes(chunk(stream))
+    with CommitLog(path) as cl:
+        series = cl.series()
+        assert len(series) == 1
+        s = series[0]
+        assert s["id"] == b"cl.ann" and s["namespace"] == b"ns1"
+        assert s["shard"] == 2 and s["tags"] == b""
+        assert list(s["ts"]) == [ts1, ts2]
+        assert list(s["vals"]) == [1.5, 2.5]
+        # annotations come back as [(point_idx, bytes)]
+        assert s["annotations"] == [(1, b"note")]
